@@ -1,0 +1,35 @@
+"""Shared building helpers: run nn.Conv1d / nn.BatchNorm1d parameter
+containers through the MI355X op layer."""
+
+import torch.nn as nn
+
+from .. import ops
+
+
+def run_conv(conv: nn.Conv1d, x, padl: int = 0, padr: int = 0):
+    """Apply an nn.Conv1d's parameters via the native conv path with
+    explicit pre-padding (module padding is added on top)."""
+    p = conv.padding[0] if isinstance(conv.padding, tuple) else int(conv.padding)
+    return ops.conv1d(
+        x, conv.weight, conv.bias,
+        stride=conv.stride[0],
+        padding=(padl + p, padr + p),
+        groups=conv.groups,
+        dilation=conv.dilation[0],
+    )
+
+
+def run_bn(bn, x, act: str = "none"):
+    """Fused BatchNorm1d (+act); falls through for Identity/other norms."""
+    if isinstance(bn, nn.BatchNorm1d):
+        y = ops.bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+                       bn.training, bn.momentum, bn.eps, act=act)
+        if bn.training and bn.track_running_stats:
+            bn.num_batches_tracked += 1
+        return y
+    y = bn(x)
+    if act == "relu":
+        y = y.relu()
+    elif act == "gelu":
+        y = ops.gelu(y)
+    return y

@@ -1,0 +1,359 @@
+"""Functional op layer: one API, two implementations.
+
+* GPU (ROCm/MI355X): hand-written HIP kernels from ``seist_amd._C``;
+  dispatch is hard — a CUDA tensor hitting an op whose kernel exists must
+  run the kernel (no silent eager fallback).
+* CPU: plain PyTorch fp32 reference with identical semantics, used by the
+  numerics tests as ground truth.
+
+Layout convention is channels-first ``(N, C, L)`` throughout, matching the
+reference design (models/seist.py:106-107 uses 1x1 Conv1d instead of Linear
+to avoid transposes — here the 1x1 conv IS a GEMM on MFMA).
+"""
+
+import math
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import has_ext, ext, use_native
+
+_SQRT_2PI = math.sqrt(2.0 * math.pi)
+
+_ACT_NONE = 0
+_ACT_GELU = 1
+_ACT_RELU = 2
+
+# ---------------------------------------------------------------------------
+# padding semantics (reference models/seist.py:12-48 `_auto_pad_1d`)
+# ---------------------------------------------------------------------------
+
+
+def auto_pad_lr(length: int, kernel_size: int, stride: int = 1) -> Tuple[int, int]:
+    """Left/right padding so conv output length is ceil(length / stride)."""
+    assert kernel_size >= stride, (
+        f"`kernel_size` must be >= `stride`, got {kernel_size}, {stride}"
+    )
+    pds = (stride - (length % stride)) % stride + kernel_size - stride
+    return pds // 2, pds - pds // 2
+
+
+def auto_pad(x: torch.Tensor, kernel_size: int, stride: int = 1,
+             value: float = 0.0) -> torch.Tensor:
+    """'same-for-strided' padding of the last dim (out = ceil(L/stride))."""
+    pl, pr = auto_pad_lr(x.size(-1), kernel_size, stride)
+    if pl == 0 and pr == 0:
+        return x
+    return F.pad(x, (pl, pr), "constant", value)
+
+
+# ---------------------------------------------------------------------------
+# gelu (erf form — PyTorch nn.GELU default)
+# ---------------------------------------------------------------------------
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    return F.gelu(x)
+
+
+def _gelu_grad(x: torch.Tensor) -> torch.Tensor:
+    cdf = 0.5 * (1.0 + torch.erf(x * (1.0 / math.sqrt(2.0))))
+    pdf = torch.exp(-0.5 * x * x) / _SQRT_2PI
+    return cdf + x * pdf
+
+
+# ---------------------------------------------------------------------------
+# pointwise (1x1) conv == batched GEMM  (MFMA kernel on GPU)
+# ---------------------------------------------------------------------------
+
+
+class _PointwiseConv(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        # x: (N, Ci, L), weight: (Co, Ci), bias: (Co,) or None
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        if use_native(x):
+            return ext().pw_conv_fwd(x, weight, bias)
+        y = torch.einsum("oc,ncl->nol", weight, x)
+        if bias is not None:
+            y = y + bias[:, None]
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_native(x):
+            dx, dw, db = ext().pw_conv_bwd(dy, x, weight, ctx.has_bias)
+        else:
+            dx = torch.einsum("oc,nol->ncl", weight, dy)
+            dw = torch.einsum("nol,ncl->oc", dy, x)
+            db = dy.sum(dim=(0, 2)) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def pointwise_conv(x: torch.Tensor, weight: torch.Tensor,
+                   bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """1x1 Conv1d: (N,Ci,L) x (Co,Ci[,1]) -> (N,Co,L)."""
+    if weight.dim() == 3:
+        weight = weight.squeeze(-1)
+    return _PointwiseConv.apply(x.contiguous(), weight.contiguous(), bias)
+
+
+# ---------------------------------------------------------------------------
+# general direct conv1d (depthwise / grouped / dense, strided, pre-padded)
+# ---------------------------------------------------------------------------
+
+
+class _Conv1d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padl, padr, groups, dilation):
+        ctx.save_for_backward(x, weight)
+        ctx.conf = (stride, padl, padr, groups, dilation, bias is not None)
+        if use_native(x):
+            return ext().conv1d_fwd(x, weight, bias, stride, padl, padr,
+                                    groups, dilation)
+        xp = F.pad(x, (padl, padr)) if (padl or padr) else x
+        return F.conv1d(xp, weight, bias, stride=stride, groups=groups,
+                        dilation=dilation)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        stride, padl, padr, groups, dilation, has_bias = ctx.conf
+        dy = dy.contiguous()
+        if use_native(x):
+            dx, dw, db = ext().conv1d_bwd(dy, x, weight, stride, padl, padr,
+                                          groups, dilation, has_bias)
+        else:
+            xp = F.pad(x, (padl, padr)) if (padl or padr) else x
+            xp = xp.detach().requires_grad_(True)
+            w = weight.detach().requires_grad_(True)
+            with torch.enable_grad():
+                out = F.conv1d(xp, w, None, stride=stride, groups=groups,
+                               dilation=dilation)
+            dxp, dw = torch.autograd.grad(out, [xp, w], dy)
+            Lp = dxp.size(-1)
+            dx = dxp[..., padl: Lp - padr] if (padl or padr) else dxp
+            db = dy.sum(dim=(0, 2)) if has_bias else None
+        return dx, dw, db, None, None, None, None, None
+
+
+def conv1d(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None, stride: int = 1,
+           padding: Tuple[int, int] = (0, 0), groups: int = 1,
+           dilation: int = 1) -> torch.Tensor:
+    """Direct Conv1d with explicit (left, right) padding.
+
+    weight: (Co, Ci/groups, K). Covers dense, grouped, depthwise
+    (groups == Ci) and dilated-causal convolutions — the K2/K3/K4/K6 kernel
+    family of SURVEY §2.4.
+    """
+    padl, padr = padding
+    if (weight.size(-1) == 1 and stride == 1 and padl == 0 and padr == 0
+            and groups == 1):
+        return pointwise_conv(x, weight, bias)
+    return _Conv1d.apply(x.contiguous(), weight.contiguous(), bias, stride,
+                         padl, padr, groups, dilation)
+
+
+# ---------------------------------------------------------------------------
+# fused BatchNorm1d (+ optional GELU) over (N, C, L)
+# ---------------------------------------------------------------------------
+
+
+class _BNAct(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training,
+                momentum, eps, act):
+        if use_native(x):
+            y, mean, invstd = ext().bn_act_fwd(
+                x, gamma, beta, running_mean, running_var, training, momentum,
+                eps, act)
+        else:
+            x32 = x.float()
+            if training:
+                mean = x32.mean(dim=(0, 2))
+                var = x32.var(dim=(0, 2), unbiased=False)
+                n = x.size(0) * x.size(2)
+                if running_mean is not None:
+                    with torch.no_grad():
+                        running_mean.mul_(1 - momentum).add_(momentum * mean)
+                        unbiased = var * (n / max(n - 1, 1))
+                        running_var.mul_(1 - momentum).add_(momentum * unbiased)
+            else:
+                mean = running_mean.float()
+                var = running_var.float()
+            invstd = torch.rsqrt(var + eps)
+            xhat = (x32 - mean[:, None]) * invstd[:, None]
+            pre = xhat * gamma.float()[:, None] + beta.float()[:, None]
+            if act == _ACT_GELU:
+                y = F.gelu(pre)
+            elif act == _ACT_RELU:
+                y = F.relu(pre)
+            else:
+                y = pre
+            y = y.to(x.dtype)
+        ctx.save_for_backward(x, gamma, beta, mean, invstd)
+        ctx.training = training
+        ctx.act = act
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, beta, mean, invstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_native(x):
+            dx, dgamma, dbeta = ext().bn_act_bwd(
+                dy, x, gamma, beta, mean, invstd, ctx.training, ctx.act)
+        else:
+            x32 = x.float()
+            dy32 = dy.float()
+            g = gamma.float()[:, None]
+            b = beta.float()[:, None]
+            xhat = (x32 - mean[:, None]) * invstd[:, None]
+            if ctx.act != _ACT_NONE:
+                pre = xhat * g + b
+                if ctx.act == _ACT_GELU:
+                    dy32 = dy32 * _gelu_grad(pre)
+                else:
+                    dy32 = dy32 * (pre > 0).to(dy32.dtype)
+            dgamma = (dy32 * xhat).sum(dim=(0, 2))
+            dbeta = dy32.sum(dim=(0, 2))
+            if ctx.training:
+                n = x.size(0) * x.size(2)
+                dx = (g * invstd[:, None] / n) * (
+                    n * dy32 - dbeta[:, None] - xhat * dgamma[:, None]
+                )
+            else:
+                dx = dy32 * g * invstd[:, None]
+            dx = dx.to(x.dtype)
+            dgamma = dgamma.to(gamma.dtype)
+            dbeta = dbeta.to(beta.dtype)
+        return dx, dgamma, dbeta, None, None, None, None, None, None
+
+
+def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+           running_mean: Optional[torch.Tensor],
+           running_var: Optional[torch.Tensor], training: bool,
+           momentum: float = 0.1, eps: float = 1e-5,
+           act: str = "none") -> torch.Tensor:
+    """Fused BatchNorm1d (+GELU) — K7/K14 of SURVEY §2.4.
+
+    BN statistics and parameters are fp32 regardless of activation dtype.
+    """
+    act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
+    return _BNAct.apply(x.contiguous(), gamma, beta, running_mean, running_var,
+                        training, momentum, eps, act_id)
+
+
+# ---------------------------------------------------------------------------
+# fused avg+max pool (ceil mode) — the LocalAwareAggregation primitive
+# ---------------------------------------------------------------------------
+
+
+class _AvgMaxPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k):
+        ctx.k = k
+        ctx.in_len = x.size(-1)
+        if use_native(x):
+            y, idx = ext().avgmax_pool_fwd(x, k)
+            ctx.save_for_backward(idx)
+            return y
+        ya = F.avg_pool1d(x, k, ceil_mode=True)
+        ym, idx = F.max_pool1d(x.float(), k, ceil_mode=True, return_indices=True)
+        ctx.save_for_backward(idx)
+        return ya + ym.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        k, in_len = ctx.k, ctx.in_len
+        if dy.is_cuda and has_ext():
+            dx = ext().avgmax_pool_bwd(dy, idx, k, in_len)
+            return dx, None
+        N, C, Lo = dy.shape
+        dx = dy.new_zeros(N, C, in_len)
+        # max part
+        dx.view(N, C, in_len).scatter_add_(2, idx, dy)
+        # avg part: each output window [i*k, min((i+1)k, L)) gets dy/len
+        for i in range(Lo):
+            lo = i * k
+            hi = min(lo + k, in_len)
+            dx[:, :, lo:hi] += (dy[:, :, i] / (hi - lo))[:, :, None]
+        return dx, None
+
+
+def avgmax_pool1d(x: torch.Tensor, k: int) -> torch.Tensor:
+    """avg_pool1d(x,k,ceil) + max_pool1d(x,k,ceil) in one kernel (K12)."""
+    if k <= 1:
+        # reference LocalAwareAggregationBlock builds no pools for k == 1
+        # (models/seist.py:79-84) — identity, not avg+max.
+        return x
+    return _AvgMaxPool.apply(x.contiguous(), k)
+
+
+# ---------------------------------------------------------------------------
+# linear interpolation resize (F.interpolate mode='linear', align_corners=False)
+# ---------------------------------------------------------------------------
+
+
+class _InterpLinear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, out_len):
+        ctx.in_len = x.size(-1)
+        if use_native(x):
+            return ext().interp_linear_fwd(x, out_len)
+        return F.interpolate(x, size=out_len, mode="linear", align_corners=False)
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous()
+        if dy.is_cuda and has_ext():
+            return ext().interp_linear_bwd(dy, ctx.in_len), None
+        # CPU: use autograd of F.interpolate
+        x = dy.new_zeros(dy.size(0), dy.size(1), ctx.in_len).requires_grad_(True)
+        with torch.enable_grad():
+            y = F.interpolate(x, size=dy.size(-1), mode="linear",
+                              align_corners=False)
+        (dx,) = torch.autograd.grad(y, [x], dy)
+        return dx, None
+
+
+def interp_linear(x: torch.Tensor, out_len: int) -> torch.Tensor:
+    """K13: 1d linear resize to arbitrary size."""
+    if out_len == x.size(-1):
+        return x
+    return _InterpLinear.apply(x.contiguous(), out_len)
+
+
+# ---------------------------------------------------------------------------
+# pooled-KV attention (K9): softmax((q/sqrt(E))^T k) @ v^T
+# ---------------------------------------------------------------------------
+
+
+def pooled_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     attn_dropout: float = 0.0,
+                     training: bool = False) -> torch.Tensor:
+    """SeisT pooled-KV attention core (reference models/seist.py:368-393).
+
+    q: (N, H, E, Lq) from the full-length sequence; k/v: (N, H, E, Lk) from
+    the aggregated sequence (Lk == Lq / aggr_ratio; 128 at every stage for
+    the published configs). Returns (N, H, E, Lq).
+
+    Runs as batched GEMM + fused softmax; a fully fused single-kernel HIP
+    path is selected when available.
+    """
+    E = q.size(2)
+    if use_native(q) and hasattr(ext(), "pooled_attn_fwd") and not training:
+        return ext().pooled_attn_fwd(q, k, v)
+    attn = torch.matmul(q.transpose(-1, -2), k) * (1.0 / math.sqrt(E))
+    attn = attn.softmax(dim=-1)
+    if attn_dropout > 0.0 and training:
+        attn = F.dropout(attn, p=attn_dropout, training=True)
+    out = torch.matmul(attn, v.transpose(-1, -2)).transpose(-1, -2)
+    return out

@@ -1,0 +1,47 @@
+"""Running-average meters and formatted progress lines.
+
+Parity with /root/reference/utils/meters.py:4-45.
+"""
+
+
+class AverageMeter:
+    """Weighted running average of a scalar."""
+
+    def __init__(self, name: str, fmt: str = ":f"):
+        self.name = name
+        self.fmt = fmt
+        self.reset()
+
+    def reset(self) -> None:
+        self.val = 0.0
+        self.sum = 0.0
+        self.count = 0
+
+    def update(self, val: float, n: int = 1) -> None:
+        self.val = val
+        self.sum += val * n
+        self.count += n
+
+    @property
+    def avg(self) -> float:
+        return self.sum / self.count if self.count else 0.0
+
+    def __str__(self) -> str:
+        return ("{name} {val" + self.fmt + "} ({avg" + self.fmt + "})").format(
+            name=self.name, val=self.val, avg=self.avg
+        )
+
+
+class ProgressMeter:
+    """Formats a step line out of a set of meters."""
+
+    def __init__(self, num_batches: int, meters, prefix: str = ""):
+        fmt = "{:" + str(len(str(num_batches))) + "d}"
+        self.batch_fmtstr = "[" + fmt + "/" + fmt.format(num_batches) + "]"
+        self.meters = meters
+        self.prefix = prefix
+
+    def display(self, batch: int) -> str:
+        entries = [self.prefix + self.batch_fmtstr.format(batch)]
+        entries += [str(m) for m in self.meters]
+        return "  ".join(entries)
