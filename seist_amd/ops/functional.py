@@ -99,6 +99,8 @@ def pointwise_conv(x: torch.Tensor, weight: torch.Tensor,
     """1x1 Conv1d: (N,Ci,L) x (Co,Ci[,1]) -> (N,Co,L)."""
     if weight.dim() == 3:
         weight = weight.squeeze(-1)
+    if x.dtype != weight.dtype:
+        x = x.to(weight.dtype)
     return _PointwiseConv.apply(x.contiguous(), weight.contiguous(), bias)
 
 
@@ -152,6 +154,8 @@ def conv1d(x: torch.Tensor, weight: torch.Tensor,
     family of SURVEY §2.4.
     """
     padl, padr = padding
+    if x.dtype != weight.dtype:
+        x = x.to(weight.dtype)
     if (weight.size(-1) == 1 and stride == 1 and padl == 0 and padr == 0
             and groups == 1):
         return pointwise_conv(x, weight, bias)

@@ -1,0 +1,16 @@
+from .logger import logger
+from .meters import AverageMeter, ProgressMeter
+from .misc import (
+    cal_snr,
+    count_parameters,
+    get_safe_path,
+    get_time_str,
+    setup_seed,
+    strfargs,
+    strftimedelta,
+)
+
+__all__ = [
+    "logger", "AverageMeter", "ProgressMeter", "cal_snr", "count_parameters",
+    "get_safe_path", "get_time_str", "setup_seed", "strfargs", "strftimedelta",
+]

@@ -56,6 +56,9 @@ class _Logger:
     def get_logdir(self) -> Optional[str]:
         return self._logdir
 
+    def logdir(self) -> Optional[str]:
+        return self._logdir
+
     def set_logger(self, name: str) -> None:
         self._ensure(name)
         self._active = name

@@ -45,3 +45,7 @@ class ProgressMeter:
         entries = [self.prefix + self.batch_fmtstr.format(batch)]
         entries += [str(m) for m in self.meters]
         return "  ".join(entries)
+
+    def get_str(self, batch_idx: int, name: str = "") -> str:
+        s = self.display(batch_idx)
+        return f"({name}) {s}" if name else s
