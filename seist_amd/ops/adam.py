@@ -1,15 +1,14 @@
 """Fused Adam/AdamW optimizer (K17 of SURVEY §2.4).
 
-Single multi-tensor HIP kernel per step on GPU (the SeisT models have
-<= 1.1 M parameters spread over hundreds of small tensors — eager Adam is
-pure launch overhead). Keeps fp32 master weights and fp32 moments when the
-model parameters are bf16, so bf16 training matches fp32 Adam trajectories
-to bf16 rounding.
+On GPU the whole update is one kernel launch per parameter dtype over
+device-resident chunk metadata packed once (pointers are stable across
+steps; repacked automatically if any grad pointer moves). Keeps fp32
+master weights and fp32 moments when parameters are bf16, so bf16
+training follows fp32 Adam trajectories to bf16 rounding.
 
-CPU path is the same math in plain PyTorch (used for tests).
+CPU path is the same math in plain PyTorch (used by tests as reference).
 """
 
-import math
 from typing import Optional
 
 import torch
@@ -23,6 +22,28 @@ class FusedAdam(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay, adamw=adamw)
         super().__init__(params, defaults)
+        self._packed = None  # list of (meta, sample, has_master, ptr_sig)
+
+    def _collect(self, group):
+        params, grads, ms, vs, masters, steps = [], [], [], [], [], []
+        for p in group["params"]:
+            if p.grad is None:
+                continue
+            state = self.state[p]
+            if len(state) == 0:
+                state["step"] = 0
+                state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                if p.dtype != torch.float32:
+                    state["master"] = p.detach().clone().float()
+            state["step"] += 1
+            params.append(p)
+            grads.append(p.grad)
+            ms.append(state["exp_avg"])
+            vs.append(state["exp_avg_sq"])
+            masters.append(state.get("master", None))
+            steps.append(state["step"])
+        return params, grads, ms, vs, masters, steps
 
     @torch.no_grad()
     def step(self, closure: Optional[callable] = None):
@@ -37,56 +58,65 @@ class FusedAdam(torch.optim.Optimizer):
             eps = group["eps"]
             wd = group["weight_decay"]
             adamw = group["adamw"]
-
-            params, grads, ms, vs, masters, steps = [], [], [], [], [], []
-            for p in group["params"]:
-                if p.grad is None:
-                    continue
-                state = self.state[p]
-                if len(state) == 0:
-                    state["step"] = 0
-                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
-                    if p.dtype != torch.float32:
-                        state["master"] = p.detach().clone().float()
-                state["step"] += 1
-                params.append(p)
-                grads.append(p.grad)
-                ms.append(state["exp_avg"])
-                vs.append(state["exp_avg_sq"])
-                masters.append(state.get("master", None))
-                steps.append(state["step"])
-
+            params, grads, ms, vs, masters, steps = self._collect(group)
             if not params:
                 continue
 
-            if params[0].is_cuda and has_ext() and hasattr(ext(), "adam_step"):
-                # steps are uniform within a group after the first call
-                step = steps[0]
-                bc1 = 1.0 - beta1 ** step
-                bc2 = 1.0 - beta2 ** step
-                master_list = [m if m is not None else p
-                               for p, m in zip(params, masters)]
-                has_master = masters[0] is not None
-                ext().adam_step(params, grads, ms, vs, master_list,
-                                has_master, lr, beta1, beta2, eps, wd,
-                                bc1, bc2, adamw)
+            native = (params[0].is_cuda and has_ext()
+                      and hasattr(ext(), "adam_step_packed")
+                      and len(set(steps)) == 1)
+            if native:
+                self._native_step(params, grads, ms, vs, masters, steps[0],
+                                  lr, beta1, beta2, eps, wd, adamw)
             else:
-                for p, g, m, v, master, step in zip(params, grads, ms, vs,
-                                                    masters, steps):
-                    w = master if master is not None else p
-                    g32 = g.float()
-                    if wd != 0.0:
-                        if adamw:
-                            w.mul_(1.0 - lr * wd)
-                        else:
-                            g32 = g32.add(w, alpha=wd)
-                    m.mul_(beta1).add_(g32, alpha=1 - beta1)
-                    v.mul_(beta2).addcmul_(g32, g32, value=1 - beta2)
-                    bc1 = 1.0 - beta1 ** step
-                    bc2 = 1.0 - beta2 ** step
-                    denom = (v / bc2).sqrt_().add_(eps)
-                    w.addcdiv_(m, denom, value=-lr / bc1)
-                    if master is not None:
-                        p.copy_(w.to(p.dtype))
+                self._torch_step(params, grads, ms, vs, masters, steps,
+                                 lr, beta1, beta2, eps, wd, adamw)
         return loss
+
+    def _native_step(self, params, grads, ms, vs, masters, step, lr, beta1,
+                     beta2, eps, wd, adamw):
+        bc1 = 1.0 - beta1 ** step
+        bc2 = 1.0 - beta2 ** step
+        # partition by dtype (bf16 convs + fp32 norms can share one group)
+        parts = {}
+        for p, g, m, v, mst in zip(params, grads, ms, vs, masters):
+            parts.setdefault(p.dtype, []).append((p, g, m, v, mst))
+        sig = tuple((g.data_ptr(), p.data_ptr())
+                    for p, g, *_ in
+                    [t for lst in parts.values() for t in lst])
+        if self._packed is None or self._packed[0] != sig:
+            packed = []
+            for dtype, lst in parts.items():
+                ps = [t[0] for t in lst]
+                gs = [t[1] for t in lst]
+                mms = [t[2] for t in lst]
+                vvs = [t[3] for t in lst]
+                has_master = dtype != torch.float32
+                msts = [t[4] if t[4] is not None else t[0] for t in lst]
+                meta = ext().adam_pack(ps, gs, mms, vvs, msts, has_master)
+                packed.append((meta, ps[0], has_master))
+            self._packed = (sig, packed)
+        for meta, sample, has_master in self._packed[1]:
+            ext().adam_step_packed(meta, sample, has_master, lr, beta1,
+                                   beta2, eps, wd, bc1, bc2, adamw)
+
+    @staticmethod
+    def _torch_step(params, grads, ms, vs, masters, steps, lr, beta1, beta2,
+                    eps, wd, adamw):
+        for p, g, m, v, master, step in zip(params, grads, ms, vs, masters,
+                                            steps):
+            w = master if master is not None else p
+            g32 = g.float()
+            if wd != 0.0:
+                if adamw:
+                    w.mul_(1.0 - lr * wd)
+                else:
+                    g32 = g32.add(w, alpha=wd)
+            m.mul_(beta1).add_(g32, alpha=1 - beta1)
+            v.mul_(beta2).addcmul_(g32, g32, value=1 - beta2)
+            bc1 = 1.0 - beta1 ** step
+            bc2 = 1.0 - beta2 ** step
+            denom = (v / bc2).sqrt_().add_(eps)
+            w.addcdiv_(m, denom, value=-lr / bc1)
+            if master is not None:
+                p.copy_(w.to(p.dtype))

@@ -1,0 +1,59 @@
+// Python bindings for the seist_amd CDNA4 (gfx950) kernel library.
+
+#include <torch/extension.h>
+
+at::Tensor pw_conv_fwd(const at::Tensor& x, const at::Tensor& w,
+                       const c10::optional<at::Tensor>& bias);
+std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                    const at::Tensor& w, bool has_bias);
+
+at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias, long stride,
+                      long padl, long padr, long groups, long dilation);
+std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                   const at::Tensor& w, long stride,
+                                   long padl, long padr, long groups,
+                                   long dilation, bool has_bias);
+
+std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                                   const at::Tensor& beta,
+                                   const c10::optional<at::Tensor>& running_mean,
+                                   const c10::optional<at::Tensor>& running_var,
+                                   bool training, double momentum, double eps,
+                                   long act);
+std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                   const at::Tensor& gamma,
+                                   const at::Tensor& beta,
+                                   const at::Tensor& mean,
+                                   const at::Tensor& invstd, bool training,
+                                   long act);
+
+std::vector<at::Tensor> avgmax_pool_fwd(const at::Tensor& x, long k);
+at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
+                           long k, long in_len);
+at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len);
+at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len);
+
+at::Tensor adam_pack(std::vector<at::Tensor> params,
+                     std::vector<at::Tensor> grads,
+                     std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                     std::vector<at::Tensor> masters, bool has_master);
+void adam_step_packed(const at::Tensor& meta, const at::Tensor& sample,
+                      bool has_master, double lr, double beta1, double beta2,
+                      double eps, double wd, double bc1, double bc2,
+                      bool adamw);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("pw_conv_fwd", &pw_conv_fwd, "pointwise conv forward (MFMA GEMM)");
+  m.def("pw_conv_bwd", &pw_conv_bwd, "pointwise conv backward");
+  m.def("conv1d_fwd", &conv1d_fwd, "direct conv1d forward");
+  m.def("conv1d_bwd", &conv1d_bwd, "direct conv1d backward");
+  m.def("bn_act_fwd", &bn_act_fwd, "fused batchnorm+act forward");
+  m.def("bn_act_bwd", &bn_act_bwd, "fused batchnorm+act backward");
+  m.def("avgmax_pool_fwd", &avgmax_pool_fwd, "fused avg+max pool forward");
+  m.def("avgmax_pool_bwd", &avgmax_pool_bwd, "fused avg+max pool backward");
+  m.def("interp_linear_fwd", &interp_linear_fwd, "linear interp forward");
+  m.def("interp_linear_bwd", &interp_linear_bwd, "linear interp backward");
+  m.def("adam_pack", &adam_pack, "pack fused-adam chunk metadata");
+  m.def("adam_step_packed", &adam_step_packed, "fused adam step");
+}

@@ -1,0 +1,275 @@
+// Fused BatchNorm1d (+GELU/ReLU) over (N, C, L) — K7 + K14 of
+// SURVEY.md §2.4. Replaces the reference's norm_layer+act pairs
+// (models/seist.py:145-154, 223-224, phasenet/eqtransformer BNs).
+//
+// Stats, parameters and running buffers are fp32 regardless of activation
+// dtype. Training forward = one split-reduction kernel (atomics into a
+// per-channel fp32 scratch) + one finalize kernel + one fused
+// normalize+activation pass; eval forward is a single pass. Backward
+// mirrors that (reduce dgamma/dbeta, then one elementwise dx pass).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "sa_common.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+template <typename scalar_t>
+__global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
+                               float* __restrict__ sums,  // (C, 2)
+                               int C, long NL, long L) {
+  __shared__ float red[kBlock / sa::kWave];
+  const int c = blockIdx.x;
+  const int split = blockIdx.y;
+  const int nsplit = gridDim.y;
+  const long chunk = (NL + nsplit - 1) / nsplit;
+  const long t0 = (long)split * chunk;
+  const long t1 = min(NL, t0 + chunk);
+
+  float s = 0.0f, s2 = 0.0f;
+  for (long t = t0 + threadIdx.x; t < t1; t += kBlock) {
+    const long n = t / L;
+    const long l = t - n * L;
+    const float v = (float)x[(n * C + c) * L + l];
+    s += v;
+    s2 += v * v;
+  }
+  s = sa::block_reduce_sum(s, red);
+  __syncthreads();
+  s2 = sa::block_reduce_sum(s2, red);
+  if (threadIdx.x == 0) {
+    atomicAdd(&sums[c * 2 + 0], s);
+    atomicAdd(&sums[c * 2 + 1], s2);
+  }
+}
+
+__global__ void bn_finalize_kernel(const float* __restrict__ sums,
+                                   float* __restrict__ mean,
+                                   float* __restrict__ invstd,
+                                   float* __restrict__ running_mean,
+                                   float* __restrict__ running_var,
+                                   int C, long NL, float momentum,
+                                   float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float m = sums[c * 2] / NL;
+  const float var = fmaxf(sums[c * 2 + 1] / NL - m * m, 0.0f);
+  mean[c] = m;
+  invstd[c] = rsqrtf(var + eps);
+  if (running_mean != nullptr) {
+    const float unbiased = var * ((float)NL / fmaxf((float)(NL - 1), 1.0f));
+    running_mean[c] = (1.0f - momentum) * running_mean[c] + momentum * m;
+    running_var[c] = (1.0f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
+template <typename scalar_t>
+__global__ void bn_apply_kernel(const scalar_t* __restrict__ x,
+                                scalar_t* __restrict__ y,
+                                const float* __restrict__ mean,
+                                const float* __restrict__ invstd,
+                                const float* __restrict__ gamma,
+                                const float* __restrict__ beta,
+                                int C, long L, long total, int act) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= total) return;
+  const int c = (int)((i / L) % C);
+  const float xh = ((float)x[i] - mean[c]) * invstd[c];
+  const float pre = xh * gamma[c] + beta[c];
+  y[i] = (scalar_t)sa::act_fwd(pre, act);
+}
+
+template <typename scalar_t>
+__global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
+                                   const scalar_t* __restrict__ x,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ invstd,
+                                   const float* __restrict__ gamma,
+                                   const float* __restrict__ beta,
+                                   float* __restrict__ out,  // (C,2): dbeta,dgamma
+                                   int C, long NL, long L, int act) {
+  __shared__ float red[kBlock / sa::kWave];
+  const int c = blockIdx.x;
+  const int split = blockIdx.y;
+  const int nsplit = gridDim.y;
+  const long chunk = (NL + nsplit - 1) / nsplit;
+  const long t0 = (long)split * chunk;
+  const long t1 = min(NL, t0 + chunk);
+
+  const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
+  float s1 = 0.0f, s2 = 0.0f;
+  for (long t = t0 + threadIdx.x; t < t1; t += kBlock) {
+    const long n = t / L;
+    const long l = t - n * L;
+    const long i = (n * C + c) * L + l;
+    const float xh = ((float)x[i] - m) * is;
+    float d = (float)dy[i];
+    if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
+    s1 += d;
+    s2 += d * xh;
+  }
+  s1 = sa::block_reduce_sum(s1, red);
+  __syncthreads();
+  s2 = sa::block_reduce_sum(s2, red);
+  if (threadIdx.x == 0) {
+    atomicAdd(&out[c * 2 + 0], s1);
+    atomicAdd(&out[c * 2 + 1], s2);
+  }
+}
+
+template <typename scalar_t, bool TRAINING>
+__global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
+                                 const scalar_t* __restrict__ x,
+                                 scalar_t* __restrict__ dx,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ beta,
+                                 const float* __restrict__ sums,  // dbeta,dgamma
+                                 int C, long L, long total, long NL, int act) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= total) return;
+  const int c = (int)((i / L) % C);
+  const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
+  const float xh = ((float)x[i] - m) * is;
+  float d = (float)dy[i];
+  if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
+  float v;
+  if (TRAINING) {
+    const float dbeta = sums[c * 2 + 0];
+    const float dgamma = sums[c * 2 + 1];
+    v = (g * is / (float)NL) * ((float)NL * d - dbeta - xh * dgamma);
+  } else {
+    v = d * g * is;
+  }
+  dx[i] = (scalar_t)v;
+}
+
+int pick_nsplit(long NL) {
+  return std::max(1, std::min<int>(64, (int)(NL / 32768) + 1));
+}
+
+}  // namespace
+
+std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                                   const at::Tensor& beta,
+                                   const c10::optional<at::Tensor>& running_mean,
+                                   const c10::optional<at::Tensor>& running_var,
+                                   bool training, double momentum, double eps,
+                                   long act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long NL = (long)N * L;
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = x.options().dtype(at::kFloat);
+
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+
+  at::Tensor mean, invstd;
+  const bool has_running = running_mean.has_value() && running_mean->defined();
+  if (training) {
+    auto sums = at::zeros({C, 2}, opts);
+    mean = at::empty({C}, opts);
+    invstd = at::empty({C}, opts);
+    TORCH_CHECK(!has_running || running_mean->scalar_type() == at::kFloat,
+                "running stats must be fp32");
+    AT_DISPATCH_FLOATING_TYPES_AND2(
+        at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+        "bn_sums", [&] {
+          hipLaunchKernelGGL((bn_sums_kernel<scalar_t>),
+                             dim3(C, pick_nsplit(NL)), dim3(kBlock), 0,
+                             stream.stream(), x.data_ptr<scalar_t>(),
+                             sums.data_ptr<float>(), C, NL, L);
+        });
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 256)),
+                       dim3(256), 0, stream.stream(),
+                       sums.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(),
+                       has_running ? running_mean->data_ptr<float>() : nullptr,
+                       has_running ? running_var->data_ptr<float>() : nullptr,
+                       C, NL, (float)momentum, (float)eps);
+  } else {
+    TORCH_CHECK(has_running, "eval mode requires running stats");
+    mean = running_mean->to(at::kFloat).contiguous();
+    invstd = at::rsqrt(running_var->to(at::kFloat) + eps).contiguous();
+  }
+
+  auto y = at::empty_like(x);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_apply", [&] {
+        hipLaunchKernelGGL((bn_apply_kernel<scalar_t>),
+                           dim3(sa::ceil_div(total, kBlock)), dim3(kBlock), 0,
+                           stream.stream(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), g32.data_ptr<float>(),
+                           b32.data_ptr<float>(), C, L, total, (int)act);
+      });
+  return {y, mean, invstd};
+}
+
+std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                   const at::Tensor& gamma,
+                                   const at::Tensor& beta,
+                                   const at::Tensor& mean,
+                                   const at::Tensor& invstd, bool training,
+                                   long act) {
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long NL = (long)N * L;
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = x.options().dtype(at::kFloat);
+
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+
+  auto sums = at::zeros({C, 2}, opts);  // dbeta, dgamma
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_bwd_sums", [&] {
+        hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t>),
+                           dim3(C, pick_nsplit(NL)), dim3(kBlock), 0,
+                           stream.stream(), dy.data_ptr<scalar_t>(),
+                           x.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), g32.data_ptr<float>(),
+                           b32.data_ptr<float>(), sums.data_ptr<float>(),
+                           C, NL, L, (int)act);
+      });
+
+  auto dx = at::empty_like(x);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_bwd_dx", [&] {
+        if (training) {
+          hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, true>),
+                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
+                             0, stream.stream(), dy.data_ptr<scalar_t>(),
+                             x.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             g32.data_ptr<float>(), b32.data_ptr<float>(),
+                             sums.data_ptr<float>(), C, L, total, NL,
+                             (int)act);
+        } else {
+          hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, false>),
+                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
+                             0, stream.stream(), dy.data_ptr<scalar_t>(),
+                             x.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             g32.data_ptr<float>(), b32.data_ptr<float>(),
+                             sums.data_ptr<float>(), C, L, total, NL,
+                             (int)act);
+        }
+      });
+
+  auto split = sums.unbind(1);
+  auto dbeta = split[0].to(gamma.scalar_type());
+  auto dgamma = split[1].to(gamma.scalar_type());
+  return {dx, dgamma, dbeta};
+}

@@ -1,0 +1,178 @@
+// Fused avg+max pooling (ceil mode) and 1D linear interpolation —
+// K12/K13 of SURVEY.md §2.4. avg+max is the LocalAwareAggregation
+// primitive (reference models/seist.py:80-93: AvgPool1d + MaxPool1d
+// summed — two kernels + an add in eager; one pass here). Linear
+// interpolation matches F.interpolate(mode='linear', align_corners=False)
+// (reference models/seist.py:563-566 upsampling head).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "sa_common.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+template <typename scalar_t>
+__global__ void avgmax_fwd_kernel(const scalar_t* __restrict__ x,
+                                  scalar_t* __restrict__ y,
+                                  int* __restrict__ argmax,
+                                  long L, long Lo, int k, long rows) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= rows * Lo) return;
+  const long row = i / Lo;
+  const long lo = i - row * Lo;
+  const long lo0 = lo * k;
+  const long lo1 = min(lo0 + (long)k, L);
+  const scalar_t* xr = x + row * L;
+  float s = 0.0f;
+  float mx = -INFINITY;
+  int mi = (int)lo0;
+  for (long l = lo0; l < lo1; ++l) {
+    const float v = (float)xr[l];
+    s += v;
+    if (v > mx) {
+      mx = v;
+      mi = (int)l;
+    }
+  }
+  y[i] = (scalar_t)(s / (float)(lo1 - lo0) + mx);
+  argmax[i] = mi;
+}
+
+template <typename scalar_t>
+__global__ void avgmax_bwd_kernel(const scalar_t* __restrict__ dy,
+                                  const int* __restrict__ argmax,
+                                  scalar_t* __restrict__ dx,
+                                  long L, long Lo, int k, long rows) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= rows * L) return;
+  const long row = i / L;
+  const long li = i - row * L;
+  const long lo = li / k;
+  const long lo0 = lo * k;
+  const long lo1 = min(lo0 + (long)k, L);
+  const long oi = row * Lo + lo;
+  const float g = (float)dy[oi];
+  float v = g / (float)(lo1 - lo0);
+  if (argmax[oi] == (int)li) v += g;
+  dx[i] = (scalar_t)v;
+}
+
+template <typename scalar_t>
+__global__ void interp_fwd_kernel(const scalar_t* __restrict__ x,
+                                  scalar_t* __restrict__ y,
+                                  long Li, long Lo, float scale, long rows) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= rows * Lo) return;
+  const long row = i / Lo;
+  const long lo = i - row * Lo;
+  float src = ((float)lo + 0.5f) * scale - 0.5f;
+  src = fmaxf(src, 0.0f);
+  long l0 = (long)src;
+  l0 = min(l0, Li - 1);
+  const long l1 = min(l0 + 1, Li - 1);
+  const float w1 = src - (float)l0;
+  const scalar_t* xr = x + row * Li;
+  y[i] = (scalar_t)((1.0f - w1) * (float)xr[l0] + w1 * (float)xr[l1]);
+}
+
+template <typename scalar_t>
+__global__ void interp_bwd_kernel(const scalar_t* __restrict__ dy,
+                                  float* __restrict__ dx32,
+                                  long Li, long Lo, float scale, long rows) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (i >= rows * Lo) return;
+  const long row = i / Lo;
+  const long lo = i - row * Lo;
+  float src = ((float)lo + 0.5f) * scale - 0.5f;
+  src = fmaxf(src, 0.0f);
+  long l0 = (long)src;
+  l0 = min(l0, Li - 1);
+  const long l1 = min(l0 + 1, Li - 1);
+  const float w1 = src - (float)l0;
+  const float g = (float)dy[i];
+  float* dxr = dx32 + row * Li;
+  atomicAdd(&dxr[l0], (1.0f - w1) * g);
+  if (w1 != 0.0f) atomicAdd(&dxr[l1], w1 * g);
+}
+
+}  // namespace
+
+std::vector<at::Tensor> avgmax_pool_fwd(const at::Tensor& x, long k) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const long rows = (long)x.size(0) * x.size(1);
+  const long L = x.size(2);
+  const long Lo = (L + k - 1) / k;
+  auto y = at::empty({x.size(0), x.size(1), Lo}, x.options());
+  auto idx = at::empty({x.size(0), x.size(1), Lo},
+                       x.options().dtype(at::kInt));
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "avgmax_fwd", [&] {
+        hipLaunchKernelGGL((avgmax_fwd_kernel<scalar_t>),
+                           dim3(sa::ceil_div(rows * Lo, kBlock)),
+                           dim3(kBlock), 0, stream.stream(),
+                           x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                           idx.data_ptr<int>(), L, Lo, (int)k, rows);
+      });
+  return {y, idx};
+}
+
+at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
+                           long k, long in_len) {
+  const long rows = (long)dy.size(0) * dy.size(1);
+  const long Lo = dy.size(2);
+  auto dx = at::empty({dy.size(0), dy.size(1), in_len}, dy.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
+      "avgmax_bwd", [&] {
+        hipLaunchKernelGGL((avgmax_bwd_kernel<scalar_t>),
+                           dim3(sa::ceil_div(rows * in_len, kBlock)),
+                           dim3(kBlock), 0, stream.stream(),
+                           dy.data_ptr<scalar_t>(), argmax.data_ptr<int>(),
+                           dx.data_ptr<scalar_t>(), in_len, Lo, (int)k, rows);
+      });
+  return dx;
+}
+
+at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const long rows = (long)x.size(0) * x.size(1);
+  const long Li = x.size(2);
+  auto y = at::empty({x.size(0), x.size(1), out_len}, x.options());
+  const float scale = (float)Li / (float)out_len;
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "interp_fwd", [&] {
+        hipLaunchKernelGGL((interp_fwd_kernel<scalar_t>),
+                           dim3(sa::ceil_div(rows * out_len, kBlock)),
+                           dim3(kBlock), 0, stream.stream(),
+                           x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                           Li, out_len, scale, rows);
+      });
+  return y;
+}
+
+at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len) {
+  const long rows = (long)dy.size(0) * dy.size(1);
+  const long Lo = dy.size(2);
+  auto dx32 = at::zeros({dy.size(0), dy.size(1), in_len},
+                        dy.options().dtype(at::kFloat));
+  const float scale = (float)in_len / (float)Lo;
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
+      "interp_bwd", [&] {
+        hipLaunchKernelGGL((interp_bwd_kernel<scalar_t>),
+                           dim3(sa::ceil_div(rows * Lo, kBlock)),
+                           dim3(kBlock), 0, stream.stream(),
+                           dy.data_ptr<scalar_t>(), dx32.data_ptr<float>(),
+                           in_len, Lo, scale, rows);
+      });
+  return dx32.to(dy.scalar_type());
+}

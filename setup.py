@@ -1,0 +1,41 @@
+"""Build the in-tree HIP/gfx950 extension:
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Produces seist_amd/_C.*.so next to the package so the GPU-box snapshot
+carries it (no JIT cache dependence).
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+HIP_DIR = os.path.join("seist_amd", "ops", "hip")
+
+ext = CUDAExtension(
+    name="seist_amd._C",
+    sources=[
+        os.path.join(HIP_DIR, "bindings.cpp"),
+        os.path.join(HIP_DIR, "pw_conv.hip"),
+        os.path.join(HIP_DIR, "conv1d.hip"),
+        os.path.join(HIP_DIR, "bn_act.hip"),
+        os.path.join(HIP_DIR, "pool_interp.hip"),
+        os.path.join(HIP_DIR, "adam.hip"),
+    ],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17"],
+    },
+)
+
+setup(
+    name="seist_amd",
+    version="0.1.0",
+    packages=["seist_amd"],
+    ext_modules=[ext],
+    cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
+)

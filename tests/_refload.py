@@ -64,6 +64,15 @@ def load_ref_module(relpath: str, name: str):
     return mod
 
 
+def load_ref_utils():
+    """Import the reference ``utils`` package (Metrics, misc, ...)."""
+    _install_stubs()
+    if REF_ROOT not in sys.path:
+        sys.path.insert(0, REF_ROOT)
+    import utils as ref_utils  # noqa
+    return ref_utils
+
+
 def load_ref_models():
     """Import the reference ``models`` package (registry + all models)."""
     _install_stubs()

@@ -1,0 +1,72 @@
+"""End-to-end engine tests on CPU with the synthetic dataset."""
+
+import glob
+import os
+
+import pytest
+import torch
+
+from seist_amd.cli import get_args, main_worker
+
+
+def _args(tmp_path, extra):
+    argv = [
+        "--mode", "train_test", "--dataset-name", "synthetic",
+        "--dataset-size", "24", "--dataset-samples", "9000",
+        "--batch-size", "4", "--epochs", "1", "--workers", "0",
+        "--device", "cpu", "--use-tensorboard", "false",
+        "--log-base", str(tmp_path), "--warmup-steps", "2",
+        "--down-steps", "3", "--log-step", "100", "--augmentation", "false",
+    ] + extra
+    return get_args(argv)
+
+
+def test_phasenet_train_test_cpu(tmp_path):
+    args = _args(tmp_path, ["--model-name", "phasenet"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    # checkpoint written, results CSV written
+    ckpts = glob.glob(str(tmp_path / "*" / "checkpoints" / "*.pth"))
+    assert len(ckpts) >= 1
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) == 1
+    losses = glob.glob(str(tmp_path / "*" / "loss" / "*.npy"))
+    assert len(losses) == 3
+
+
+def test_seist_regression_task_cpu(tmp_path):
+    args = _args(tmp_path, ["--model-name", "seist_s_emg"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) == 1
+
+
+def test_resume_from_checkpoint(tmp_path):
+    args = _args(tmp_path, ["--model-name", "phasenet", "--mode", "train"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    ckpts = sorted(glob.glob(str(tmp_path / "*" / "checkpoints" / "*.pth")))
+    assert ckpts
+    # resume training from the saved checkpoint
+    args2 = _args(tmp_path, ["--model-name", "phasenet", "--mode", "train",
+                             "--start-epoch", "1", "--epochs", "2"])
+    args2.checkpoint = ckpts[-1]
+    args2.distributed = False
+    main_worker(args2, torch.device("cpu"))
+
+
+def test_early_stopping(tmp_path):
+    args = _args(tmp_path, ["--model-name", "phasenet", "--mode", "train",
+                            "--epochs", "4", "--patience", "0"])
+    args.distributed = False
+    # with patience 0 the run must stop after the first non-improving epoch
+    main_worker(args, torch.device("cpu"))
+
+
+def test_scaled_activation_heads_cpu(tmp_path):
+    # baz head uses cos/sin transforms end to end
+    args = _args(tmp_path, ["--model-name", "baz_network", "--mode",
+                            "train"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))

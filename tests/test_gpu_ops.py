@@ -1,0 +1,188 @@
+"""HIP/gfx950 kernel numerics vs the plain-PyTorch fp32 CPU reference.
+All tests are gpu-marked; each op runs the native kernel (dispatch is hard
+on CUDA tensors) and is compared against the same functional API on CPU."""
+
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+from seist_amd import ops  # noqa: E402
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    from seist_amd.ops import has_ext
+    assert has_ext(), "native extension must be built on the GPU box"
+    return torch.device("cuda:0")
+
+
+def _cmp(a, b, atol, rtol=1e-4, msg=""):
+    a = a.detach().float().cpu()
+    b = b.detach().float().cpu()
+    diff = (a - b).abs().max().item()
+    denom = b.abs().max().item() + 1e-8
+    assert diff <= atol + rtol * denom, f"{msg} max diff {diff} (ref {denom})"
+
+
+@pytest.mark.parametrize("dtype,atol", [(torch.float32, 1e-4),
+                                        (torch.bfloat16, 5e-2)])
+@pytest.mark.parametrize("Ci,Co,L", [(3, 16, 8192), (48, 16, 2048),
+                                     (96, 192, 128), (64, 64, 1024)])
+def test_pw_conv_fwd_bwd(dev, dtype, atol, Ci, Co, L):
+    torch.manual_seed(0)
+    N = 4
+    x32 = torch.randn(N, Ci, L)
+    w32 = torch.randn(Co, Ci, 1) * 0.1
+    b32 = torch.randn(Co) * 0.1
+
+    xg = x32.to(dev, dtype).requires_grad_(True)
+    wg = w32.to(dev, dtype).requires_grad_(True)
+    bg = b32.to(dev, dtype).requires_grad_(True)
+    y = ops.pointwise_conv(xg, wg, bg)
+
+    xc = x32.clone().requires_grad_(True)
+    wc = w32.clone().requires_grad_(True)
+    bc = b32.clone().requires_grad_(True)
+    y_ref = ops.pointwise_conv(xc, wc, bc)
+    _cmp(y, y_ref, atol, msg="pw fwd")
+
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev, dtype))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, atol * 4, msg="pw dx")
+    _cmp(wg.grad.squeeze(), wc.grad.squeeze(), atol * 40, 1e-3, msg="pw dw")
+    _cmp(bg.grad, bc.grad, atol * 40, 1e-3, msg="pw db")
+
+
+@pytest.mark.parametrize("groups,k,stride,dil,Ci,Co", [
+    (16, 11, 2, 1, 16, 16),   # depthwise stem
+    (16, 19, 1, 1, 16, 16),   # widest stem kernel
+    (8, 3, 1, 1, 64, 64),     # grouped conv
+    (1, 7, 1, 1, 16, 8),      # dense head conv
+    (1, 7, 4, 1, 8, 8),       # strided phasenet conv
+    (1, 6, 1, 64, 20, 20),    # dilated causal (dist-PT)
+])
+def test_conv1d_fwd_bwd(dev, groups, k, stride, dil, Ci, Co):
+    torch.manual_seed(1)
+    N, L = 3, 1024
+    padl, padr = (k - 1) * dil // 2, (k - 1) * dil - (k - 1) * dil // 2
+    x32 = torch.randn(N, Ci, L)
+    w32 = torch.randn(Co, Ci // groups, k) * 0.2
+
+    xg = x32.to(dev).requires_grad_(True)
+    wg = w32.to(dev).requires_grad_(True)
+    y = ops.conv1d(xg, wg, None, stride=stride, padding=(padl, padr),
+                   groups=groups, dilation=dil)
+    xc = x32.clone().requires_grad_(True)
+    wc = w32.clone().requires_grad_(True)
+    y_ref = ops.conv1d(xc, wc, None, stride=stride, padding=(padl, padr),
+                       groups=groups, dilation=dil)
+    _cmp(y, y_ref, 1e-4, msg="conv fwd")
+
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, 1e-4, msg="conv dx")
+    _cmp(wg.grad, wc.grad, 1e-3, 1e-3, msg="conv dw")
+
+
+def test_conv1d_with_bias(dev):
+    x32 = torch.randn(2, 8, 256)
+    w32 = torch.randn(4, 8, 7) * 0.2
+    b32 = torch.randn(4)
+    y = ops.conv1d(x32.to(dev), w32.to(dev), b32.to(dev), padding=(3, 3))
+    y_ref = ops.conv1d(x32, w32, b32, padding=(3, 3))
+    _cmp(y, y_ref, 1e-4, msg="conv bias fwd")
+
+
+@pytest.mark.parametrize("training", [True, False])
+@pytest.mark.parametrize("act", ["none", "gelu", "relu"])
+@pytest.mark.parametrize("dtype,atol", [(torch.float32, 1e-4),
+                                        (torch.bfloat16, 3e-2)])
+def test_bn_act_fwd_bwd(dev, training, act, dtype, atol):
+    torch.manual_seed(2)
+    N, C, L = 8, 24, 512
+    x32 = torch.randn(N, C, L) * 2 + 0.5
+    gamma32 = torch.rand(C) + 0.5
+    beta32 = torch.randn(C) * 0.2
+    rm = torch.randn(C) * 0.1
+    rv = torch.rand(C) + 0.5
+
+    rm_g, rv_g = rm.clone().to(dev), rv.clone().to(dev)
+    xg = x32.to(dev, dtype).requires_grad_(True)
+    gg = gamma32.clone().to(dev).requires_grad_(True)
+    bg = beta32.clone().to(dev).requires_grad_(True)
+    y = ops.bn_act(xg, gg, bg, rm_g, rv_g, training, 0.1, 1e-5, act)
+
+    rm_c, rv_c = rm.clone(), rv.clone()
+    xc = x32.clone().requires_grad_(True)
+    gc = gamma32.clone().requires_grad_(True)
+    bc = beta32.clone().requires_grad_(True)
+    y_ref = ops.bn_act(xc, gc, bc, rm_c, rv_c, training, 0.1, 1e-5, act)
+
+    _cmp(y, y_ref, atol, msg="bn fwd")
+    _cmp(rm_g, rm_c, atol, msg="running mean")
+    _cmp(rv_g, rv_c, atol, msg="running var")
+
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev, dtype))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, atol * 4, msg="bn dx")
+    _cmp(gg.grad, gc.grad, atol * 40, 1e-3, msg="bn dgamma")
+    _cmp(bg.grad, bc.grad, atol * 40, 1e-3, msg="bn dbeta")
+
+
+@pytest.mark.parametrize("L,k", [(1024, 2), (1023, 2), (2048, 8), (127, 4)])
+def test_avgmax_pool_fwd_bwd(dev, L, k):
+    x32 = torch.randn(4, 16, L)
+    xg = x32.to(dev).requires_grad_(True)
+    y = ops.avgmax_pool1d(xg, k)
+    xc = x32.clone().requires_grad_(True)
+    y_ref = ops.avgmax_pool1d(xc, k)
+    _cmp(y, y_ref, 1e-5, msg="avgmax fwd")
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, 1e-5, msg="avgmax bwd")
+
+
+@pytest.mark.parametrize("Li,Lo", [(128, 203), (203, 128), (64, 8192),
+                                   (8192, 64)])
+def test_interp_linear_fwd_bwd(dev, Li, Lo):
+    x32 = torch.randn(2, 8, Li)
+    xg = x32.to(dev).requires_grad_(True)
+    y = ops.interp_linear(xg, Lo)
+    xc = x32.clone().requires_grad_(True)
+    y_ref = ops.interp_linear(xc, Lo)
+    _cmp(y, y_ref, 1e-5, msg="interp fwd")
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, 1e-4, msg="interp bwd")
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_adam_gpu_matches_cpu(dev, dtype):
+    from seist_amd.ops import FusedAdam
+    torch.manual_seed(3)
+    w0 = torch.randn(1000)
+    pg = torch.nn.Parameter(w0.to(dev, dtype))
+    pc = torch.nn.Parameter(w0.clone())
+    og = FusedAdam([pg], lr=1e-2, weight_decay=0.01)
+    oc = FusedAdam([pc], lr=1e-2, weight_decay=0.01)
+    for i in range(5):
+        g = torch.randn(1000)
+        pg.grad = g.to(dev, dtype)
+        pc.grad = g.clone()
+        og.step()
+        oc.step()
+    atol = 1e-5 if dtype == torch.float32 else 2e-2
+    _cmp(pg, pc, atol, msg="adam params")
+    if dtype == torch.bfloat16:
+        # fp32 master must track the fp32 trajectory closely
+        _cmp(og.state[pg]["master"], pc, 1e-3, msg="adam master")

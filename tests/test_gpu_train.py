@@ -1,0 +1,100 @@
+"""GPU integration: model forward/backward with the native op path, a full
+bench-style training step, and verification that the in-tree extension is
+the code that actually ran."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda:0")
+
+
+def test_native_extension_loaded(dev):
+    import seist_amd._C as C
+    assert "seist_amd" in C.__file__, C.__file__
+    # the op layer must dispatch to it for CUDA tensors
+    from seist_amd.ops import has_ext, use_native
+    assert has_ext()
+    assert use_native(torch.zeros(1, device=dev))
+
+
+@pytest.mark.parametrize("name", ["seist_m_dpk", "seist_s_dpk", "phasenet"])
+def test_model_gpu_forward_matches_cpu(dev, name):
+    from seist_amd.models import create_model
+    torch.manual_seed(0)
+    m = create_model(name).eval()
+    x = torch.randn(2, 3, 8192)
+    with torch.no_grad():
+        y_cpu = m(x)
+        y_gpu = m.to(dev)(x.to(dev))
+    diff = (y_gpu.float().cpu() - y_cpu).abs().max().item()
+    assert diff < 1e-3, f"{name}: GPU/CPU forward diff {diff}"
+
+
+def test_seist_bf16_train_step(dev):
+    from seist_amd.config import Config
+    from seist_amd.engine.precision import convert_to_bf16
+    from seist_amd.models import create_model
+    from seist_amd.ops import FusedAdam
+    from seist_amd.parallel.ddp import FlatReplica
+
+    torch.manual_seed(0)
+    model = convert_to_bf16(
+        create_model("seist_m_dpk", in_channels=3, in_samples=8192))
+    model = model.to(dev).train()
+    rep = FlatReplica(model)
+    opt = FusedAdam(model.parameters(), lr=1e-4)
+    loss_fn = Config.get_loss("seist_m_dpk").to(dev)
+
+    x = torch.randn(8, 3, 8192, device=dev, dtype=torch.bfloat16)
+    t = torch.rand(8, 3, 8192, device=dev)
+    losses = []
+    for _ in range(5):
+        rep.zero_grad()
+        loss = loss_fn(model(x).float(), t)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses)))
+    # optimizing on a fixed batch must reduce the loss
+    assert losses[-1] < losses[0]
+
+
+def test_eqtransformer_gpu_step(dev):
+    from seist_amd.config import Config
+    from seist_amd.models import create_model
+    from seist_amd.ops import FusedAdam
+
+    torch.manual_seed(0)
+    model = create_model("eqtransformer").to(dev).train()
+    opt = FusedAdam(model.parameters(), lr=1e-4)
+    loss_fn = Config.get_loss("eqtransformer").to(dev)
+    x = torch.randn(4, 3, 8192, device=dev)
+    t = torch.rand(4, 3, 8192, device=dev)
+    loss = loss_fn(model(x), t)
+    opt.zero_grad()
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss).item()
+
+
+def test_train_engine_one_epoch_gpu(dev, tmp_path):
+    from seist_amd.cli import get_args, main_worker
+    args = get_args([
+        "--mode", "train_test", "--model-name", "seist_s_dpk",
+        "--dataset-name", "synthetic", "--dataset-size", "32",
+        "--dataset-samples", "9000", "--batch-size", "8", "--epochs", "1",
+        "--workers", "0", "--device", "cuda:0", "--use-tensorboard",
+        "false", "--log-base", str(tmp_path), "--warmup-steps", "2",
+        "--down-steps", "3", "--log-step", "100", "--augmentation",
+        "false", "--precision", "bf16",
+    ])
+    args.distributed = False
+    main_worker(args, dev)
+    import glob
+    assert glob.glob(str(tmp_path / "*" / "checkpoints" / "*.pth"))

@@ -1,0 +1,91 @@
+"""Peak picking / trigger semantics vs the reference (and obspy-equivalent
+trigger behavior for equal thresholds)."""
+
+import numpy as np
+import pytest
+import torch
+
+from seist_amd.engine.postprocess import (_detect_event, _pick_phase,
+                                          detect_peaks, trigger_onset)
+
+from _refload import load_ref_module, reference_available
+
+
+def test_trigger_onset_runs():
+    x = np.zeros(100)
+    x[10:20] = 0.9
+    x[50:52] = 0.8
+    picks = trigger_onset(x, 0.5, 0.5)
+    assert picks == [[10, 19], [50, 51]]
+
+
+def test_trigger_onset_empty():
+    assert trigger_onset(np.zeros(50), 0.5, 0.5) == []
+
+
+def test_trigger_onset_two_thresholds():
+    x = np.zeros(100)
+    x[10:30] = 0.6   # above thres2 only
+    x[15:20] = 0.9   # crosses thres1
+    picks = trigger_onset(x, 0.8, 0.5)
+    assert picks == [[15, 29]]
+
+
+def test_detect_peaks_basic():
+    x = np.zeros(100, dtype=np.float32)
+    x[30] = 1.0
+    x[60] = 0.8
+    ind = detect_peaks(x, mph=0.5, mpd=10)
+    assert ind.tolist() == [30, 60]
+
+
+def test_detect_peaks_mpd_suppression():
+    x = np.zeros(100, dtype=np.float32)
+    x[30] = 1.0
+    x[35] = 0.9   # within mpd of the higher peak -> suppressed
+    ind = detect_peaks(x, mph=0.5, mpd=10)
+    assert ind.tolist() == [30]
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_detect_peaks_matches_reference():
+    ref = load_ref_module("training/postprocess.py", "ref_post")
+    rng = np.random.default_rng(0)
+    for _ in range(20):
+        x = rng.random(512).astype(np.float32)
+        for mph, mpd, topk in ((0.5, 10, 3), (0.8, 50, 1), (None, 1, None)):
+            a = detect_peaks(x, mph=mph, mpd=mpd, topk=topk)
+            b = ref._detect_peaks(x.copy(), mph=mph, mpd=mpd, topk=topk)
+            assert np.array_equal(a, b), (mph, mpd, topk)
+
+
+def test_pick_phase_batch():
+    out = torch.zeros(3, 256)
+    out[0, 100] = 0.9
+    out[1, 50] = 0.7
+    out[1, 51] = 0.6
+    phases = _pick_phase(out, prob_threshold=0.5, min_peak_dist=10, topk=2,
+                         padding_value=-7)
+    assert phases[0].tolist() == [100, -7]
+    assert phases[1].tolist() == [50, -7]   # 51 suppressed by mpd
+    assert phases[2].tolist() == [-7, -7]
+
+
+def test_detect_event_batch():
+    out = torch.zeros(2, 256)
+    out[0, 20:60] = 0.9
+    dets = _detect_event(out, prob_threshold=0.5, topk=2)
+    assert dets[0].tolist() == [20, 59, 1, 0]
+    assert dets[1].tolist() == [1, 0, 1, 0]
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_pick_phase_matches_reference_loops():
+    """Our batched picker vs the reference's per-trace loop (the reference's
+    obspy dependency is stubbed; only _pick_phase is exercised)."""
+    ref = load_ref_module("training/postprocess.py", "ref_post")
+    rng = np.random.default_rng(3)
+    out = torch.tensor(rng.random((16, 512)), dtype=torch.float32)
+    a = _pick_phase(out, 0.6, 25, 3, -7)
+    b = ref._pick_phase(out, 0.6, 25, 3, -7)
+    assert torch.equal(a, b)

@@ -1,0 +1,144 @@
+"""Golden parity of preprocessing / augmentation / label generation vs the
+reference implementation (and standalone sanity without it)."""
+
+import copy
+
+import numpy as np
+import pytest
+
+from seist_amd.data.preprocess import DataPreprocessor, _pad_array, _pad_phases
+
+from _refload import load_ref_module, reference_available
+
+KW = dict(data_channels=["z", "n", "e"], sampling_rate=50, in_samples=8192,
+          min_snr=-10, p_position_ratio=-1, coda_ratio=1.4, norm_mode="std",
+          add_event_rate=0.3, add_noise_rate=0.3, add_gap_rate=0.3,
+          drop_channel_rate=0.3, scale_amplitude_rate=0.3,
+          pre_emphasis_rate=0.3, pre_emphasis_ratio=0.97, max_event_num=2,
+          generate_noise_rate=0.1, shift_event_rate=0.3, mask_percent=0,
+          noise_percent=0, min_event_gap_sec=1.0)
+
+
+def _pp(shape="gaussian", width=100, **over):
+    kw = dict(KW, **over)
+    return DataPreprocessor(soft_label_shape=shape, soft_label_width=width,
+                            **kw)
+
+
+def _event(L=8192, ppks=(1000,), spks=(1500,)):
+    rng = np.random.default_rng(0)
+    return {"data": rng.standard_normal((3, L)).astype(np.float32),
+            "ppks": list(ppks), "spks": list(spks),
+            "emg": [3.0], "smg": [3.0], "pmp": [0], "clr": [1],
+            "baz": [10.0], "dis": [50.0],
+            "snr": np.array([20.0, 20.0, 20.0])}
+
+
+def test_pad_phases_sentinels():
+    ppks, spks = _pad_phases([5], [9], 7, 100)
+    assert ppks == [5] and spks == [9]
+    ppks, spks = _pad_phases([], [9], 7, 100)
+    assert ppks == [-7] and spks == [9]
+    ppks, spks = _pad_phases([5], [], 7, 100)
+    assert ppks == [5] and spks == [107]
+
+
+def test_pad_array():
+    out = _pad_array([1, 2], 5, -1)
+    assert out.tolist() == [1, 2, -1, -1, -1]
+    with pytest.raises(Exception):
+        _pad_array([1, 2, 3], 2, 0)
+
+
+def test_soft_label_peak_at_pick():
+    pp = _pp()
+    ev = _event()
+    lab = pp._generate_soft_label("ppk", ev, 100, "gaussian")
+    assert lab.shape == (8192,)
+    assert lab.argmax() == 1000
+    assert lab.max() == pytest.approx(1.0)
+
+
+def test_det_label_covers_interval():
+    pp = _pp()
+    ev = _event()
+    lab = pp._generate_soft_label("det", ev, 100, "gaussian")
+    assert lab[1200] == 1.0  # inside [ppk, coda_end)
+    assert lab.max() <= 1.0
+
+
+def test_normalize_modes():
+    pp = _pp()
+    d = np.random.default_rng(1).standard_normal((3, 100)).astype(np.float32)
+    out = pp._normalize(d.copy(), "std")
+    assert np.allclose(out.mean(1), 0, atol=1e-5)
+    assert np.allclose(out.std(1), 1, atol=1e-4)
+    zeros = np.zeros((3, 100), dtype=np.float32)
+    out = pp._normalize(zeros, "std")  # zero std must not divide by zero
+    assert np.all(np.isfinite(out))
+
+
+def test_cut_window_fixed_p_position():
+    pp = _pp(p_position_ratio=0.25)
+    ev = _event(L=12000, ppks=(6000,), spks=(7000,))
+    data, ppks, spks = pp._cut_window(ev["data"], ev["ppks"], ev["spks"], 8192)
+    assert data.shape == (3, 8192)
+    assert ppks[0] == int(8192 * 0.25)
+    assert spks[0] - ppks[0] == 1000
+
+
+def test_onehot_and_value_items():
+    pp = _pp()
+    ev = _event()
+    one = pp._get_io_item("pmp", ev)
+    assert one.tolist() == [1, 0]
+    val = pp._get_io_item("emg", ev)
+    assert val.tolist() == [3.0]
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_labels_match_reference_all_shapes():
+    refpp = load_ref_module("training/preprocess.py", "refpp")
+    for shape in ("gaussian", "triangle", "box", "sigmoid"):
+        ours = _pp(shape)
+        theirs = refpp.DataPreprocessor(soft_label_shape=shape,
+                                        soft_label_width=100, **KW)
+        ev = _event(ppks=(1000, 4100), spks=(1500, 4600))
+        for name in ("ppk", "spk", "non", "det", "ppk+", "z", "dz"):
+            la = ours._generate_soft_label(name, dict(ev), 100, shape)
+            lb = theirs._generate_soft_label(name, dict(ev), 100, shape)
+            assert np.abs(la - lb).max() == 0, (shape, name)
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_augmented_process_matches_reference_rng_stream():
+    refpp = load_ref_module("training/preprocess.py", "refpp")
+    for seed in range(4):
+        ev1 = _event(L=12000, ppks=(3000,), spks=(4500,))
+        ev1["data"] = np.random.default_rng(seed).standard_normal(
+            (3, 12000)).astype(np.float32)
+        ev2 = copy.deepcopy(ev1)
+        np.random.seed(seed)
+        pa = _pp().process(ev1, augmentation=True)
+        np.random.seed(seed)
+        pb = refpp.DataPreprocessor(
+            soft_label_shape="gaussian", soft_label_width=100,
+            **KW).process(ev2, augmentation=True)
+        assert np.allclose(pa["data"], pb["data"])
+        assert pa["ppks"] == pb["ppks"] and pa["spks"] == pb["spks"]
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_metric_targets_match_reference():
+    refpp = load_ref_module("training/preprocess.py", "refpp")
+    ours = _pp()
+    theirs = refpp.DataPreprocessor(soft_label_shape="gaussian",
+                                    soft_label_width=100, **KW)
+    ev = _event()
+    ta = ours.get_targets_for_metrics(copy.deepcopy(ev), 5,
+                                      ["ppk", "spk", "det", "emg", "pmp"])
+    tb = theirs.get_targets_for_metrics(
+        copy.deepcopy(ev), max_event_num=5,
+        task_names=["ppk", "spk", "det", "emg", "pmp"])
+    for k in ta:
+        assert np.array_equal(ta[k], tb[k]), k
