@@ -124,12 +124,15 @@ __global__ void pw_dw_kernel(const scalar_t* __restrict__ dy,
 #pragma unroll 16
     for (int kk = 0; kk < KS; ++kk) {
       acc += dy_s[po][kk] * x_s[pi][kk];
-      if (HAS_BIAS && pi == 0) bacc += dy_s[po][kk];
+      // db only once per (o, k) — not per ci-tile
+      if (HAS_BIAS && pi == 0 && blockIdx.y == 0) bacc += dy_s[po][kk];
     }
   }
 
   if (valid) atomicAdd(&dw[(long)(o0 + po) * Ci + i0 + pi], acc);
-  if (HAS_BIAS && pi == 0 && o0 + po < Co) atomicAdd(&db[o0 + po], bacc);
+  if (HAS_BIAS && pi == 0 && blockIdx.y == 0 && o0 + po < Co) {
+    atomicAdd(&db[o0 + po], bacc);
+  }
 }
 
 }  // namespace
