@@ -104,9 +104,19 @@ __global__ void conv1d_dx_kernel(const scalar_t* __restrict__ dy,
 }
 
 // dw[co][cig][k] = sum_{n,lo} dy[n][co][lo] * x[n][g*Cig+cig][lo*s-padl+k*d]
-// one block per (co, cig*K), grid-stride over (n, lo); fp32 atomics with
-// split over blockIdx.z.
-template <typename scalar_t, bool HAS_BIAS>
+//
+// One block per (co, cig-pair, n-split). The dy row chunk is staged once
+// in LDS and reused for every (cig, k); each thread accumulates all
+// CIG_T*K tap products for its own lo positions, then the block reduces.
+// This reads x only Co times and dy only ceil(Cig/CIG_T) times — the
+// previous per-(co,cig,k)-block design read x Co*K times and was the
+// second-largest kernel cost of the training step.
+constexpr int kCigT = 2;    // cig channels per block
+constexpr int kMaxK = 24;   // max kernel taps supported (zoo max is 19)
+
+// KT is the compile-time tap-count bound so the accumulator array stays in
+// registers (runtime-indexed register arrays spill to scratch on gfx950).
+template <typename scalar_t, bool HAS_BIAS, int KT>
 __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
                                  const scalar_t* __restrict__ x,
                                  float* __restrict__ dw,
@@ -114,39 +124,73 @@ __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
                                  int N, int Ci, int Co, long L, long Lo,
                                  int K, int stride, int padl, int dil,
                                  int G, int nsplit) {
+  __shared__ float dy_s[kBlock];
   __shared__ float red[kBlock / sa::kWave];
 
   const int co = blockIdx.x;
-  const int wi = blockIdx.y;  // cig * K + k
-  const int cig = wi / K;
-  const int k = wi - cig * K;
+  const int cig0 = blockIdx.y * kCigT;
   const int Cig = Ci / G;
   const int g = co / (Co / G);
-  const int ci = g * Cig + cig;
+  const int cig_n = min(kCigT, Cig - cig0);
 
-  const long total = (long)N * Lo;
-  const long chunk = (total + nsplit - 1) / nsplit;
-  const long k0 = (long)blockIdx.z * chunk;
-  const long k1 = min(total, k0 + chunk);
+  const long nchunk = ((long)N + nsplit - 1) / nsplit;
+  const long n0 = (long)blockIdx.z * nchunk;
+  const long n1 = min((long)N, n0 + nchunk);
 
-  float acc = 0.0f;
+  float acc[kCigT][KT];
+#pragma unroll
+  for (int c = 0; c < kCigT; ++c)
+#pragma unroll
+    for (int k = 0; k < KT; ++k) acc[c][k] = 0.0f;
   float bacc = 0.0f;
-  for (long t = k0 + threadIdx.x; t < k1; t += kBlock) {
-    const long n = t / Lo;
-    const long lo = t - n * Lo;
-    const float dyv = (float)dy[((long)n * Co + co) * Lo + lo];
-    const long li = lo * stride - padl + (long)k * dil;
-    if (li >= 0 && li < L) {
-      acc += dyv * (float)x[((long)n * Ci + ci) * L + li];
+
+  for (long n = n0; n < n1; ++n) {
+    const scalar_t* dyr = dy + ((long)n * Co + co) * Lo;
+    const scalar_t* xr0 = x + ((long)n * Ci + g * Cig + cig0) * L;
+    for (long lo0 = 0; lo0 < Lo; lo0 += kBlock) {
+      const long lo = lo0 + threadIdx.x;
+      __syncthreads();
+      dy_s[threadIdx.x] = (lo < Lo) ? (float)dyr[lo] : 0.0f;
+      __syncthreads();
+      if (lo < Lo) {
+        const float dyv = dy_s[threadIdx.x];
+        if (HAS_BIAS && blockIdx.y == 0) bacc += dyv;
+        const long li0 = lo * stride - padl;
+#pragma unroll
+        for (int c = 0; c < kCigT; ++c) {
+          if (c >= cig_n) break;
+          const scalar_t* xr = xr0 + (long)c * L;
+#pragma unroll
+          for (int k = 0; k < KT; ++k) {
+            if (k >= K) break;
+            const long li = li0 + (long)k * dil;
+            if (li >= 0 && li < L) {
+              acc[c][k] += dyv * (float)xr[li];
+            }
+          }
+        }
+      }
     }
-    if (HAS_BIAS && wi == 0) bacc += dyv;
   }
-  acc = sa::block_reduce_sum(acc, red);
-  if (threadIdx.x == 0) atomicAdd(&dw[((long)co * Cig + cig) * K + k], acc);
-  if (HAS_BIAS && wi == 0) {
+
+  // block-reduce each accumulator and emit
+#pragma unroll
+  for (int c = 0; c < kCigT; ++c) {
+#pragma unroll
+    for (int k = 0; k < KT; ++k) {
+      if (c < cig_n && k < K) {
+        __syncthreads();
+        const float v = sa::block_reduce_sum(acc[c][k], red);
+        if (threadIdx.x == 0) {
+          atomicAdd(&dw[((long)co * Cig + cig0 + c) * K + k], v);
+        }
+      }
+    }
+  }
+  if (HAS_BIAS && blockIdx.y == 0) {
     __syncthreads();
-    bacc = sa::block_reduce_sum(bacc, red);
-    if (threadIdx.x == 0) atomicAdd(&db[co], bacc);
+    const float v = sa::block_reduce_sum(bacc, red);
+    if (threadIdx.x == 0) atomicAdd(&db[co], v);
   }
 }
 
@@ -229,27 +273,32 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   at::Tensor db32;
   if (has_bias) db32 = at::zeros({Co}, w.options().dtype(at::kFloat));
   {
-    const int nsplit = std::max(
-        1, std::min<int>(32, (int)(((long)N * Lo) / 65536) + 1));
-    dim3 grid(Co, Cig * K, nsplit);
+    TORCH_CHECK(K <= kMaxK, "conv1d_dw: kernel taps > ", kMaxK);
+    const int nsplit = std::max(1, std::min<int>(
+        (int)N, 4096 / (Co * sa::ceil_div(Cig, kCigT))));
+    dim3 grid(Co, sa::ceil_div(Cig, kCigT), nsplit);
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
         "conv1d_dw", [&] {
-          if (has_bias) {
-            hipLaunchKernelGGL((conv1d_dw_kernel<scalar_t, true>), grid,
-                               dim3(kBlock), 0, stream.stream(),
-                               dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                               dw32.data_ptr<float>(), db32.data_ptr<float>(),
-                               N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
-                               (int)dilation, (int)groups, nsplit);
-          } else {
-            hipLaunchKernelGGL((conv1d_dw_kernel<scalar_t, false>), grid,
-                               dim3(kBlock), 0, stream.stream(),
-                               dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                               dw32.data_ptr<float>(), nullptr,
-                               N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
-                               (int)dilation, (int)groups, nsplit);
-          }
+          auto launch_kt = [&](auto kt, auto hb) {
+            hipLaunchKernelGGL(
+                (conv1d_dw_kernel<scalar_t, decltype(hb)::value,
+                                  decltype(kt)::value>),
+                grid, dim3(kBlock), 0, stream.stream(),
+                dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                dw32.data_ptr<float>(),
+                decltype(hb)::value ? db32.data_ptr<float>() : nullptr,
+                N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
+                (int)dilation, (int)groups, nsplit);
+          };
+          auto launch_hb = [&](auto kt) {
+            if (has_bias) launch_kt(kt, std::true_type{});
+            else launch_kt(kt, std::false_type{});
+          };
+          if (K <= 4) launch_hb(std::integral_constant<int, 4>{});
+          else if (K <= 8) launch_hb(std::integral_constant<int, 8>{});
+          else if (K <= 12) launch_hb(std::integral_constant<int, 12>{});
+          else launch_hb(std::integral_constant<int, kMaxK>{});
         });
   }
   auto dw = dw32.to(w.scalar_type());

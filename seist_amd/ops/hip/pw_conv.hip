@@ -73,68 +73,6 @@ __global__ void pw_gemm_kernel(const scalar_t* __restrict__ x,
   }
 }
 
-// dw[o][i] = sum_{n,l} dy[n][o][l] * x[n][i][l] — a (Co x Ci) GEMM whose
-// K dimension is the flattened (n,l) axis. 16x16 output tile per block,
-// one (o,i) pair per thread, K staged through LDS in 64-wide slabs,
-// split-K over blockIdx.z with fp32 atomics. db folded in.
-template <typename scalar_t, bool HAS_BIAS>
-__global__ void pw_dw_kernel(const scalar_t* __restrict__ dy,
-                             const scalar_t* __restrict__ x,
-                             float* __restrict__ dw,
-                             float* __restrict__ db,
-                             int N, int Ci, int Co, long L, int nsplit) {
-  constexpr int T = 16;   // tile side; T*T == blockDim.x
-  constexpr int KS = 64;  // k-slab width
-  __shared__ float dy_s[T][KS + 1];
-  __shared__ float x_s[T][KS + 1];
-
-  const int o0 = blockIdx.x * T;
-  const int i0 = blockIdx.y * T;
-  const long total = (long)N * L;
-  const long chunk = (total + nsplit - 1) / nsplit;
-  const long k0 = (long)blockIdx.z * chunk;
-  const long k1 = min(total, k0 + chunk);
-
-  const int po = threadIdx.x / T;
-  const int pi = threadIdx.x % T;
-  const bool valid = (o0 + po < Co) && (i0 + pi < Ci);
-
-  float acc = 0.0f;
-  float bacc = 0.0f;
-
-  for (long ks = k0; ks < k1; ks += KS) {
-    const int kn = (int)min((long)KS, k1 - ks);
-    // stage dy rows and x rows: 4 rows per wave pass
-    __syncthreads();
-    for (int idx = threadIdx.x; idx < T * KS; idx += kBlock) {
-      const int r = idx / KS;
-      const int kk = idx % KS;
-      const long k = ks + kk;
-      float dv = 0.0f, xv = 0.0f;
-      if (kk < kn && k < total) {
-        const long n = k / L;
-        const long l = k - n * L;
-        if (o0 + r < Co) dv = (float)dy[((long)n * Co + o0 + r) * L + l];
-        if (i0 + r < Ci) xv = (float)x[((long)n * Ci + i0 + r) * L + l];
-      }
-      dy_s[r][kk] = dv;
-      x_s[r][kk] = xv;
-    }
-    __syncthreads();
-#pragma unroll 16
-    for (int kk = 0; kk < KS; ++kk) {
-      acc += dy_s[po][kk] * x_s[pi][kk];
-      // db only once per (o, k) — not per ci-tile
-      if (HAS_BIAS && pi == 0 && blockIdx.y == 0) bacc += dy_s[po][kk];
-    }
-  }
-
-  if (valid) atomicAdd(&dw[(long)(o0 + po) * Ci + i0 + pi], acc);
-  if (HAS_BIAS && pi == 0 && blockIdx.y == 0 && o0 + po < Co) {
-    atomicAdd(&db[o0 + po], bacc);
-  }
-}
-
 }  // namespace
 
 at::Tensor pw_conv_fwd(const at::Tensor& x, const at::Tensor& w,
@@ -196,33 +134,18 @@ std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,
         });
   }
 
-  auto dw32 = at::zeros({Co, Ci}, x.options().dtype(at::kFloat));
-  at::Tensor db32;
-  if (has_bias) db32 = at::zeros({Co}, x.options().dtype(at::kFloat));
-  {
-    const int nsplit = std::max(
-        1, std::min<int>(128, (int)(((long)N * L) / 65536) + 1));
-    dim3 grid(sa::ceil_div(Co, 16), sa::ceil_div(Ci, 16), nsplit);
-    AT_DISPATCH_FLOATING_TYPES_AND2(
-        at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
-        "pw_conv_dw", [&] {
-          if (has_bias) {
-            hipLaunchKernelGGL((pw_dw_kernel<scalar_t, true>), grid,
-                               dim3(kBlock), 0, stream.stream(),
-                               dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                               dw32.data_ptr<float>(), db32.data_ptr<float>(),
-                               N, Ci, Co, L, nsplit);
-          } else {
-            hipLaunchKernelGGL((pw_dw_kernel<scalar_t, false>), grid,
-                               dim3(kBlock), 0, stream.stream(),
-                               dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
-                               dw32.data_ptr<float>(), nullptr,
-                               N, Ci, Co, L, nsplit);
-          }
-        });
-  }
-  auto dw = dw32.to(w.scalar_type());
+  // dw[o][i] = sum_n dy_n (Co x L) @ x_n^T (L x Ci) — a plain batched GEMM:
+  // run it on the MFMA matrix cores via rocBLAS (guide rule: hand-write the
+  // fused hot ops, use the BLAS library for plain GEMMs), reduce the batch
+  // axis in fp32. The bespoke pw_dw_kernel above measured 79% of the whole
+  // training step (rocprofv3, profiles/); this path is >10x faster.
+  auto dw = at::bmm(dy, x.transpose(1, 2))
+                .sum(/*dim=*/{0}, /*keepdim=*/false, at::kFloat)
+                .to(w.scalar_type());
   at::Tensor db;
-  if (has_bias) db = db32.to(w.scalar_type());
+  if (has_bias) {
+    db = dy.sum(std::vector<int64_t>{0, 2}, false, at::kFloat)
+             .to(w.scalar_type());
+  }
   return {dx, dw, db};
 }
