@@ -25,17 +25,19 @@ __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
   const int c = blockIdx.x;
   const int split = blockIdx.y;
   const int nsplit = gridDim.y;
-  const long chunk = (NL + nsplit - 1) / nsplit;
-  const long t0 = (long)split * chunk;
-  const long t1 = min(NL, t0 + chunk);
+  const long N = NL / L;
+  const long nchunk = (N + nsplit - 1) / nsplit;
+  const long n0 = (long)split * nchunk;
+  const long n1 = min(N, n0 + nchunk);
 
   float s = 0.0f, s2 = 0.0f;
-  for (long t = t0 + threadIdx.x; t < t1; t += kBlock) {
-    const long n = t / L;
-    const long l = t - n * L;
-    const float v = (float)x[(n * C + c) * L + l];
-    s += v;
-    s2 += v * v;
+  for (long n = n0; n < n1; ++n) {
+    const scalar_t* xr = x + (n * C + c) * L;
+    for (long l = threadIdx.x; l < L; l += kBlock) {
+      const float v = (float)xr[l];
+      s += v;
+      s2 += v * v;
+    }
   }
   s = sa::block_reduce_sum(s, red);
   __syncthreads();
@@ -95,21 +97,23 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
   const int c = blockIdx.x;
   const int split = blockIdx.y;
   const int nsplit = gridDim.y;
-  const long chunk = (NL + nsplit - 1) / nsplit;
-  const long t0 = (long)split * chunk;
-  const long t1 = min(NL, t0 + chunk);
+  const long N = NL / L;
+  const long nchunk = (N + nsplit - 1) / nsplit;
+  const long n0 = (long)split * nchunk;
+  const long n1 = min(N, n0 + nchunk);
 
   const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
   float s1 = 0.0f, s2 = 0.0f;
-  for (long t = t0 + threadIdx.x; t < t1; t += kBlock) {
-    const long n = t / L;
-    const long l = t - n * L;
-    const long i = (n * C + c) * L + l;
-    const float xh = ((float)x[i] - m) * is;
-    float d = (float)dy[i];
-    if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
-    s1 += d;
-    s2 += d * xh;
+  for (long n = n0; n < n1; ++n) {
+    const scalar_t* xr = x + (n * C + c) * L;
+    const scalar_t* dyr = dy + (n * C + c) * L;
+    for (long l = threadIdx.x; l < L; l += kBlock) {
+      const float xh = ((float)xr[l] - m) * is;
+      float d = (float)dyr[l];
+      if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
+      s1 += d;
+      s2 += d * xh;
+    }
   }
   s1 = sa::block_reduce_sum(s1, red);
   __syncthreads();

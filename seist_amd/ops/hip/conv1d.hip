@@ -7,10 +7,15 @@
 // x (N, Ci, L), w (Co, Cig=Ci/G, K), y (N, Co, Lo)
 // y[n][co][lo] = sum_{cig,k} w[co][cig][k] * x[n][g*Cig+cig][lo*s - padl + k*d]
 //
-// Weights are staged in LDS (Cig*K <= a few KB for every model in the
-// zoo); each block computes 256 consecutive lo of one (n, co) row, so all
-// global traffic is coalesced and each x row is read once per k-tap from
-// L1/L2.
+// Two forward/dx structures:
+//  * depthwise (Cog == 1): direct per-(n,co) kernel, weights in LDS, x
+//    window through L1 — each x element is consumed by only one output
+//    channel, so there is nothing to share.
+//  * dense/grouped (Cog > 1): "conv-GEMM" — a 32-wide output-channel chunk
+//    per block shares every staged x (or dy) read across 32 accumulators,
+//    exactly like the pointwise GEMM kernel but with a tap loop. This cut
+//    the head-conv forward ~10x vs the one-channel-per-block design
+//    (rocprofv3, profiles/).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -20,102 +25,199 @@
 namespace {
 
 constexpr int kBlock = 256;
-constexpr int kMaxWLds = 4096;  // floats of LDS weight stage per block
+constexpr int kMaxWLds = 8192;  // floats of LDS weight stage per block
+constexpr int kJT = 32;         // output-channel chunk of the conv-GEMM
 
+// ---------------- depthwise direct ----------------
 template <typename scalar_t, bool HAS_BIAS>
-__global__ void conv1d_fwd_kernel(const scalar_t* __restrict__ x,
+__global__ void dwconv_fwd_kernel(const scalar_t* __restrict__ x,
                                   const scalar_t* __restrict__ w,
-                                  const float* __restrict__ bias,
+                                  const scalar_t* __restrict__ bias,
                                   scalar_t* __restrict__ y,
                                   int N, int Ci, int Co, long L, long Lo,
-                                  int K, int stride, int padl, int dil,
-                                  int G) {
-  extern __shared__ float w_lds[];  // [Cig][K] for this co
-
+                                  int K, int stride, int padl, int dil) {
+  extern __shared__ float w_lds[];  // [K]
   const int n = blockIdx.y;
   const int co = blockIdx.z;
-  const int Cig = Ci / G;
-  const int g = co / (Co / G);
   const long lo = (long)blockIdx.x * kBlock + threadIdx.x;
-
-  for (int idx = threadIdx.x; idx < Cig * K; idx += kBlock) {
-    w_lds[idx] = (float)w[(long)co * Cig * K + idx];
+  for (int idx = threadIdx.x; idx < K; idx += kBlock) {
+    w_lds[idx] = (float)w[(long)co * K + idx];
   }
   __syncthreads();
-
   if (lo >= Lo) return;
-
   const long li0 = lo * stride - padl;
-  float acc = HAS_BIAS ? bias[co] : 0.0f;
-  const scalar_t* xb = x + ((long)n * Ci + (long)g * Cig) * L;
-  for (int cig = 0; cig < Cig; ++cig) {
-    const scalar_t* xr = xb + (long)cig * L;
-    const float* wr = w_lds + cig * K;
-    for (int k = 0; k < K; ++k) {
-      const long li = li0 + (long)k * dil;
-      if (li >= 0 && li < L) acc += wr[k] * (float)xr[li];
-    }
+  float acc = HAS_BIAS ? (float)bias[co] : 0.0f;
+  const scalar_t* xr = x + ((long)n * Ci + co) * L;  // depthwise: ci == co
+  for (int k = 0; k < K; ++k) {
+    const long li = li0 + (long)k * dil;
+    if (li >= 0 && li < L) acc += w_lds[k] * (float)xr[li];
   }
   y[((long)n * Co + co) * Lo + lo] = (scalar_t)acc;
 }
 
-// dx[n][ci][li] = sum_{co in group, k} dy[n][co][lo] * w[co][cig][k]
-//   where lo = (li + padl - k*d) / s  (when divisible and in range)
 template <typename scalar_t>
-__global__ void conv1d_dx_kernel(const scalar_t* __restrict__ dy,
+__global__ void dwconv_dx_kernel(const scalar_t* __restrict__ dy,
                                  const scalar_t* __restrict__ w,
                                  scalar_t* __restrict__ dx,
                                  int N, int Ci, int Co, long L, long Lo,
-                                 int K, int stride, int padl, int dil,
-                                 int G) {
-  extern __shared__ float w_lds[];  // [Cog][K] for this ci's group
-
+                                 int K, int stride, int padl, int dil) {
+  extern __shared__ float w_lds[];
   const int n = blockIdx.y;
   const int ci = blockIdx.z;
-  const int Cig = Ci / G;
-  const int Cog = Co / G;
-  const int g = ci / Cig;
-  const int cig = ci - g * Cig;
   const long li = (long)blockIdx.x * kBlock + threadIdx.x;
-
-  // stage w[g*Cog + j][cig][k] for j in [0, Cog)
-  for (int idx = threadIdx.x; idx < Cog * K; idx += kBlock) {
-    const int j = idx / K;
-    const int k = idx - j * K;
-    w_lds[idx] = (float)w[(((long)(g * Cog + j)) * Cig + cig) * K + k];
+  for (int idx = threadIdx.x; idx < K; idx += kBlock) {
+    w_lds[idx] = (float)w[(long)ci * K + idx];
   }
   __syncthreads();
-
   if (li >= L) return;
-
   float acc = 0.0f;
-  for (int j = 0; j < Cog; ++j) {
-    const scalar_t* dyr = dy + ((long)n * Co + g * Cog + j) * Lo;
-    const float* wr = w_lds + j * K;
-    for (int k = 0; k < K; ++k) {
-      const long num = li + padl - (long)k * dil;
-      if (num < 0) continue;
-      if (num % stride) continue;
-      const long lo = num / stride;
-      if (lo < Lo) acc += wr[k] * (float)dyr[lo];
-    }
+  const scalar_t* dyr = dy + ((long)n * Co + ci) * Lo;
+  for (int k = 0; k < K; ++k) {
+    const long num = li + padl - (long)k * dil;
+    if (num < 0) continue;
+    if (num % stride) continue;
+    const long lo = num / stride;
+    if (lo < Lo) acc += w_lds[k] * (float)dyr[lo];
   }
   dx[((long)n * Ci + ci) * L + li] = (scalar_t)acc;
 }
 
-// dw[co][cig][k] = sum_{n,lo} dy[n][co][lo] * x[n][g*Cig+cig][lo*s-padl+k*d]
-//
-// One block per (co, cig-pair, n-split). The dy row chunk is staged once
-// in LDS and reused for every (cig, k); each thread accumulates all
-// CIG_T*K tap products for its own lo positions, then the block reduces.
-// This reads x only Co times and dy only ceil(Cig/CIG_T) times — the
-// previous per-(co,cig,k)-block design read x Co*K times and was the
-// second-largest kernel cost of the training step.
-constexpr int kCigT = 2;    // cig channels per block
-constexpr int kMaxK = 24;   // max kernel taps supported (zoo max is 19)
+// ---------------- dense / grouped conv-GEMM ----------------
+// block: (lo-chunk, n, group-co-chunk). Weights for the 32-co chunk live
+// in LDS ([jo][cig_chunk*K], padded); each staged x value feeds 32 FMAs.
+template <typename scalar_t, bool HAS_BIAS>
+__global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
+                                    const scalar_t* __restrict__ w,
+                                    const scalar_t* __restrict__ bias,
+                                    scalar_t* __restrict__ y,
+                                    int N, int Ci, int Co, long L, long Lo,
+                                    int K, int stride, int padl, int dil,
+                                    int G, int cig_chunk) {
+  extern __shared__ float w_lds[];  // [kJT][cig_chunk*K]
+  const int n = blockIdx.y;
+  const int Cog = Co / G;
+  const int Cig = Ci / G;
+  const int chunks_per_g = (Cog + kJT - 1) / kJT;
+  const int g = blockIdx.z / chunks_per_g;
+  const int j0 = (blockIdx.z - g * chunks_per_g) * kJT;
+  const int co0 = g * Cog + j0;
+  const int jn = min(kJT, Cog - j0);
+  const long lo = (long)blockIdx.x * kBlock + threadIdx.x;
+  const int wrow = cig_chunk * K;
 
-// KT is the compile-time tap-count bound so the accumulator array stays in
-// registers (runtime-indexed register arrays spill to scratch on gfx950).
+  float acc[kJT];
+#pragma unroll
+  for (int j = 0; j < kJT; ++j) acc[j] = 0.0f;
+
+  const long li0 = lo * stride - padl;
+  for (int c0 = 0; c0 < Cig; c0 += cig_chunk) {
+    const int cn = min(cig_chunk, Cig - c0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < jn * cn * K; idx += kBlock) {
+      const int jo = idx / (cn * K);
+      const int r = idx - jo * cn * K;  // cig*K + k within chunk
+      w_lds[jo * wrow + r] = (float)w[((long)(co0 + jo) * Cig + c0) * K + r];
+    }
+    __syncthreads();
+    if (lo < Lo) {
+      const scalar_t* xb = x + ((long)n * Ci + (long)g * Cig + c0) * L;
+      for (int c = 0; c < cn; ++c) {
+        const scalar_t* xr = xb + (long)c * L;
+        for (int k = 0; k < K; ++k) {
+          const long li = li0 + (long)k * dil;
+          if (li < 0 || li >= L) continue;
+          const float xv = (float)xr[li];
+          const float* wp = w_lds + c * K + k;
+#pragma unroll
+          for (int j = 0; j < kJT; ++j) {
+            acc[j] += wp[j * wrow] * xv;
+          }
+        }
+      }
+    }
+  }
+
+  if (lo < Lo) {
+    scalar_t* yp = y + ((long)n * Co + co0) * Lo + lo;
+    for (int j = 0; j < jn; ++j) {
+      float v = acc[j];
+      if (HAS_BIAS) v += (float)bias[co0 + j];
+      yp[(long)j * Lo] = (scalar_t)v;
+    }
+  }
+}
+
+// dx chunked the same way: 32 input channels per block share dy reads.
+template <typename scalar_t>
+__global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
+                                   const scalar_t* __restrict__ w,
+                                   scalar_t* __restrict__ dx,
+                                   int N, int Ci, int Co, long L, long Lo,
+                                   int K, int stride, int padl, int dil,
+                                   int G, int cog_chunk) {
+  extern __shared__ float w_lds[];  // [kJT(ci)][cog_chunk*K]
+  const int n = blockIdx.y;
+  const int Cog = Co / G;
+  const int Cig = Ci / G;
+  const int chunks_per_g = (Cig + kJT - 1) / kJT;
+  const int g = blockIdx.z / chunks_per_g;
+  const int i0 = (blockIdx.z - g * chunks_per_g) * kJT;
+  const int ci0 = g * Cig + i0;
+  const int in_ = min(kJT, Cig - i0);
+  const long li = (long)blockIdx.x * kBlock + threadIdx.x;
+  const int wrow = cog_chunk * K;
+
+  float acc[kJT];
+#pragma unroll
+  for (int j = 0; j < kJT; ++j) acc[j] = 0.0f;
+
+  for (int j0_ = 0; j0_ < Cog; j0_ += cog_chunk) {
+    const int jn = min(cog_chunk, Cog - j0_);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < in_ * jn * K; idx += kBlock) {
+      const int ii = idx / (jn * K);
+      const int r = idx - ii * jn * K;
+      const int jo = r / K;
+      const int k = r - jo * K;
+      w_lds[ii * wrow + r] =
+          (float)w[(((long)(g * Cog + j0_ + jo)) * Cig + i0 + ii) * K + k];
+    }
+    __syncthreads();
+    if (li < L) {
+      for (int jo = 0; jo < jn; ++jo) {
+        const scalar_t* dyr = dy + ((long)n * Co + g * Cog + j0_ + jo) * Lo;
+        for (int k = 0; k < K; ++k) {
+          const long num = li + padl - (long)k * dil;
+          if (num < 0) continue;
+          if (stride > 1 && (num % stride)) continue;
+          const long lo = (stride > 1) ? num / stride : num;
+          if (lo >= Lo) continue;
+          const float dyv = (float)dyr[lo];
+          const float* wp = w_lds + jo * K + k;
+#pragma unroll
+          for (int j = 0; j < kJT; ++j) {
+            acc[j] += wp[j * wrow] * dyv;
+          }
+        }
+      }
+    }
+  }
+
+  if (li < L) {
+    scalar_t* dxp = dx + ((long)n * Ci + ci0) * L + li;
+    for (int j = 0; j < in_; ++j) {
+      dxp[(long)j * L] = (scalar_t)acc[j];
+    }
+  }
+}
+
+// ---------------- weight gradient ----------------
+// One block per (co, cig-pair, n-split): dy chunk staged in LDS once and
+// reused for every (cig, k); KT is the compile-time tap bound so the
+// accumulators stay in registers.
+constexpr int kCigT = 2;
+constexpr int kMaxK = 24;
+
 template <typename scalar_t, bool HAS_BIAS, int KT>
 __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
                                  const scalar_t* __restrict__ x,
@@ -173,7 +275,6 @@ __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
     }
   }
 
-  // block-reduce each accumulator and emit
 #pragma unroll
   for (int c = 0; c < kCigT; ++c) {
 #pragma unroll
@@ -194,6 +295,12 @@ __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
   }
 }
 
+int pick_cig_chunk(int Cig, int K) {
+  int chunk = Cig;
+  while (kJT * chunk * K > kMaxWLds) chunk = (chunk + 1) / 2;
+  return std::max(1, chunk);
+}
+
 }  // namespace
 
 at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
@@ -209,32 +316,56 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
   TORCH_CHECK(Lo > 0, "empty conv output");
   auto y = at::empty({N, Co, Lo}, x.options());
 
-  at::Tensor b32;
   const bool has_bias = bias.has_value() && bias->defined();
-  if (has_bias) b32 = bias->to(at::kFloat).contiguous();
+  at::Tensor bct;
+  if (has_bias) bct = bias->to(x.scalar_type()).contiguous();
 
   const int Cig = Ci / groups;
-  TORCH_CHECK(Cig * K <= kMaxWLds, "weight tile too large for LDS stage");
-  const size_t lds = sizeof(float) * Cig * K;
-  dim3 grid(sa::ceil_div(Lo, kBlock), N, Co);
+  const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
+
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "conv1d_fwd", [&] {
-        if (has_bias) {
-          hipLaunchKernelGGL((conv1d_fwd_kernel<scalar_t, true>), grid,
-                             dim3(kBlock), lds, stream.stream(),
-                             x.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
-                             b32.data_ptr<float>(), y.data_ptr<scalar_t>(),
-                             N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
-                             (int)dilation, (int)groups);
+        const scalar_t* bp = has_bias ? bct.data_ptr<scalar_t>() : nullptr;
+        if (Cog == 1) {  // depthwise
+          TORCH_CHECK(Cig == 1, "depthwise expects Cig==1");
+          dim3 grid(sa::ceil_div(Lo, kBlock), N, Co);
+          const size_t lds = sizeof(float) * K;
+          if (has_bias) {
+            hipLaunchKernelGGL((dwconv_fwd_kernel<scalar_t, true>), grid,
+                               dim3(kBlock), lds, stream.stream(),
+                               x.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
+                               bp, y.data_ptr<scalar_t>(), N, Ci, Co, L, Lo,
+                               K, (int)stride, (int)padl, (int)dilation);
+          } else {
+            hipLaunchKernelGGL((dwconv_fwd_kernel<scalar_t, false>), grid,
+                               dim3(kBlock), lds, stream.stream(),
+                               x.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
+                               bp, y.data_ptr<scalar_t>(), N, Ci, Co, L, Lo,
+                               K, (int)stride, (int)padl, (int)dilation);
+          }
         } else {
-          hipLaunchKernelGGL((conv1d_fwd_kernel<scalar_t, false>), grid,
-                             dim3(kBlock), lds, stream.stream(),
-                             x.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
-                             nullptr, y.data_ptr<scalar_t>(),
-                             N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
-                             (int)dilation, (int)groups);
+          const int cig_chunk = pick_cig_chunk(Cig, K);
+          const int chunks_per_g = sa::ceil_div(Cog, kJT);
+          dim3 grid(sa::ceil_div(Lo, kBlock), N,
+                    (int)groups * chunks_per_g);
+          const size_t lds = sizeof(float) * kJT * cig_chunk * K;
+          if (has_bias) {
+            hipLaunchKernelGGL((convgemm_fwd_kernel<scalar_t, true>), grid,
+                               dim3(kBlock), lds, stream.stream(),
+                               x.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
+                               bp, y.data_ptr<scalar_t>(), N, Ci, Co, L, Lo,
+                               K, (int)stride, (int)padl, (int)dilation,
+                               (int)groups, cig_chunk);
+          } else {
+            hipLaunchKernelGGL((convgemm_fwd_kernel<scalar_t, false>), grid,
+                               dim3(kBlock), lds, stream.stream(),
+                               x.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
+                               bp, y.data_ptr<scalar_t>(), N, Ci, Co, L, Lo,
+                               K, (int)stride, (int)padl, (int)dilation,
+                               (int)groups, cig_chunk);
+          }
         }
       });
   return y;
@@ -253,21 +384,30 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto stream = at::hip::getCurrentHIPStream();
 
   auto dx = at::empty_like(x);
-  {
-    TORCH_CHECK(Cog * K <= kMaxWLds, "weight tile too large for LDS stage");
-    const size_t lds = sizeof(float) * Cog * K;
-    dim3 grid(sa::ceil_div(L, kBlock), N, Ci);
-    AT_DISPATCH_FLOATING_TYPES_AND2(
-        at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
-        "conv1d_dx", [&] {
-          hipLaunchKernelGGL((conv1d_dx_kernel<scalar_t>), grid,
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "conv1d_dx", [&] {
+        if (Cog == 1) {
+          dim3 grid(sa::ceil_div(L, kBlock), N, Ci);
+          const size_t lds = sizeof(float) * K;
+          hipLaunchKernelGGL((dwconv_dx_kernel<scalar_t>), grid,
                              dim3(kBlock), lds, stream.stream(),
                              dy.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
-                             dx.data_ptr<scalar_t>(),
-                             N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
-                             (int)dilation, (int)groups);
-        });
-  }
+                             dx.data_ptr<scalar_t>(), N, Ci, Co, L, Lo, K,
+                             (int)stride, (int)padl, (int)dilation);
+        } else {
+          const int cog_chunk = pick_cig_chunk(Cog, K);
+          const int chunks_per_g = sa::ceil_div(Cig, kJT);
+          dim3 grid(sa::ceil_div(L, kBlock), N, (int)groups * chunks_per_g);
+          const size_t lds = sizeof(float) * kJT * cog_chunk * K;
+          hipLaunchKernelGGL((convgemm_dx_kernel<scalar_t>), grid,
+                             dim3(kBlock), lds, stream.stream(),
+                             dy.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
+                             dx.data_ptr<scalar_t>(), N, Ci, Co, L, Lo, K,
+                             (int)stride, (int)padl, (int)dilation,
+                             (int)groups, cog_chunk);
+        }
+      });
 
   auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
   at::Tensor db32;

@@ -78,24 +78,42 @@ __global__ void interp_fwd_kernel(const scalar_t* __restrict__ x,
   y[i] = (scalar_t)((1.0f - w1) * (float)xr[l0] + w1 * (float)xr[l1]);
 }
 
+// Gather formulation: each dx element sums the dy window that maps onto
+// it (deterministic, no atomics — the scatter version serialised on
+// fp32 atomics when Lo >> Li, e.g. the 8192 -> 64 upsampling-head
+// backward).
 template <typename scalar_t>
 __global__ void interp_bwd_kernel(const scalar_t* __restrict__ dy,
-                                  float* __restrict__ dx32,
+                                  scalar_t* __restrict__ dx,
                                   long Li, long Lo, float scale, long rows) {
   const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= rows * Lo) return;
-  const long row = i / Lo;
-  const long lo = i - row * Lo;
-  float src = ((float)lo + 0.5f) * scale - 0.5f;
-  src = fmaxf(src, 0.0f);
-  long l0 = (long)src;
-  l0 = min(l0, Li - 1);
-  const long l1 = min(l0 + 1, Li - 1);
-  const float w1 = src - (float)l0;
-  const float g = (float)dy[i];
-  float* dxr = dx32 + row * Li;
-  atomicAdd(&dxr[l0], (1.0f - w1) * g);
-  if (w1 != 0.0f) atomicAdd(&dxr[l1], w1 * g);
+  if (i >= rows * Li) return;
+  const long row = i / Li;
+  const long li = i - row * Li;
+  // dy[lo] touches dx indices l0(lo) and l0(lo)+1 where
+  // src(lo) = max((lo+0.5)*scale - 0.5, 0), l0 = min(floor(src), Li-1).
+  // Conservative window of lo whose src lies in (li-1, li+1):
+  const float inv = 1.0f / scale;
+  long lo_lo = (long)floorf(((float)li - 1.0f + 0.5f) * inv - 0.5f) - 1;
+  long lo_hi = (long)ceilf(((float)li + 1.0f + 0.5f) * inv - 0.5f) + 1;
+  if (li == 0) lo_lo = 0;           // src clamp maps early lo to l0 = 0
+  lo_lo = max(lo_lo, (long)0);
+  lo_hi = min(lo_hi, Lo - 1);
+  const scalar_t* dyr = dy + row * Lo;
+  float acc = 0.0f;
+  for (long lo = lo_lo; lo <= lo_hi; ++lo) {
+    float src = ((float)lo + 0.5f) * scale - 0.5f;
+    src = fmaxf(src, 0.0f);
+    long l0 = (long)src;
+    l0 = min(l0, Li - 1);
+    const long l1 = min(l0 + 1, Li - 1);
+    const float w1 = src - (float)l0;
+    float wgt = 0.0f;
+    if (l0 == li) wgt += 1.0f - w1;
+    if (l1 == li && w1 != 0.0f) wgt += w1;
+    acc += wgt * (float)dyr[lo];
+  }
+  dx[i] = (scalar_t)acc;
 }
 
 }  // namespace
@@ -161,18 +179,17 @@ at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len) {
 at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len) {
   const long rows = (long)dy.size(0) * dy.size(1);
   const long Lo = dy.size(2);
-  auto dx32 = at::zeros({dy.size(0), dy.size(1), in_len},
-                        dy.options().dtype(at::kFloat));
+  auto dx = at::empty({dy.size(0), dy.size(1), in_len}, dy.options());
   const float scale = (float)in_len / (float)Lo;
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
       "interp_bwd", [&] {
         hipLaunchKernelGGL((interp_bwd_kernel<scalar_t>),
-                           dim3(sa::ceil_div(rows * Lo, kBlock)),
+                           dim3(sa::ceil_div(rows * in_len, kBlock)),
                            dim3(kBlock), 0, stream.stream(),
-                           dy.data_ptr<scalar_t>(), dx32.data_ptr<float>(),
+                           dy.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                            in_len, Lo, scale, rows);
       });
-  return dx32.to(dy.scalar_type());
+  return dx;
 }
