@@ -152,8 +152,10 @@ __global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
   dx[i] = (scalar_t)v;
 }
 
-int pick_nsplit(long NL) {
-  return std::max(1, std::min<int>(64, (int)(NL / 32768) + 1));
+int pick_nsplit(long N, int C) {
+  // enough blocks to fill 256 CUs several times over, bounded by N
+  const int want = std::max(1, 2048 / std::max(C, 1));
+  return std::max(1, std::min<int>((int)N, want));
 }
 
 }  // namespace
@@ -187,7 +189,7 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
         at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
         "bn_sums", [&] {
           hipLaunchKernelGGL((bn_sums_kernel<scalar_t>),
-                             dim3(C, pick_nsplit(NL)), dim3(kBlock), 0,
+                             dim3(C, pick_nsplit(N, C)), dim3(kBlock), 0,
                              stream.stream(), x.data_ptr<scalar_t>(),
                              sums.data_ptr<float>(), C, NL, L);
         });
@@ -239,7 +241,7 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "bn_bwd_sums", [&] {
         hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t>),
-                           dim3(C, pick_nsplit(NL)), dim3(kBlock), 0,
+                           dim3(C, pick_nsplit(N, C)), dim3(kBlock), 0,
                            stream.stream(), dy.data_ptr<scalar_t>(),
                            x.data_ptr<scalar_t>(), mean.data_ptr<float>(),
                            invstd.data_ptr<float>(), g32.data_ptr<float>(),

@@ -328,8 +328,7 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "conv1d_fwd", [&] {
         const scalar_t* bp = has_bias ? bct.data_ptr<scalar_t>() : nullptr;
-        if (Cog == 1) {  // depthwise
-          TORCH_CHECK(Cig == 1, "depthwise expects Cig==1");
+        if (Cog == 1 && Cig == 1) {  // true depthwise (groups == Ci == Co)
           dim3 grid(sa::ceil_div(Lo, kBlock), N, Co);
           const size_t lds = sizeof(float) * K;
           if (has_bias) {
@@ -387,7 +386,7 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "conv1d_dx", [&] {
-        if (Cog == 1) {
+        if (Cog == 1 && Cig == 1) {
           dim3 grid(sa::ceil_div(L, kBlock), N, Ci);
           const size_t lds = sizeof(float) * K;
           hipLaunchKernelGGL((dwconv_dx_kernel<scalar_t>), grid,
@@ -408,6 +407,30 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
                              (int)groups, cog_chunk);
         }
       });
+
+  // dense stride-1 convs (head/EQT/dist-PT): dw is K shifted plain GEMMs —
+  // run them on the matrix cores via rocBLAS strided-batched bmm over
+  // zero-copy slices instead of the direct-reduction kernel.
+  if (groups == 1 && stride == 1) {
+    auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
+    for (int k = 0; k < K; ++k) {
+      const long off = (long)k * dilation - padl;
+      const long lo0 = std::max<long>(0, -off);
+      const long lo1 = std::min<long>(Lo, L - off);
+      if (lo1 <= lo0) continue;
+      auto dyv = dy.slice(2, lo0, lo1);
+      auto xv = x.slice(2, lo0 + off, lo1 + off);
+      dw32.select(2, k).add_(
+          at::bmm(dyv, xv.transpose(1, 2)).sum({0}, false, at::kFloat));
+    }
+    auto dw = dw32.to(w.scalar_type());
+    at::Tensor db;
+    if (has_bias) {
+      db = dy.sum(std::vector<int64_t>{0, 2}, false, at::kFloat)
+               .to(w.scalar_type());
+    }
+    return {dx, dw, db};
+  }
 
   auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
   at::Tensor db32;
