@@ -19,7 +19,7 @@ constexpr int kBlock = 256;
 
 template <typename scalar_t>
 __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
-                               float* __restrict__ sums,  // (C, 2)
+                               float* __restrict__ part,  // (C, nsplit, 2)
                                int C, long NL, long L) {
   __shared__ float red[kBlock / sa::kWave];
   const int c = blockIdx.x;
@@ -43,12 +43,13 @@ __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
   __syncthreads();
   s2 = sa::block_reduce_sum(s2, red);
   if (threadIdx.x == 0) {
-    atomicAdd(&sums[c * 2 + 0], s);
-    atomicAdd(&sums[c * 2 + 1], s2);
+    part[((long)c * nsplit + split) * 2 + 0] = s;
+    part[((long)c * nsplit + split) * 2 + 1] = s2;
   }
 }
 
-__global__ void bn_finalize_kernel(const float* __restrict__ sums,
+__global__ void bn_finalize_kernel(const float* __restrict__ part,
+                                   int nsplit,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
@@ -57,8 +58,13 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sums,
                                    float eps) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  const float m = sums[c * 2] / NL;
-  const float var = fmaxf(sums[c * 2 + 1] / NL - m * m, 0.0f);
+  float s = 0.0f, s2 = 0.0f;
+  for (int j = 0; j < nsplit; ++j) {
+    s += part[((long)c * nsplit + j) * 2 + 0];
+    s2 += part[((long)c * nsplit + j) * 2 + 1];
+  }
+  const float m = s / NL;
+  const float var = fmaxf(s2 / NL - m * m, 0.0f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
   if (running_mean != nullptr) {
@@ -91,7 +97,7 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
                                    const float* __restrict__ invstd,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
-                                   float* __restrict__ out,  // (C,2): dbeta,dgamma
+                                   float* __restrict__ part,  // (C,nsplit,2)
                                    int C, long NL, long L, int act) {
   __shared__ float red[kBlock / sa::kWave];
   const int c = blockIdx.x;
@@ -119,9 +125,24 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
   __syncthreads();
   s2 = sa::block_reduce_sum(s2, red);
   if (threadIdx.x == 0) {
-    atomicAdd(&out[c * 2 + 0], s1);
-    atomicAdd(&out[c * 2 + 1], s2);
+    part[((long)c * nsplit + split) * 2 + 0] = s1;
+    part[((long)c * nsplit + split) * 2 + 1] = s2;
   }
+}
+
+// reduce (C, nsplit, 2) partials -> (C, 2)
+__global__ void bn_part_reduce_kernel(const float* __restrict__ part,
+                                      float* __restrict__ out,
+                                      int C, int nsplit) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s1 = 0.0f, s2 = 0.0f;
+  for (int j = 0; j < nsplit; ++j) {
+    s1 += part[((long)c * nsplit + j) * 2 + 0];
+    s2 += part[((long)c * nsplit + j) * 2 + 1];
+  }
+  out[c * 2 + 0] = s1;
+  out[c * 2 + 1] = s2;
 }
 
 template <typename scalar_t, bool TRAINING>
@@ -180,7 +201,8 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
   at::Tensor mean, invstd;
   const bool has_running = running_mean.has_value() && running_mean->defined();
   if (training) {
-    auto sums = at::zeros({C, 2}, opts);
+    const int nsplit = pick_nsplit(N, C);
+    auto part = at::empty({C, nsplit, 2}, opts);
     mean = at::empty({C}, opts);
     invstd = at::empty({C}, opts);
     TORCH_CHECK(!has_running || running_mean->scalar_type() == at::kFloat,
@@ -189,13 +211,13 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
         at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
         "bn_sums", [&] {
           hipLaunchKernelGGL((bn_sums_kernel<scalar_t>),
-                             dim3(C, pick_nsplit(N, C)), dim3(kBlock), 0,
+                             dim3(C, nsplit), dim3(kBlock), 0,
                              stream.stream(), x.data_ptr<scalar_t>(),
-                             sums.data_ptr<float>(), C, NL, L);
+                             part.data_ptr<float>(), C, NL, L);
         });
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 256)),
                        dim3(256), 0, stream.stream(),
-                       sums.data_ptr<float>(), mean.data_ptr<float>(),
+                       part.data_ptr<float>(), nsplit, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(),
                        has_running ? running_mean->data_ptr<float>() : nullptr,
                        has_running ? running_var->data_ptr<float>() : nullptr,
@@ -236,18 +258,23 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto g32 = gamma.to(at::kFloat).contiguous();
   auto b32 = beta.to(at::kFloat).contiguous();
 
-  auto sums = at::zeros({C, 2}, opts);  // dbeta, dgamma
+  const int nsplit = pick_nsplit(N, C);
+  auto part = at::empty({C, nsplit, 2}, opts);
+  auto sums = at::empty({C, 2}, opts);  // dbeta, dgamma
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "bn_bwd_sums", [&] {
         hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t>),
-                           dim3(C, pick_nsplit(N, C)), dim3(kBlock), 0,
+                           dim3(C, nsplit), dim3(kBlock), 0,
                            stream.stream(), dy.data_ptr<scalar_t>(),
                            x.data_ptr<scalar_t>(), mean.data_ptr<float>(),
                            invstd.data_ptr<float>(), g32.data_ptr<float>(),
-                           b32.data_ptr<float>(), sums.data_ptr<float>(),
+                           b32.data_ptr<float>(), part.data_ptr<float>(),
                            C, NL, L, (int)act);
       });
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 256)),
+                     dim3(256), 0, stream.stream(), part.data_ptr<float>(),
+                     sums.data_ptr<float>(), C, nsplit);
 
   auto dx = at::empty_like(x);
   AT_DISPATCH_FLOATING_TYPES_AND2(

@@ -408,20 +408,30 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
         }
       });
 
-  // dense stride-1 convs (head/EQT/dist-PT): dw is K shifted plain GEMMs —
-  // run them on the matrix cores via rocBLAS strided-batched bmm over
-  // zero-copy slices instead of the direct-reduction kernel.
-  if (groups == 1 && stride == 1) {
+  // stride-1 convs (dense AND grouped): dw is K shifted plain GEMMs — run
+  // them on the matrix cores via rocBLAS strided-batched bmm over zero-copy
+  // strided views. For groups > 1 the (N, G*Cog, l) slice is viewed as an
+  // (N*G)-batch of (Cog, l) panels (the N and G strides compose exactly),
+  // so one batched GEMM covers every group.
+  if (stride == 1 && Lo + (long)(K - 1) * dilation <= L + padl + padr) {
     auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
+    const long G = groups;
     for (int k = 0; k < K; ++k) {
       const long off = (long)k * dilation - padl;
       const long lo0 = std::max<long>(0, -off);
       const long lo1 = std::min<long>(Lo, L - off);
       if (lo1 <= lo0) continue;
-      auto dyv = dy.slice(2, lo0, lo1);
-      auto xv = x.slice(2, lo0 + off, lo1 + off);
-      dw32.select(2, k).add_(
-          at::bmm(dyv, xv.transpose(1, 2)).sum({0}, false, at::kFloat));
+      const long l = lo1 - lo0;
+      auto dyv = at::as_strided(dy, {N * G, Cog, l},
+                                {(long)Cog * Lo, Lo, 1},
+                                dy.storage_offset() + lo0);
+      auto xv = at::as_strided(x, {N * G, Cig, l},
+                               {(long)Cig * L, L, 1},
+                               x.storage_offset() + lo0 + off);
+      auto prod = at::bmm(dyv, xv.transpose(1, 2))
+                      .view({(long)N, G, (long)Cog, (long)Cig})
+                      .sum({0}, false, at::kFloat);     // (G, Cog, Cig)
+      dw32.select(2, k).add_(prod.view({(long)Co, (long)Cig}));
     }
     auto dw = dw32.to(w.scalar_type());
     at::Tensor db;
