@@ -413,7 +413,9 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   // strided views. For groups > 1 the (N, G*Cog, l) slice is viewed as an
   // (N*G)-batch of (Cog, l) panels (the N and G strides compose exactly),
   // so one batched GEMM covers every group.
-  if (stride == 1 && Lo + (long)(K - 1) * dilation <= L + padl + padr) {
+  // grouped convs keep the direct kernel: at Cog=Cig=8 the per-tap
+  // batched GEMM is launch/overhead-bound and measured slower.
+  if (groups == 1 && stride == 1) {
     auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
     const long G = groups;
     for (int k = 0; k < K; ++k) {
