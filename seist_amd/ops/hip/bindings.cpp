@@ -38,6 +38,9 @@ at::Tensor adam_pack(std::vector<at::Tensor> params,
                      std::vector<at::Tensor> grads,
                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                      std::vector<at::Tensor> masters, bool has_master);
+at::Tensor sum_batch(const at::Tensor& in);
+at::Tensor channel_sum(const at::Tensor& in);
+
 void adam_step_packed(const at::Tensor& meta, const at::Tensor& sample,
                       bool has_master, double lr, double beta1, double beta2,
                       double eps, double wd, double bc1, double bc2,
@@ -55,5 +58,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("interp_linear_fwd", &interp_linear_fwd, "linear interp forward");
   m.def("interp_linear_bwd", &interp_linear_bwd, "linear interp backward");
   m.def("adam_pack", &adam_pack, "pack fused-adam chunk metadata");
+  m.def("sum_batch", &sum_batch, "batch-axis sum to fp32");
+  m.def("channel_sum", &channel_sum, "per-channel sum to fp32");
   m.def("adam_step_packed", &adam_step_packed, "fused adam step");
 }

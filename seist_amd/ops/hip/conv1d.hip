@@ -22,6 +22,9 @@
 
 #include "sa_common.h"
 
+at::Tensor sum_batch(const at::Tensor& in);
+at::Tensor channel_sum(const at::Tensor& in);
+
 namespace {
 
 constexpr int kBlock = 256;
@@ -55,7 +58,7 @@ __global__ void dwconv_fwd_kernel(const scalar_t* __restrict__ x,
   y[((long)n * Co + co) * Lo + lo] = (scalar_t)acc;
 }
 
-template <typename scalar_t>
+template <typename scalar_t, int STRIDE>
 __global__ void dwconv_dx_kernel(const scalar_t* __restrict__ dy,
                                  const scalar_t* __restrict__ w,
                                  scalar_t* __restrict__ dx,
@@ -70,13 +73,14 @@ __global__ void dwconv_dx_kernel(const scalar_t* __restrict__ dy,
   }
   __syncthreads();
   if (li >= L) return;
+  const int s = (STRIDE > 0) ? STRIDE : stride;
   float acc = 0.0f;
   const scalar_t* dyr = dy + ((long)n * Co + ci) * Lo;
   for (int k = 0; k < K; ++k) {
     const long num = li + padl - (long)k * dil;
     if (num < 0) continue;
-    if (num % stride) continue;
-    const long lo = num / stride;
+    if (STRIDE != 1 && (num % s)) continue;
+    const long lo = (STRIDE == 1) ? num : num / s;
     if (lo < Lo) acc += w_lds[k] * (float)dyr[lo];
   }
   dx[((long)n * Ci + ci) * L + li] = (scalar_t)acc;
@@ -93,7 +97,7 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
                                     int N, int Ci, int Co, long L, long Lo,
                                     int K, int stride, int padl, int dil,
                                     int G, int cig_chunk) {
-  extern __shared__ float w_lds[];  // [kJT][cig_chunk*K]
+  extern __shared__ float w_lds[];  // [cig_chunk*K][kJT] — co contiguous
   const int n = blockIdx.y;
   const int Cog = Co / G;
   const int Cig = Ci / G;
@@ -103,20 +107,23 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
   const int co0 = g * Cog + j0;
   const int jn = min(kJT, Cog - j0);
   const long lo = (long)blockIdx.x * kBlock + threadIdx.x;
-  const int wrow = cig_chunk * K;
 
-  float acc[kJT];
+  float2 acc[kJT / 2];
 #pragma unroll
-  for (int j = 0; j < kJT; ++j) acc[j] = 0.0f;
+  for (int j = 0; j < kJT / 2; ++j) acc[j] = make_float2(0.f, 0.f);
 
   const long li0 = lo * stride - padl;
   for (int c0 = 0; c0 < Cig; c0 += cig_chunk) {
     const int cn = min(cig_chunk, Cig - c0);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < jn * cn * K; idx += kBlock) {
-      const int jo = idx / (cn * K);
-      const int r = idx - jo * cn * K;  // cig*K + k within chunk
-      w_lds[jo * wrow + r] = (float)w[((long)(co0 + jo) * Cig + c0) * K + r];
+    for (int idx = threadIdx.x; idx < cn * K * kJT; idx += kBlock) {
+      const int r = idx / kJT;          // c*K + k within chunk
+      const int jo = idx - r * kJT;
+      float v = 0.0f;
+      if (jo < jn) {
+        v = (float)w[((long)(co0 + jo) * Cig + c0) * K + r];
+      }
+      w_lds[idx] = v;
     }
     __syncthreads();
     if (lo < Lo) {
@@ -127,10 +134,11 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
           const long li = li0 + (long)k * dil;
           if (li < 0 || li >= L) continue;
           const float xv = (float)xr[li];
-          const float* wp = w_lds + c * K + k;
+          const float2* wp = (const float2*)(w_lds + (c * K + k) * kJT);
 #pragma unroll
-          for (int j = 0; j < kJT; ++j) {
-            acc[j] += wp[j * wrow] * xv;
+          for (int j = 0; j < kJT / 2; ++j) {
+            acc[j].x = fmaf(wp[j].x, xv, acc[j].x);
+            acc[j].y = fmaf(wp[j].y, xv, acc[j].y);
           }
         }
       }
@@ -140,7 +148,7 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
   if (lo < Lo) {
     scalar_t* yp = y + ((long)n * Co + co0) * Lo + lo;
     for (int j = 0; j < jn; ++j) {
-      float v = acc[j];
+      float v = (j & 1) ? acc[j / 2].y : acc[j / 2].x;
       if (HAS_BIAS) v += (float)bias[co0 + j];
       yp[(long)j * Lo] = (scalar_t)v;
     }
@@ -155,7 +163,7 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
                                    int N, int Ci, int Co, long L, long Lo,
                                    int K, int stride, int padl, int dil,
                                    int G, int cog_chunk) {
-  extern __shared__ float w_lds[];  // [kJT(ci)][cog_chunk*K]
+  extern __shared__ float w_lds[];  // [cog_chunk*K][kJT] — ci contiguous
   const int n = blockIdx.y;
   const int Cog = Co / G;
   const int Cig = Ci / G;
@@ -165,22 +173,24 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
   const int ci0 = g * Cig + i0;
   const int in_ = min(kJT, Cig - i0);
   const long li = (long)blockIdx.x * kBlock + threadIdx.x;
-  const int wrow = cog_chunk * K;
 
-  float acc[kJT];
+  float2 acc[kJT / 2];
 #pragma unroll
-  for (int j = 0; j < kJT; ++j) acc[j] = 0.0f;
+  for (int j = 0; j < kJT / 2; ++j) acc[j] = make_float2(0.f, 0.f);
 
   for (int j0_ = 0; j0_ < Cog; j0_ += cog_chunk) {
     const int jn = min(cog_chunk, Cog - j0_);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < in_ * jn * K; idx += kBlock) {
-      const int ii = idx / (jn * K);
-      const int r = idx - ii * jn * K;
+    for (int idx = threadIdx.x; idx < jn * K * kJT; idx += kBlock) {
+      const int r = idx / kJT;          // jo*K + k within chunk
+      const int ii = idx - r * kJT;
       const int jo = r / K;
       const int k = r - jo * K;
-      w_lds[ii * wrow + r] =
-          (float)w[(((long)(g * Cog + j0_ + jo)) * Cig + i0 + ii) * K + k];
+      float v = 0.0f;
+      if (ii < in_) {
+        v = (float)w[(((long)(g * Cog + j0_ + jo)) * Cig + i0 + ii) * K + k];
+      }
+      w_lds[idx] = v;
     }
     __syncthreads();
     if (li < L) {
@@ -193,10 +203,11 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
           const long lo = (stride > 1) ? num / stride : num;
           if (lo >= Lo) continue;
           const float dyv = (float)dyr[lo];
-          const float* wp = w_lds + jo * K + k;
+          const float2* wp = (const float2*)(w_lds + (jo * K + k) * kJT);
 #pragma unroll
-          for (int j = 0; j < kJT; ++j) {
-            acc[j] += wp[j * wrow] * dyv;
+          for (int j = 0; j < kJT / 2; ++j) {
+            acc[j].x = fmaf(wp[j].x, dyv, acc[j].x);
+            acc[j].y = fmaf(wp[j].y, dyv, acc[j].y);
           }
         }
       }
@@ -206,7 +217,7 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
   if (li < L) {
     scalar_t* dxp = dx + ((long)n * Ci + ci0) * L + li;
     for (int j = 0; j < in_; ++j) {
-      dxp[(long)j * L] = (scalar_t)acc[j];
+      dxp[(long)j * L] = (scalar_t)((j & 1) ? acc[j / 2].y : acc[j / 2].x);
     }
   }
 }
@@ -389,11 +400,17 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
         if (Cog == 1 && Cig == 1) {
           dim3 grid(sa::ceil_div(L, kBlock), N, Ci);
           const size_t lds = sizeof(float) * K;
-          hipLaunchKernelGGL((dwconv_dx_kernel<scalar_t>), grid,
-                             dim3(kBlock), lds, stream.stream(),
-                             dy.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
-                             dx.data_ptr<scalar_t>(), N, Ci, Co, L, Lo, K,
-                             (int)stride, (int)padl, (int)dilation);
+          auto launch_dw = [&](auto st) {
+            hipLaunchKernelGGL((dwconv_dx_kernel<scalar_t,
+                                                 decltype(st)::value>),
+                               grid, dim3(kBlock), lds, stream.stream(),
+                               dy.data_ptr<scalar_t>(), w.data_ptr<scalar_t>(),
+                               dx.data_ptr<scalar_t>(), N, Ci, Co, L, Lo, K,
+                               (int)stride, (int)padl, (int)dilation);
+          };
+          if (stride == 1) launch_dw(std::integral_constant<int, 1>{});
+          else if (stride == 2) launch_dw(std::integral_constant<int, 2>{});
+          else launch_dw(std::integral_constant<int, 0>{});
         } else {
           const int cog_chunk = pick_cig_chunk(Cog, K);
           const int chunks_per_g = sa::ceil_div(Cig, kJT);
@@ -430,16 +447,14 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
       auto xv = at::as_strided(x, {N * G, Cig, l},
                                {(long)Cig * L, L, 1},
                                x.storage_offset() + lo0 + off);
-      auto prod = at::bmm(dyv, xv.transpose(1, 2))
-                      .view({(long)N, G, (long)Cog, (long)Cig})
-                      .sum({0}, false, at::kFloat);     // (G, Cog, Cig)
+      auto prod = sum_batch(at::bmm(dyv, xv.transpose(1, 2))
+                                .view({(long)N, G * Cog, (long)Cig}));
       dw32.select(2, k).add_(prod.view({(long)Co, (long)Cig}));
     }
     auto dw = dw32.to(w.scalar_type());
     at::Tensor db;
     if (has_bias) {
-      db = dy.sum(std::vector<int64_t>{0, 2}, false, at::kFloat)
-               .to(w.scalar_type());
+      db = channel_sum(dy).to(w.scalar_type());
     }
     return {dx, dw, db};
   }
