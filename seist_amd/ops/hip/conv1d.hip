@@ -97,7 +97,7 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
                                     int N, int Ci, int Co, long L, long Lo,
                                     int K, int stride, int padl, int dil,
                                     int G, int cig_chunk) {
-  extern __shared__ float w_lds[];  // [cig_chunk*K][kJT] — co contiguous
+  extern __shared__ float w_lds[];  // [kJT][cig_chunk*K]
   const int n = blockIdx.y;
   const int Cog = Co / G;
   const int Cig = Ci / G;
@@ -107,23 +107,20 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
   const int co0 = g * Cog + j0;
   const int jn = min(kJT, Cog - j0);
   const long lo = (long)blockIdx.x * kBlock + threadIdx.x;
+  const int wrow = cig_chunk * K;
 
-  float2 acc[kJT / 2];
+  float acc[kJT];
 #pragma unroll
-  for (int j = 0; j < kJT / 2; ++j) acc[j] = make_float2(0.f, 0.f);
+  for (int j = 0; j < kJT; ++j) acc[j] = 0.0f;
 
   const long li0 = lo * stride - padl;
   for (int c0 = 0; c0 < Cig; c0 += cig_chunk) {
     const int cn = min(cig_chunk, Cig - c0);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < cn * K * kJT; idx += kBlock) {
-      const int r = idx / kJT;          // c*K + k within chunk
-      const int jo = idx - r * kJT;
-      float v = 0.0f;
-      if (jo < jn) {
-        v = (float)w[((long)(co0 + jo) * Cig + c0) * K + r];
-      }
-      w_lds[idx] = v;
+    for (int idx = threadIdx.x; idx < jn * cn * K; idx += kBlock) {
+      const int jo = idx / (cn * K);
+      const int r = idx - jo * cn * K;  // cig*K + k within chunk
+      w_lds[jo * wrow + r] = (float)w[((long)(co0 + jo) * Cig + c0) * K + r];
     }
     __syncthreads();
     if (lo < Lo) {
@@ -134,11 +131,10 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
           const long li = li0 + (long)k * dil;
           if (li < 0 || li >= L) continue;
           const float xv = (float)xr[li];
-          const float2* wp = (const float2*)(w_lds + (c * K + k) * kJT);
+          const float* wp = w_lds + c * K + k;
 #pragma unroll
-          for (int j = 0; j < kJT / 2; ++j) {
-            acc[j].x = fmaf(wp[j].x, xv, acc[j].x);
-            acc[j].y = fmaf(wp[j].y, xv, acc[j].y);
+          for (int j = 0; j < kJT; ++j) {
+            acc[j] += wp[j * wrow] * xv;
           }
         }
       }
@@ -148,7 +144,7 @@ __global__ void convgemm_fwd_kernel(const scalar_t* __restrict__ x,
   if (lo < Lo) {
     scalar_t* yp = y + ((long)n * Co + co0) * Lo + lo;
     for (int j = 0; j < jn; ++j) {
-      float v = (j & 1) ? acc[j / 2].y : acc[j / 2].x;
+      float v = acc[j];
       if (HAS_BIAS) v += (float)bias[co0 + j];
       yp[(long)j * Lo] = (scalar_t)v;
     }
@@ -163,7 +159,7 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
                                    int N, int Ci, int Co, long L, long Lo,
                                    int K, int stride, int padl, int dil,
                                    int G, int cog_chunk) {
-  extern __shared__ float w_lds[];  // [cog_chunk*K][kJT] — ci contiguous
+  extern __shared__ float w_lds[];  // [kJT(ci)][cog_chunk*K]
   const int n = blockIdx.y;
   const int Cog = Co / G;
   const int Cig = Ci / G;
@@ -173,24 +169,22 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
   const int ci0 = g * Cig + i0;
   const int in_ = min(kJT, Cig - i0);
   const long li = (long)blockIdx.x * kBlock + threadIdx.x;
+  const int wrow = cog_chunk * K;
 
-  float2 acc[kJT / 2];
+  float acc[kJT];
 #pragma unroll
-  for (int j = 0; j < kJT / 2; ++j) acc[j] = make_float2(0.f, 0.f);
+  for (int j = 0; j < kJT; ++j) acc[j] = 0.0f;
 
   for (int j0_ = 0; j0_ < Cog; j0_ += cog_chunk) {
     const int jn = min(cog_chunk, Cog - j0_);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < jn * K * kJT; idx += kBlock) {
-      const int r = idx / kJT;          // jo*K + k within chunk
-      const int ii = idx - r * kJT;
+    for (int idx = threadIdx.x; idx < in_ * jn * K; idx += kBlock) {
+      const int ii = idx / (jn * K);
+      const int r = idx - ii * jn * K;
       const int jo = r / K;
       const int k = r - jo * K;
-      float v = 0.0f;
-      if (ii < in_) {
-        v = (float)w[(((long)(g * Cog + j0_ + jo)) * Cig + i0 + ii) * K + k];
-      }
-      w_lds[idx] = v;
+      w_lds[ii * wrow + r] =
+          (float)w[(((long)(g * Cog + j0_ + jo)) * Cig + i0 + ii) * K + k];
     }
     __syncthreads();
     if (li < L) {
@@ -203,11 +197,10 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
           const long lo = (stride > 1) ? num / stride : num;
           if (lo >= Lo) continue;
           const float dyv = (float)dyr[lo];
-          const float2* wp = (const float2*)(w_lds + (jo * K + k) * kJT);
+          const float* wp = w_lds + jo * K + k;
 #pragma unroll
-          for (int j = 0; j < kJT / 2; ++j) {
-            acc[j].x = fmaf(wp[j].x, dyv, acc[j].x);
-            acc[j].y = fmaf(wp[j].y, dyv, acc[j].y);
+          for (int j = 0; j < kJT; ++j) {
+            acc[j] += wp[j * wrow] * dyv;
           }
         }
       }
@@ -217,7 +210,7 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
   if (li < L) {
     scalar_t* dxp = dx + ((long)n * Ci + ci0) * L + li;
     for (int j = 0; j < in_; ++j) {
-      dxp[(long)j * L] = (scalar_t)((j & 1) ? acc[j / 2].y : acc[j / 2].x);
+      dxp[(long)j * L] = (scalar_t)acc[j];
     }
   }
 }

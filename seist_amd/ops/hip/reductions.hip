@@ -15,18 +15,32 @@ constexpr int kBlock = 256;
 
 template <typename scalar_t>
 __global__ void sum_batch_kernel(const scalar_t* __restrict__ in,
-                                 float* __restrict__ out,
+                                 float* __restrict__ part,  // (nsplit, M)
                                  long B, long M) {
   const long j = (long)blockIdx.x * kBlock + threadIdx.x;
   if (j >= M) return;
+  const int nsplit = gridDim.y;
+  const long bchunk = (B + nsplit - 1) / nsplit;
+  const long b0 = (long)blockIdx.y * bchunk;
+  const long b1 = min(B, b0 + bchunk);
   float s = 0.0f;
   const scalar_t* p = in + j;
-  long b = 0;
-  for (; b + 4 <= B; b += 4) {
+  long b = b0;
+  for (; b + 4 <= b1; b += 4) {
     s += (float)p[b * M] + (float)p[(b + 1) * M]
          + (float)p[(b + 2) * M] + (float)p[(b + 3) * M];
   }
-  for (; b < B; ++b) s += (float)p[b * M];
+  for (; b < b1; ++b) s += (float)p[b * M];
+  part[(long)blockIdx.y * M + j] = s;
+}
+
+__global__ void sum_batch_final_kernel(const float* __restrict__ part,
+                                       float* __restrict__ out,
+                                       int nsplit, long M) {
+  const long j = (long)blockIdx.x * kBlock + threadIdx.x;
+  if (j >= M) return;
+  float s = 0.0f;
+  for (int i = 0; i < nsplit; ++i) s += part[(long)i * M + j];
   out[j] = s;
 }
 
@@ -69,14 +83,23 @@ at::Tensor sum_batch(const at::Tensor& in) {
   out_sizes.erase(out_sizes.begin());
   auto out = at::empty(out_sizes, in.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
+  // enough blocks to fill the chip: M/256 columns x B-split rows
+  const int nsplit = std::max(1, std::min<int>(
+      (int)B, (int)(1024 / std::max<long>(M / kBlock, 1))));
+  auto part = at::empty({nsplit, M}, in.options().dtype(at::kFloat));
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, in.scalar_type(),
       "sum_batch", [&] {
         hipLaunchKernelGGL((sum_batch_kernel<scalar_t>),
-                           dim3(sa::ceil_div(M, kBlock)), dim3(kBlock), 0,
+                           dim3(sa::ceil_div(M, kBlock), nsplit),
+                           dim3(kBlock), 0,
                            stream.stream(), in.data_ptr<scalar_t>(),
-                           out.data_ptr<float>(), B, M);
+                           part.data_ptr<float>(), B, M);
       });
+  hipLaunchKernelGGL(sum_batch_final_kernel,
+                     dim3(sa::ceil_div(M, kBlock)), dim3(kBlock), 0,
+                     stream.stream(), part.data_ptr<float>(),
+                     out.data_ptr<float>(), nsplit, M);
   return out;
 }
 
