@@ -83,9 +83,10 @@ at::Tensor sum_batch(const at::Tensor& in) {
   out_sizes.erase(out_sizes.begin());
   auto out = at::empty(out_sizes, in.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
-  // enough blocks to fill the chip: M/256 columns x B-split rows
-  const int nsplit = std::max(1, std::min<int>(
-      (int)B, (int)(1024 / std::max<long>(M / kBlock, 1))));
+  // target ~512 blocks total, nsplit bounded so the final pass stays tiny
+  const long col_blocks = std::max<long>(M / kBlock, 1);
+  const int nsplit = std::max(1, (int)std::min<long>(
+      std::min<long>(B, 64), 512 / col_blocks));
   auto part = at::empty({nsplit, M}, in.options().dtype(at::kFloat));
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, in.scalar_type(),
