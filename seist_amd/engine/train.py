@@ -37,6 +37,7 @@ from .metrics import Metrics
 from .postprocess import process_outputs
 from .precision import cast_inputs, convert_to_bf16
 from .scalars import ScalarWriter
+from .tracing import StepTimer
 from .validate import validate
 
 
@@ -79,63 +80,85 @@ def train(args, tasks, model, optimizer, scheduler, loss_fn, train_loader,
         "outputs_transform_for_loss", "outputs_transform_for_results")
 
     sync_per_step = pdist.is_dist() and args.sync_metrics_per_step
+    timer = StepTimer(getattr(args, "trace_step_time", False),
+                      torch.device(device))
 
-    for step, (x, loss_targets, metrics_targets, _) in enumerate(train_loader):
-        x = cast_inputs(_to_device(x, device), compute_dtype)
-        loss_targets = _to_device(loss_targets, device)
+    loader_it = iter(train_loader)
+    step = -1
+    while True:
+        with timer.phase("data"):
+            batch = next(loader_it, None)
+        if batch is None:
+            break
+        step += 1
+        x, loss_targets, metrics_targets, _ = batch
+        with timer.phase("h2d"):
+            x = cast_inputs(_to_device(x, device), compute_dtype)
+            loss_targets = _to_device(loss_targets, device)
 
-        outputs = model(x)
+        with timer.phase("forward"):
+            outputs = model(x)
 
-        outputs_for_loss = (outs_trans_for_loss(outputs)
-                            if outs_trans_for_loss is not None else outputs)
-        loss_targets = (tgts_trans_for_loss(loss_targets)
-                        if tgts_trans_for_loss is not None else loss_targets)
-        if isinstance(outputs_for_loss, (list, tuple)):
-            outputs_for_loss = [o.float() for o in outputs_for_loss]
-        else:
-            outputs_for_loss = outputs_for_loss.float()
-        loss = loss_fn(outputs_for_loss, loss_targets)
+        with timer.phase("loss"):
+            outputs_for_loss = (outs_trans_for_loss(outputs)
+                                if outs_trans_for_loss is not None
+                                else outputs)
+            loss_targets = (tgts_trans_for_loss(loss_targets)
+                            if tgts_trans_for_loss is not None
+                            else loss_targets)
+            if isinstance(outputs_for_loss, (list, tuple)):
+                outputs_for_loss = [o.float() for o in outputs_for_loss]
+            else:
+                outputs_for_loss = outputs_for_loss.float()
+            loss = loss_fn(outputs_for_loss, loss_targets)
 
-        optimizer.zero_grad(set_to_none=True)
-        loss.backward()
-        optimizer.step()
-
-        if scheduler is not None:
-            scheduler.step()
-            lr = scheduler.get_last_lr()[0]
-        else:
-            lr = optimizer.param_groups[0]["lr"]
+        with timer.phase("backward"):
+            optimizer.zero_grad(set_to_none=True)
+            loss.backward()
+        with timer.phase("optimizer"):
+            optimizer.step()
+            if scheduler is not None:
+                scheduler.step()
+                lr = scheduler.get_last_lr()[0]
+            else:
+                lr = optimizer.param_groups[0]["lr"]
 
         step_batch_size = (x[0] if isinstance(x, (list, tuple))
                            else x).size(0)
-        if pdist.is_dist():
-            loss = pdist.reduce_tensor(loss, "AVG")
-            sbs = torch.tensor(step_batch_size, device=device,
-                               dtype=torch.int32)
-            step_batch_size = pdist.reduce_tensor(sbs).item()
+        with timer.phase("comm"):
+            if pdist.is_dist():
+                loss = pdist.reduce_tensor(loss, "AVG")
+                sbs = torch.tensor(step_batch_size, device=device,
+                                   dtype=torch.int32)
+                step_batch_size = pdist.reduce_tensor(sbs).item()
 
         average_meters["loss"].update(loss.item(), step_batch_size)
         train_loss_per_step.append(loss.item())
 
-        outputs_for_metrics = (outs_trans_for_res(outputs)
-                               if outs_trans_for_res is not None else outputs)
-        if isinstance(outputs_for_metrics, (list, tuple)):
-            outputs_for_metrics = [o.float() for o in outputs_for_metrics]
-        else:
-            outputs_for_metrics = outputs_for_metrics.float()
-        results = process_outputs(args, outputs_for_metrics, label_names,
-                                  sampling_rate)
+        with timer.phase("postprocess"):
+            outputs_for_metrics = (outs_trans_for_res(outputs)
+                                   if outs_trans_for_res is not None
+                                   else outputs)
+            if isinstance(outputs_for_metrics, (list, tuple)):
+                outputs_for_metrics = [o.float()
+                                       for o in outputs_for_metrics]
+            else:
+                outputs_for_metrics = outputs_for_metrics.float()
+            results = process_outputs(args, outputs_for_metrics, label_names,
+                                      sampling_rate)
 
-        tasks_metrics = {}
-        for task in tasks:
-            metrics = new_metrics(task)
-            tasks_metrics[task] = metrics
-            metrics.compute(targets=metrics_targets[task],
-                            preds=results[task], reduce=sync_per_step)
-            for metric in metrics.metric_names():
-                average_meters[f"{task}_{metric}"].update(
-                    metrics.get_metric(metric), step_batch_size)
-            metrics_merged[task].add(metrics)
+        with timer.phase("metrics"):
+            tasks_metrics = {}
+            for task in tasks:
+                metrics = new_metrics(task)
+                tasks_metrics[task] = metrics
+                metrics.compute(targets=metrics_targets[task],
+                                preds=results[task], reduce=sync_per_step)
+                for metric in metrics.metric_names():
+                    average_meters[f"{task}_{metric}"].update(
+                        metrics.get_metric(metric), step_batch_size)
+                metrics_merged[task].add(metrics)
+        timer.step()
 
         if scalar_writer is not None and pdist.is_main_process():
             gstep = epoch * len(train_loader) + step
@@ -154,6 +177,11 @@ def train(args, tasks, model, optimizer, scheduler, loss_fn, train_loader,
     if pdist.is_dist() and not sync_per_step:
         for task in tasks:
             metrics_merged[task].synchronize_between_processes()
+
+    if timer.enabled and pdist.is_main_process():
+        msg = timer.format()
+        if msg:
+            logger.info(f"* {msg}")
 
     return train_loss_per_step, metrics_merged
 

@@ -70,3 +70,32 @@ def test_scaled_activation_heads_cpu(tmp_path):
                             "train"])
     args.distributed = False
     main_worker(args, torch.device("cpu"))
+
+
+def test_step_time_tracing(tmp_path, capsys):
+    args = _args(tmp_path, ["--model-name", "phasenet", "--mode", "train",
+                            "--trace-step-time", "true"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    # breakdown line must be in the train log
+    logs = glob.glob(str(tmp_path / "*" / "train.log"))
+    text = open(logs[0]).read() if logs else ""
+    assert "step breakdown" in text
+
+
+def test_torchrun_bench_cpu(tmp_path):
+    """The driver's exact multi-rank launch contract must work (gloo/CPU)."""
+    import subprocess
+    import sys
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29541", "bench.py", "--gpus", "2",
+         "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert res.returncode == 0, res.stderr[-2000:]
+    import json as _json
+    line = [l for l in res.stdout.splitlines() if l.startswith("{")][-1]
+    out = _json.loads(line)
+    assert out["n_gpus"] == 2 and out["value"] > 0
