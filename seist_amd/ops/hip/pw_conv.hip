@@ -18,6 +18,9 @@
 
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor channel_sum(const at::Tensor& in);
+bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
+                  const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                  bool trans);
 
 namespace {
 
@@ -92,6 +95,9 @@ at::Tensor pw_conv_fwd(const at::Tensor& x, const at::Tensor& w,
   const bool has_bias = bias.has_value() && bias->defined();
   if (has_bias) b32 = bias->to(at::kFloat).contiguous();
 
+  // bf16 path runs on the matrix cores
+  if (pw_mfma_gemm(x, w, bias, y, /*trans=*/false)) return y;
+
   dim3 grid(sa::ceil_div(L, kBlock), N, sa::ceil_div(Co, kCoChunk));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
@@ -124,7 +130,7 @@ std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,
 
   // dx = W^T @ dy (same kernel, transposed weight view)
   auto dx = at::empty_like(x);
-  {
+  if (!pw_mfma_gemm(dy, w, c10::nullopt, dx, /*trans=*/true)) {
     dim3 grid(sa::ceil_div(L, kBlock), N, sa::ceil_div(Ci, kCoChunk));
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),

@@ -1,0 +1,188 @@
+// MFMA (matrix-core) pointwise-conv GEMM for bf16 — the K1 hot op of
+// SURVEY.md §2.4 on the CDNA4 matrix pipe.
+//
+//   y[n, m, l] = sum_k Wv[m][k] * x[n, k, l]   (+ bias[m])
+//   Wv[m][k] = TRANS ? w[k*Co + m] : w[m*Ci + k]
+//
+// Tiling: one 256-thread block (4 waves) owns a (32 co x 128 l) output
+// tile of one sample; waves split 2x2 over (co, l); each wave computes
+// 16x64 via four v_mfma_f32_16x16x32_bf16 accumulators, K-stepping over
+// ci in 32-wide chunks staged through LDS.
+//
+// Fragment maps (gfx950, 16x16x32): A[i = lane&15][k = (lane>>4)*8 + j],
+// B[k = (lane>>4)*8 + j][n = lane&15], D col = lane&15,
+// row = (lane>>4)*4 + reg (cdna_hip_programming.md §3).
+//
+// The X chunk stays row-major in LDS (coalesced b128 staging); the
+// B-fragment gather pays 8 narrow LDS reads per fragment, which the four
+// MFMAs per K-step hide. The W chunk is read as one contiguous bf16x8 per
+// lane from a padded-row LDS image (conflict-free: 40-element row pitch).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "sa_common.h"
+
+typedef __bf16 sa_bf16;
+typedef sa_bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kCoT = 32;    // co per block
+constexpr int kLT = 128;    // l per block
+constexpr int kKT = 32;     // ci per K-step
+constexpr int kWPitch = 40;   // LDS row pitch of the W chunk (bf16)
+constexpr int kXPitch = 136;  // LDS row pitch of the X chunk (bf16)
+
+template <bool TRANS, bool HAS_BIAS>
+__global__ __launch_bounds__(kBlock)
+void pw_mfma_kernel(const sa_bf16* __restrict__ x,
+                    const sa_bf16* __restrict__ w,
+                    const sa_bf16* __restrict__ bias,
+                    sa_bf16* __restrict__ y,
+                    int N, int Ci, int Co, long L) {
+  __shared__ sa_bf16 w_s[kCoT * kWPitch];
+  __shared__ sa_bf16 x_s[kKT * kXPitch];
+
+  const int n = blockIdx.y;
+  const int co0 = blockIdx.z * kCoT;
+  const long l0 = (long)blockIdx.x * kLT;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;   // 0..1: co sub-block of 16
+  const int wc = wid & 1;    // 0..1: l sub-chunk of 64
+
+  const int frag_m = lane & 15;          // A row / D col index
+  const int kbase = (lane >> 4) * 8;     // fragment k base
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc3 = {0.f, 0.f, 0.f, 0.f};
+
+  const sa_bf16* xb = x + (long)n * Ci * L;
+
+  for (int k0 = 0; k0 < Ci; k0 += kKT) {
+    // ---- stage W chunk [kCoT][kKT] (guarded, zero-padded) ----
+    __syncthreads();
+    for (int idx = tid; idx < kCoT * kKT; idx += kBlock) {
+      const int m = idx / kKT;
+      const int k = idx - m * kKT;
+      const int mg = co0 + m;
+      const int kg = k0 + k;
+      float v = 0.0f;
+      if (mg < Co && kg < Ci) {
+        v = TRANS ? (float)w[(long)kg * Co + mg]
+                  : (float)w[(long)mg * Ci + kg];
+      }
+      w_s[m * kWPitch + k] = (sa_bf16)v;
+    }
+    // ---- stage X chunk [kKT][kLT] row-major (b128 loads/stores) ----
+    for (int idx = tid; idx < kKT * (kLT / 8); idx += kBlock) {
+      const int k = idx / (kLT / 8);
+      const int c8 = idx - k * (kLT / 8);
+      const long lg = l0 + c8 * 8;
+      const int kg = k0 + k;
+      bf16x8 v = {};
+      if (kg < Ci) {
+        if (lg + 8 <= L) {
+          v = *(const bf16x8*)(xb + (long)kg * L + lg);
+        } else {
+          for (int j = 0; j < 8; ++j) {
+            v[j] = (lg + j < L) ? xb[(long)kg * L + lg + j] : (sa_bf16)0.f;
+          }
+        }
+      }
+      *(bf16x8*)(x_s + k * kXPitch + c8 * 8) = v;
+    }
+    __syncthreads();
+
+    // ---- fragments + MFMA ----
+    const bf16x8 a =
+        *(const bf16x8*)(w_s + (wr * 16 + frag_m) * kWPitch + kbase);
+    const int lb = wc * 64 + frag_m;  // this lane's l column within tile
+#pragma unroll
+    for (int nrep = 0; nrep < 4; ++nrep) {
+      bf16x8 b;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        b[j] = x_s[(kbase + j) * kXPitch + nrep * 16 + lb];
+      }
+      switch (nrep) {
+        case 0: acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0); break;
+        case 1: acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc1, 0, 0, 0); break;
+        case 2: acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc2, 0, 0, 0); break;
+        case 3: acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc3, 0, 0, 0); break;
+      }
+    }
+  }
+
+  // ---- epilogue: D col = lane&15 (l), row = (lane>>4)*4 + r (co) ----
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int nrep = 0; nrep < 4; ++nrep) {
+    const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
+                      : (nrep == 2) ? acc2 : acc3;
+    const long lg = l0 + wc * 64 + nrep * 16 + d_col;
+    if (lg >= L) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mg = co0 + wr * 16 + d_row0 + r;
+      if (mg < Co) {
+        float v = acc[r];
+        if (HAS_BIAS) v += (float)bias[mg];
+        y[((long)n * Co + mg) * L + lg] = (sa_bf16)v;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// host entry; returns false if the shape/dtype is not handled
+bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
+                  const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                  bool trans) {
+  if (x.scalar_type() != at::ScalarType::BFloat16) return false;
+  const int N = x.size(0), Ci = x.size(1);
+  const long L = x.size(2);
+  const int Co = trans ? w.size(1) : w.size(0);
+  // MFMA pays off once the K dimension covers at least one 32-chunk
+  // reasonably; tiny-K layers stay on the VALU kernel.
+  if (Ci < 16) return false;
+
+  const bool has_bias = bias.has_value() && bias->defined();
+  at::Tensor bct;
+  if (has_bias) bct = bias->to(x.scalar_type()).contiguous();
+
+  dim3 grid(sa::ceil_div(L, kLT), N, sa::ceil_div(Co, kCoT));
+  auto stream = at::hip::getCurrentHIPStream();
+  const sa_bf16* xp = (const sa_bf16*)x.data_ptr();
+  const sa_bf16* wp = (const sa_bf16*)w.data_ptr();
+  const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
+  sa_bf16* yp = (sa_bf16*)y.data_ptr();
+
+  if (trans) {
+    if (has_bias) {
+      hipLaunchKernelGGL((pw_mfma_kernel<true, true>), grid, dim3(kBlock), 0,
+                         stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
+    } else {
+      hipLaunchKernelGGL((pw_mfma_kernel<true, false>), grid, dim3(kBlock),
+                         0, stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
+    }
+  } else {
+    if (has_bias) {
+      hipLaunchKernelGGL((pw_mfma_kernel<false, true>), grid, dim3(kBlock),
+                         0, stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
+    } else {
+      hipLaunchKernelGGL((pw_mfma_kernel<false, false>), grid, dim3(kBlock),
+                         0, stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
+    }
+  }
+  return true;
+}

@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "pool_interp.hip"),
         os.path.join(HIP_DIR, "adam.hip"),
         os.path.join(HIP_DIR, "reductions.hip"),
+        os.path.join(HIP_DIR, "pw_mfma.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
