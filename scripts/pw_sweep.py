@@ -1,4 +1,5 @@
-"""A/B sweep: VALU pw_gemm vs MFMA pw kernel on the seist_m_dpk pointwise
+"""Shape sweep: VALU pw_gemm vs MFMA pw kernel on the seist_m_dpk pointwise
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 shapes (within-process interleaved timing)."""
 
 import torch
