@@ -113,7 +113,9 @@ def main():
         torch.cuda.synchronize()
 
     graphed = False
-    if args.graph and use_cuda and not distributed:
+    has_rnn = any(isinstance(m, (torch.nn.LSTM, torch.nn.GRU))
+                  for m in model.modules())
+    if args.graph and use_cuda and not distributed and not has_rnn:
         try:
             g = torch.cuda.CUDAGraph()
             stream = torch.cuda.Stream()

@@ -1,6 +1,8 @@
 """Shape sweep: VALU pw_gemm vs MFMA pw kernel on the seist_m_dpk pointwise
-import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 shapes (within-process interleaved timing)."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import seist_amd._C as C
