@@ -206,10 +206,11 @@ class GroupConvBlock(nn.Module):
         y = _conv(self.conv, x, auto_pad=True)
         y = _norm(self.norm0, y, act="gelu" if _is_gelu(self.act) else "none")
         y = ops.pointwise_conv(y, self.proj.weight, self.proj.bias)
-        x = x + self.droppath0(y)
+        x = ops.droppath_add(x, y, self.droppath0.drop_prob, self.training)
         y = _norm(self.norm1, x)
         y = self.mlp(y)
-        return x + self.droppath1(y)
+        return ops.droppath_add(x, y, self.droppath1.drop_prob,
+                                self.training)
 
 
 class MultiScaleMixedConv(nn.Module):
@@ -355,16 +356,21 @@ class MultiPathTransformerLayer(nn.Module):
         if self.has_attn:
             x1 = _norm(self.norm0, ops.pointwise_conv(
                 x, self.attn_proj.weight, self.attn_proj.bias))
-            x1 = x1 + self.attn_droppath(self.attention(x1))
+            x1 = ops.droppath_add(x1, self.attention(x1),
+                                  self.attn_droppath.drop_prob,
+                                  self.training)
             outs.append(x1)
         if self.has_conv:
             x2 = _norm(self.norm1, ops.pointwise_conv(
                 x, self.conv_proj.weight, self.conv_proj.bias))
-            x2 = x2 + self.gconv_droppath(self.gconv(x2))
+            x2 = ops.droppath_add(x2, self.gconv(x2),
+                                  self.gconv_droppath.drop_prob,
+                                  self.training)
             outs.append(x2)
         x = torch.cat(outs, dim=1)
         x = _norm(self.norm2, x)
-        return x + self.mlp_droppath(self.mlp(x))
+        return ops.droppath_add(x, self.mlp(x),
+                                self.mlp_droppath.drop_prob, self.training)
 
 
 class HeadDetectionPicking(nn.Module):

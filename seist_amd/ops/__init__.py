@@ -53,6 +53,7 @@ def use_native(x: torch.Tensor) -> bool:
 from .functional import (  # noqa: E402,F401
     auto_pad,
     avgmax_pool1d,
+    droppath_add,
     bn_act,
     conv1d,
     gelu,

@@ -361,3 +361,46 @@ def pooled_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         attn = F.dropout(attn, p=attn_dropout, training=True)
     out = torch.matmul(attn, v.transpose(-1, -2)).transpose(-1, -2)
     return out
+
+
+# ---------------------------------------------------------------------------
+# fused residual + DropPath (stochastic depth): z = x + mask[n]/keep * y
+# ---------------------------------------------------------------------------
+
+
+class _DropPathAdd(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, y, mask, scale):
+        ctx.scale = scale
+        ctx.save_for_backward(mask if mask is not None else torch.empty(0))
+        if use_native(x) and hasattr(ext(), "row_scale_add"):
+            return ext().row_scale_add(x, y, mask, scale)
+        m = scale if mask is None else mask.view(
+            -1, *([1] * (x.dim() - 1))) * scale
+        return x + m * y
+
+    @staticmethod
+    def backward(ctx, dz):
+        (mask,) = ctx.saved_tensors
+        mask = mask if mask.numel() else None
+        dz = dz.contiguous()
+        if dz.is_cuda and has_ext() and hasattr(ext(), "row_scale"):
+            dy = ext().row_scale(dz, mask, ctx.scale)
+        else:
+            m = ctx.scale if mask is None else mask.view(
+                -1, *([1] * (dz.dim() - 1))) * ctx.scale
+            dy = dz * m
+        return dz, dy, None, None
+
+
+def droppath_add(x: torch.Tensor, y: torch.Tensor, drop_prob: float,
+                 training: bool) -> torch.Tensor:
+    """Fused residual + stochastic depth: replaces ``x + DropPath(p)(y)``
+    (one elementwise pass; the reference chain is bernoulli/div/mul/add)."""
+    if drop_prob == 0.0 or not training:
+        return _DropPathAdd.apply(x.contiguous(), y.contiguous(), None, 1.0)
+    keep = 1.0 - drop_prob
+    mask = torch.bernoulli(
+        torch.full((x.size(0),), keep, device=x.device, dtype=torch.float32))
+    return _DropPathAdd.apply(x.contiguous(), y.contiguous(), mask,
+                              1.0 / keep)

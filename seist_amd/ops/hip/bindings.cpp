@@ -40,6 +40,10 @@ at::Tensor adam_pack(std::vector<at::Tensor> params,
                      std::vector<at::Tensor> masters, bool has_master);
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor channel_sum(const at::Tensor& in);
+at::Tensor row_scale_add(const at::Tensor& x, const at::Tensor& y,
+                         const c10::optional<at::Tensor>& mask, double scale);
+at::Tensor row_scale(const at::Tensor& y,
+                     const c10::optional<at::Tensor>& mask, double scale);
 
 void adam_step_packed(const at::Tensor& meta, const at::Tensor& sample,
                       bool has_master, double lr, double beta1, double beta2,
@@ -60,5 +64,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_pack", &adam_pack, "pack fused-adam chunk metadata");
   m.def("sum_batch", &sum_batch, "batch-axis sum to fp32");
   m.def("channel_sum", &channel_sum, "per-channel sum to fp32");
+  m.def("row_scale_add", &row_scale_add, "z = x + mask[n]*scale*y");
+  m.def("row_scale", &row_scale, "z = mask[n]*scale*y");
   m.def("adam_step_packed", &adam_step_packed, "fused adam step");
 }

@@ -23,6 +23,7 @@
 #include "sa_common.h"
 
 at::Tensor sum_batch(const at::Tensor& in);
+at::Tensor sum_mid(const at::Tensor& in);
 at::Tensor channel_sum(const at::Tensor& in);
 
 namespace {
@@ -426,25 +427,30 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   // grouped convs keep the direct kernel: at Cog=Cig=8 the per-tap
   // batched GEMM is launch/overhead-bound and measured slower.
   if (groups == 1 && stride == 1) {
-    auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
-    const long G = groups;
+    // one bmm per tap into a (K, N, Co, Ci) slab, one middle-axis sum over
+    // N for all taps, one strided scatter into the (Co, Ci, K) layout
+    auto slab = at::zeros({(long)K, (long)N, (long)Co, (long)Cig},
+                          x.options());
     for (int k = 0; k < K; ++k) {
       const long off = (long)k * dilation - padl;
       const long lo0 = std::max<long>(0, -off);
       const long lo1 = std::min<long>(Lo, L - off);
       if (lo1 <= lo0) continue;
       const long l = lo1 - lo0;
-      auto dyv = at::as_strided(dy, {N * G, Cog, l},
-                                {(long)Cog * Lo, Lo, 1},
+      auto dyv = at::as_strided(dy, {(long)N, (long)Co, l},
+                                {(long)Co * Lo, Lo, 1},
                                 dy.storage_offset() + lo0);
-      auto xv = at::as_strided(x, {N * G, Cig, l},
+      auto xv = at::as_strided(x, {(long)N, (long)Cig, l},
                                {(long)Cig * L, L, 1},
                                x.storage_offset() + lo0 + off);
-      auto prod = sum_batch(at::bmm(dyv, xv.transpose(1, 2))
-                                .view({(long)N, G * Cog, (long)Cig}));
-      dw32.select(2, k).add_(prod.view({(long)Co, (long)Cig}));
+      auto out_k = slab.select(0, k);
+      at::bmm_out(out_k, dyv, xv.transpose(1, 2));
     }
-    auto dw = dw32.to(w.scalar_type());
+    auto dw = sum_mid(slab.view({(long)K, (long)N, (long)Co * Cig}))
+                  .view({(long)K, (long)Co, (long)Cig})
+                  .permute({1, 2, 0})
+                  .contiguous()
+                  .to(w.scalar_type());
     at::Tensor db;
     if (has_bias) {
       db = channel_sum(dy).to(w.scalar_type());

@@ -61,7 +61,7 @@ __global__ void channel_sum_kernel(const scalar_t* __restrict__ in,
     for (long l = threadIdx.x; l < L; l += kBlock) s += (float)r[l];
   }
   s = sa::block_reduce_sum(s, red);
-  if (threadIdx.x == 0) part[(long)c * nsplit + split] = s;
+  if (threadIdx.x == 0) part[(long)split * C + c] = s;
 }
 
 __global__ void part_sum_kernel(const float* __restrict__ part,
@@ -69,11 +69,50 @@ __global__ void part_sum_kernel(const float* __restrict__ part,
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float s = 0.0f;
-  for (int j = 0; j < nsplit; ++j) s += part[(long)c * nsplit + j];
+  for (int j = 0; j < nsplit; ++j) s += part[(long)j * C + c];
   out[c] = s;
 }
 
+template <typename scalar_t>
+__global__ void sum_mid_kernel(const scalar_t* __restrict__ in,
+                               float* __restrict__ out,
+                               long A, long B, long M) {
+  const long i = (long)blockIdx.x * kBlock + threadIdx.x;  // a*M + j
+  if (i >= A * M) return;
+  const long a = i / M;
+  const long j = i - a * M;
+  const scalar_t* p = in + (a * B) * M + j;
+  float s = 0.0f;
+  long b = 0;
+  for (; b + 4 <= B; b += 4) {
+    s += (float)p[b * M] + (float)p[(b + 1) * M]
+         + (float)p[(b + 2) * M] + (float)p[(b + 3) * M];
+  }
+  for (; b < B; ++b) s += (float)p[b * M];
+  out[i] = s;
+}
+
 }  // namespace
+
+// (A, B, M) -> (A, M) fp32, middle axis reduced — used to collapse the
+// batch axis of all K tap-GEMMs of a conv weight gradient in one launch.
+at::Tensor sum_mid(const at::Tensor& in) {
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() >= 3);
+  const long A = in.size(0);
+  const long B = in.size(1);
+  const long M = in.numel() / (A * B);
+  auto out = at::empty({A, M}, in.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, in.scalar_type(),
+      "sum_mid", [&] {
+        hipLaunchKernelGGL((sum_mid_kernel<scalar_t>),
+                           dim3(sa::ceil_div(A * M, kBlock)), dim3(kBlock),
+                           0, stream.stream(), in.data_ptr<scalar_t>(),
+                           out.data_ptr<float>(), A, B, M);
+      });
+  return out;
+}
 
 at::Tensor sum_batch(const at::Tensor& in) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() >= 2);
@@ -111,7 +150,7 @@ at::Tensor channel_sum(const at::Tensor& in) {
   const long L = in.size(2);
   const int nsplit = std::max(1, std::min<int>(
       (int)N, 2048 / std::max(C, 1)));
-  auto part = at::empty({C, nsplit}, in.options().dtype(at::kFloat));
+  auto part = at::empty({nsplit, C}, in.options().dtype(at::kFloat));
   auto out = at::empty({C}, in.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(

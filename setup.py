@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "adam.hip"),
         os.path.join(HIP_DIR, "reductions.hip"),
         os.path.join(HIP_DIR, "pw_mfma.hip"),
+        os.path.join(HIP_DIR, "rowscale.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
