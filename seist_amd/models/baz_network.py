@@ -14,17 +14,53 @@ from ._registry import register_model
 
 
 def _sym3_eig(cov: torch.Tensor):
-    """Batched eigendecomposition of (near-)symmetric 3x3 covariance
-    matrices. Uses torch.linalg.eigh (covariances are symmetric PSD by
-    construction), returning eigenvalues descending to match the magnitude
-    ordering that `torch.linalg.eig` yields for PSD input."""
-    vals, vecs = torch.linalg.eigh(cov)
-    # eigh returns ascending; reference torch.linalg.eig returns unordered
-    # but for PSD matrices the downstream net only consumes the set —
-    # fix descending for determinism.
-    vals = vals.flip(-1)
-    vecs = vecs.flip(-1)
-    return vals.unsqueeze(-1), vecs
+    """Batched closed-form eigendecomposition of symmetric 3x3 covariance
+    matrices (K16 of SURVEY.md §2.4) — the analytic trigonometric method
+    for symmetric 3x3 eigenvalues plus null-space cross products for the
+    eigenvectors. Pure tensor math: runs on the GPU with no LAPACK/MAGMA
+    dependency, identically on CPU. Eigenvalues descending (the reference's
+    `torch.linalg.eig` ordering is unspecified; the downstream conv only
+    consumes the eigenstructure)."""
+    A = cov.double()
+    N = A.shape[0]
+    a, b, c = A[:, 0, 0], A[:, 1, 1], A[:, 2, 2]
+    d, e, f = A[:, 0, 1], A[:, 1, 2], A[:, 0, 2]
+
+    q = (a + b + c) / 3.0
+    p1 = d * d + e * e + f * f
+    p2 = ((a - q) ** 2 + (b - q) ** 2 + (c - q) ** 2 + 2.0 * p1)
+    p = torch.sqrt(torch.clamp(p2 / 6.0, min=1e-300))
+    eye = torch.eye(3, dtype=A.dtype, device=A.device).expand(N, 3, 3)
+    B = (A - q[:, None, None] * eye) / p[:, None, None]
+    r = torch.linalg.det(B) / 2.0  # 3x3 determinant: closed-form in ATen
+    r = torch.clamp(r, -1.0, 1.0)
+    phi = torch.acos(r) / 3.0
+    two_pi_3 = 2.0943951023931953
+    l0 = q + 2.0 * p * torch.cos(phi)
+    l2 = q + 2.0 * p * torch.cos(phi + two_pi_3)
+    l1 = 3.0 * q - l0 - l2
+    vals = torch.stack([l0, l1, l2], dim=-1)  # descending
+
+    # eigenvector for each eigenvalue: cross product of two rows of A - l*I
+    def eigvec(lam):
+        M = A - lam[:, None, None] * eye
+        c01 = torch.cross(M[:, 0], M[:, 1], dim=-1)
+        c02 = torch.cross(M[:, 0], M[:, 2], dim=-1)
+        c12 = torch.cross(M[:, 1], M[:, 2], dim=-1)
+        cands = torch.stack([c01, c02, c12], dim=1)  # (N,3,3)
+        norms = cands.norm(dim=-1)
+        best = norms.argmax(dim=1)
+        v = cands[torch.arange(N, device=A.device), best]
+        return v / torch.clamp(v.norm(dim=-1, keepdim=True), min=1e-30)
+
+    vecs = torch.stack([eigvec(l0), eigvec(l1), eigvec(l2)], dim=-1)
+    # degenerate (isotropic) case: A ~ q*I -> any orthonormal basis
+    iso = p2 < 1e-24
+    if iso.any():
+        vals = torch.where(iso[:, None], q[:, None].expand(N, 3), vals)
+        vecs = torch.where(iso[:, None, None],
+                           eye.to(vecs.dtype), vecs)
+    return (vals.unsqueeze(-1).to(cov.dtype), vecs.to(cov.dtype))
 
 
 class BAZ_Network(nn.Module):

@@ -429,13 +429,16 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   if (groups == 1 && stride == 1) {
     // one bmm per tap into a (K, N, Co, Ci) slab, one middle-axis sum over
     // N for all taps, one strided scatter into the (Co, Ci, K) layout
-    auto slab = at::zeros({(long)K, (long)N, (long)Co, (long)Cig},
+    auto slab = at::empty({(long)K, (long)N, (long)Co, (long)Cig},
                           x.options());
     for (int k = 0; k < K; ++k) {
       const long off = (long)k * dilation - padl;
       const long lo0 = std::max<long>(0, -off);
       const long lo1 = std::min<long>(Lo, L - off);
-      if (lo1 <= lo0) continue;
+      if (lo1 <= lo0) {
+        slab.select(0, k).zero_();  // fully clipped tap
+        continue;
+      }
       const long l = lo1 - lo0;
       auto dyv = at::as_strided(dy, {(long)N, (long)Co, l},
                                 {(long)Co * Lo, Lo, 1},

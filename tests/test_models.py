@@ -131,3 +131,18 @@ def test_checkpoint_strips_wrapper_prefixes(tmp_path):
     from seist_amd.models import load_checkpoint
     ck = load_checkpoint(path)
     m.load_state_dict(ck["model_dict"])
+
+
+def test_sym3_eig_analytic():
+    """K16: closed-form symmetric 3x3 eig vs LAPACK eigh."""
+    from seist_amd.models.baz_network import _sym3_eig
+    torch.manual_seed(0)
+    x = torch.randn(32, 3, 500)
+    d = x - x.mean(-1, keepdim=True)
+    cov = torch.matmul(d, d.transpose(1, 2)) / 499
+    vals, vecs = _sym3_eig(cov)
+    ref = torch.linalg.eigh(cov)[0].flip(-1)
+    assert torch.allclose(vals.squeeze(-1), ref, atol=1e-4, rtol=1e-4)
+    # eigen equation A v = lambda v
+    err = (torch.matmul(cov, vecs) - vals.transpose(1, 2) * vecs).abs().max()
+    assert err < 1e-4
