@@ -24,6 +24,9 @@
 
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor sum_mid(const at::Tensor& in);
+bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
+               const c10::optional<at::Tensor>& bias, at::Tensor& y,
+               long padl, long dilation, bool is_dx);
 at::Tensor channel_sum(const at::Tensor& in);
 
 namespace {
@@ -329,6 +332,12 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
   const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
 
+  // dense stride-1 convs go to the matrix cores
+  if (groups == 1 && stride == 1
+      && conv_mfma(x, w, bias, y, padl, dilation, /*is_dx=*/false)) {
+    return y;
+  }
+
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "conv1d_fwd", [&] {
@@ -388,6 +397,10 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto stream = at::hip::getCurrentHIPStream();
 
   auto dx = at::empty_like(x);
+  bool dx_done = (groups == 1 && stride == 1
+                  && conv_mfma(dy, w, c10::nullopt, dx, padl, dilation,
+                               /*is_dx=*/true));
+  if (!dx_done)
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "conv1d_dx", [&] {

@@ -186,3 +186,172 @@ bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
   }
   return true;
 }
+
+// ---------------------------------------------------------------------------
+// Dense stride-1 Conv1d on the matrix cores: the conv is a GEMM with
+// K_eff = Ci*K where the B operand rows are k-shifted x rows ("im2col in
+// LDS" — nothing is materialized in HBM). Handles forward
+// (y[co][l] = sum_{ci,k} w[co][ci][k] x[ci][l - padl + k*d]) and the
+// input gradient (IS_DX: dx[ci][l] = sum_{co,k} w[co][ci][k]
+// dy[co][l + padl - k*d]) with the same structure.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <bool IS_DX, bool HAS_BIAS>
+__global__ __launch_bounds__(kBlock)
+void conv_mfma_kernel(const sa_bf16* __restrict__ x,
+                      const sa_bf16* __restrict__ w,
+                      const sa_bf16* __restrict__ bias,
+                      sa_bf16* __restrict__ y,
+                      int N, int Cin, int Cout, long Lin, long Lout,
+                      int K, int padl, int dil) {
+  __shared__ sa_bf16 w_s[kCoT * kWPitch];
+  __shared__ sa_bf16 x_s[kKT * kXPitch];
+
+  const int n = blockIdx.y;
+  const int m0 = blockIdx.z * kCoT;
+  const long l0 = (long)blockIdx.x * kLT;
+  const int KK = Cin * K;  // effective GEMM K
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+  const int frag_m = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc3 = {0.f, 0.f, 0.f, 0.f};
+
+  const sa_bf16* xb = x + (long)n * Cin * Lin;
+
+  for (int k0 = 0; k0 < KK; k0 += kKT) {
+    __syncthreads();
+    // ---- weight chunk [kCoT m][kKT kk] ----
+    for (int idx = tid; idx < kCoT * kKT; idx += kBlock) {
+      const int m = idx / kKT;
+      const int kk = idx - m * kKT;
+      const int mg = m0 + m;
+      const int kkg = k0 + kk;
+      float v = 0.0f;
+      if (mg < Cout && kkg < KK) {
+        if (IS_DX) {
+          const int co = kkg / K;
+          const int k = kkg - co * K;
+          v = (float)w[((long)co * Cout + mg) * K + k];
+        } else {
+          v = (float)w[(long)mg * KK + kkg];
+        }
+      }
+      w_s[m * kWPitch + kk] = (sa_bf16)v;
+    }
+    // ---- shifted-row chunk [kKT kk][kLT l] ----
+    for (int idx = tid; idx < kKT * (kLT / 8); idx += kBlock) {
+      const int kk = idx / (kLT / 8);
+      const int c8 = idx - kk * (kLT / 8);
+      const int kkg = k0 + kk;
+      bf16x8 v = {};
+      if (kkg < KK) {
+        const int c = kkg / K;          // ci (fwd) or co (dx)
+        const int k = kkg - c * K;
+        const long shift = IS_DX ? (long)padl - (long)k * dil
+                                 : (long)k * dil - (long)padl;
+        const long lg = l0 + c8 * 8 + shift;
+        const sa_bf16* row = xb + (long)c * Lin;
+        if (lg >= 0 && lg + 8 <= Lin) {
+          v = *(const bf16x8*)(row + lg);
+        } else {
+          for (int j = 0; j < 8; ++j) {
+            const long lj = lg + j;
+            v[j] = (lj >= 0 && lj < Lin) ? row[lj] : (sa_bf16)0.f;
+          }
+        }
+      }
+      *(bf16x8*)(x_s + kk * kXPitch + c8 * 8) = v;
+    }
+    __syncthreads();
+
+    const bf16x8 a =
+        *(const bf16x8*)(w_s + (wr * 16 + frag_m) * kWPitch + kbase);
+    const int lb = wc * 64 + frag_m;
+#pragma unroll
+    for (int nrep = 0; nrep < 4; ++nrep) {
+      bf16x8 b;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        b[j] = x_s[(kbase + j) * kXPitch + nrep * 16 + lb];
+      }
+      switch (nrep) {
+        case 0: acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0); break;
+        case 1: acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc1, 0, 0, 0); break;
+        case 2: acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc2, 0, 0, 0); break;
+        case 3: acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc3, 0, 0, 0); break;
+      }
+    }
+  }
+
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int nrep = 0; nrep < 4; ++nrep) {
+    const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
+                      : (nrep == 2) ? acc2 : acc3;
+    const long lg = l0 + wc * 64 + nrep * 16 + d_col;
+    if (lg >= Lout) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mg = m0 + wr * 16 + d_row0 + r;
+      if (mg < Cout) {
+        float v = acc[r];
+        if (HAS_BIAS) v += (float)bias[mg];
+        y[((long)n * Cout + mg) * Lout + lg] = (sa_bf16)v;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// Dense stride-1 conv fwd/dx on MFMA; returns false if not applicable.
+bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
+               const c10::optional<at::Tensor>& bias, at::Tensor& y,
+               long padl, long dilation, bool is_dx) {
+  if (x.scalar_type() != at::ScalarType::BFloat16) return false;
+  const int N = x.size(0), Cin = x.size(1);
+  const long Lin = x.size(2);
+  const int Cout = y.size(1);
+  const long Lout = y.size(2);
+  const int K = w.size(2);
+  if (Cin * K < 32) return false;
+
+  const bool has_bias = bias.has_value() && bias->defined();
+  at::Tensor bct;
+  if (has_bias) bct = bias->to(x.scalar_type()).contiguous();
+
+  dim3 grid(sa::ceil_div(Lout, kLT), N, sa::ceil_div(Cout, kCoT));
+  auto stream = at::hip::getCurrentHIPStream();
+  const sa_bf16* xp = (const sa_bf16*)x.data_ptr();
+  const sa_bf16* wp = (const sa_bf16*)w.data_ptr();
+  const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
+  sa_bf16* yp = (sa_bf16*)y.data_ptr();
+
+  auto launch = [&](auto dxp_, auto hb_) {
+    hipLaunchKernelGGL((conv_mfma_kernel<decltype(dxp_)::value,
+                                         decltype(hb_)::value>),
+                       grid, dim3(kBlock), 0, stream.stream(), xp, wp, bp,
+                       yp, N, Cin, Cout, Lin, Lout, K, (int)padl,
+                       (int)dilation);
+  };
+  if (is_dx) {
+    if (has_bias) launch(std::true_type{}, std::true_type{});
+    else launch(std::true_type{}, std::false_type{});
+  } else {
+    if (has_bias) launch(std::false_type{}, std::true_type{});
+    else launch(std::false_type{}, std::false_type{});
+  }
+  return true;
+}

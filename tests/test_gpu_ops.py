@@ -188,3 +188,37 @@ def test_fused_adam_gpu_matches_cpu(dev, dtype):
         # fp32 masters on both sides must agree closely
         _cmp(og.state[pg]["master"], oc.state[pc]["master"], 1e-4,
              msg="adam master")
+
+
+@pytest.mark.parametrize("Ci,Co,K,pad,dil", [
+    (16, 8, 7, 3, 1), (64, 64, 3, 1, 1), (20, 20, 6, 320, 64),
+    (96, 64, 7, 3, 1), (8, 16, 11, 5, 1),
+])
+def test_dense_conv_mfma_bf16(dev, Ci, Co, K, pad, dil):
+    """Dense stride-1 conv fwd/dx on the matrix cores vs bf16-quantized
+    CPU reference."""
+    torch.manual_seed(4)
+    N, L = 3, 1024
+    padl = padr = pad if dil == 1 else (K - 1) * dil // 2
+    x32 = torch.randn(N, Ci, L)
+    w32 = torch.randn(Co, Ci, K) * 0.2
+    b32 = torch.randn(Co) * 0.1
+
+    xg = x32.to(dev, torch.bfloat16).requires_grad_(True)
+    wg = w32.to(dev, torch.bfloat16).requires_grad_(True)
+    bg = b32.to(dev, torch.bfloat16).requires_grad_(True)
+    y = ops.conv1d(xg, wg, bg, stride=1, padding=(padl, padr), dilation=dil)
+
+    xc = x32.to(torch.bfloat16).float().requires_grad_(True)
+    wc = w32.to(torch.bfloat16).float().requires_grad_(True)
+    bc = b32.to(torch.bfloat16).float().requires_grad_(True)
+    y_ref = ops.conv1d(xc, wc, bc, stride=1, padding=(padl, padr),
+                       dilation=dil)
+    _cmp(y, y_ref, 5e-2, 1e-3, msg="conv mfma fwd")
+
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev, torch.bfloat16))
+    y_ref.backward(g32.to(torch.bfloat16).float())
+    _cmp(xg.grad, xc.grad, 1e-1, 2e-3, msg="conv mfma dx")
+    _cmp(wg.grad, wc.grad, 1.0, 5e-3, msg="conv mfma dw")
+    _cmp(bg.grad, bc.grad, 1.0, 5e-3, msg="conv mfma db")
