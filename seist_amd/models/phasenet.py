@@ -8,7 +8,7 @@ match the reference for checkpoint interop.
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
+import torch.nn.functional as F  # noqa: F401
 
 from .. import ops
 from ._blocks import run_bn, run_conv
@@ -79,7 +79,8 @@ class ConvTransBlock(nn.Module):
             x = run_bn(self.bn0, x, act="relu")
         x = self.drop0(x)
         if isinstance(self.convt, nn.ConvTranspose1d):
-            x = self.convt(x)
+            x = ops.conv_transpose1d(x, self.convt.weight, self.convt.bias,
+                                     stride=self.convt.stride[0])
             x = run_bn(self.bn1, x, act="relu")
         return self.drop1(x)
 

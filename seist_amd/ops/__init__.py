@@ -56,6 +56,7 @@ from .functional import (  # noqa: E402,F401
     droppath_add,
     bn_act,
     conv1d,
+    conv_transpose1d,
     gelu,
     interp_linear,
     pointwise_conv,

@@ -404,3 +404,43 @@ def droppath_add(x: torch.Tensor, y: torch.Tensor, drop_prob: float,
         torch.full((x.size(0),), keep, device=x.device, dtype=torch.float32))
     return _DropPathAdd.apply(x.contiguous(), y.contiguous(), mask,
                               1.0 / keep)
+
+
+# ---------------------------------------------------------------------------
+# ConvTranspose1d (K5) — PhaseNet decoder
+# ---------------------------------------------------------------------------
+
+
+class _ConvTranspose1d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride):
+        ctx.save_for_backward(x, weight)
+        ctx.stride = stride
+        ctx.has_bias = bias is not None
+        if use_native(x) and hasattr(ext(), "conv_transpose1d_fwd"):
+            return ext().conv_transpose1d_fwd(x, weight, bias, stride)
+        return F.conv_transpose1d(x, weight, bias, stride=stride)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        if use_native(x) and hasattr(ext(), "conv_transpose1d_bwd"):
+            dx, dw, db = ext().conv_transpose1d_bwd(dy, x, weight,
+                                                    ctx.stride, ctx.has_bias)
+        else:
+            xd = x.detach().requires_grad_(True)
+            wd = weight.detach().requires_grad_(True)
+            with torch.enable_grad():
+                out = F.conv_transpose1d(xd, wd, None, stride=ctx.stride)
+            dx, dw = torch.autograd.grad(out, [xd, wd], dy)
+            db = dy.sum(dim=(0, 2)) if ctx.has_bias else None
+        return dx, dw, db, None
+
+
+def conv_transpose1d(x: torch.Tensor, weight: torch.Tensor,
+                     bias: Optional[torch.Tensor] = None,
+                     stride: int = 1) -> torch.Tensor:
+    """ConvTranspose1d, weight (Ci, Co, K), no padding (PhaseNet's usage)."""
+    return _ConvTranspose1d.apply(x.contiguous(), weight.contiguous(), bias,
+                                  stride)

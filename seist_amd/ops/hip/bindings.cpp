@@ -14,6 +14,13 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
                                    const at::Tensor& w, long stride,
                                    long padl, long padr, long groups,
                                    long dilation, bool has_bias);
+at::Tensor conv_transpose1d_fwd(const at::Tensor& x, const at::Tensor& w,
+                                const c10::optional<at::Tensor>& bias,
+                                long stride);
+std::vector<at::Tensor> conv_transpose1d_bwd(const at::Tensor& dy,
+                                             const at::Tensor& x,
+                                             const at::Tensor& w,
+                                             long stride, bool has_bias);
 
 std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
                                    const at::Tensor& beta,
@@ -55,6 +62,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pw_conv_bwd", &pw_conv_bwd, "pointwise conv backward");
   m.def("conv1d_fwd", &conv1d_fwd, "direct conv1d forward");
   m.def("conv1d_bwd", &conv1d_bwd, "direct conv1d backward");
+  m.def("conv_transpose1d_fwd", &conv_transpose1d_fwd,
+        "transposed conv1d forward");
+  m.def("conv_transpose1d_bwd", &conv_transpose1d_bwd,
+        "transposed conv1d backward");
   m.def("bn_act_fwd", &bn_act_fwd, "fused batchnorm+act forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused batchnorm+act backward");
   m.def("avgmax_pool_fwd", &avgmax_pool_fwd, "fused avg+max pool forward");

@@ -384,25 +384,26 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
   return y;
 }
 
-std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
-                                   const at::Tensor& w, long stride,
-                                   long padl, long padr, long groups,
-                                   long dilation, bool has_bias) {
-  const int N = x.size(0), Ci = x.size(1);
-  const long L = x.size(2);
+// Gather pass shared by the conv input-gradient and the transposed-conv
+// forward: out[n][ci][li] = sum_{co in group, k} in[n][co][(li + padl -
+// k*d)/stride] * w[co][cig][k] (only where divisible/in range).
+void conv1d_dx_into(const at::Tensor& dy, const at::Tensor& w,
+                    at::Tensor& dx, long stride, long padl, long groups,
+                    long dilation) {
+  const int N = dx.size(0), Ci = dx.size(1);
+  const long L = dx.size(2);
   const int Co = w.size(0), K = w.size(2);
   const long Lo = dy.size(2);
   const int Cig = Ci / groups;
   const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
 
-  auto dx = at::empty_like(x);
-  bool dx_done = (groups == 1 && stride == 1
-                  && conv_mfma(dy, w, c10::nullopt, dx, padl, dilation,
-                               /*is_dx=*/true));
-  if (!dx_done)
+  if (groups == 1 && stride == 1
+      && conv_mfma(dy, w, c10::nullopt, dx, padl, dilation, /*is_dx=*/true)) {
+    return;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(
-      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      at::ScalarType::BFloat16, at::ScalarType::Half, dx.scalar_type(),
       "conv1d_dx", [&] {
         if (Cog == 1 && Cig == 1) {
           dim3 grid(sa::ceil_div(L, kBlock), N, Ci);
@@ -431,6 +432,22 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
                              (int)groups, cog_chunk);
         }
       });
+}
+
+std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                   const at::Tensor& w, long stride,
+                                   long padl, long padr, long groups,
+                                   long dilation, bool has_bias) {
+  const int N = x.size(0), Ci = x.size(1);
+  const long L = x.size(2);
+  const int Co = w.size(0), K = w.size(2);
+  const long Lo = dy.size(2);
+  const int Cig = Ci / groups;
+  const int Cog = Co / groups;
+  auto stream = at::hip::getCurrentHIPStream();
+
+  auto dx = at::empty_like(x);
+  conv1d_dx_into(dy, w, dx, stride, padl, groups, dilation);
 
   // stride-1 convs (dense AND grouped): dw is K shifted plain GEMMs — run
   // them on the matrix cores via rocBLAS strided-batched bmm over zero-copy
@@ -509,5 +526,51 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dw = dw32.to(w.scalar_type());
   at::Tensor db;
   if (has_bias) db = db32.to(w.scalar_type());
+  return {dx, dw, db};
+}
+
+// ---------------------------------------------------------------------------
+// ConvTranspose1d (K5) — PhaseNet's up path (reference phasenet.py:118-127).
+// Forward is the conv input-gradient gather with roles swapped; the input
+// gradient is a strided conv forward of dY with w viewed (Ci, Co, K) — the
+// exact layout ConvTranspose1d stores; the weight gradient is the conv
+// weight gradient with x and dY swapped.
+// ---------------------------------------------------------------------------
+
+at::Tensor conv_transpose1d_fwd(const at::Tensor& x, const at::Tensor& w,
+                                const c10::optional<at::Tensor>& bias,
+                                long stride) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
+  const int N = x.size(0);
+  const int Ci = x.size(1);   // convT input channels
+  const long Li = x.size(2);
+  const int Co = w.size(1);   // w: (Ci, Co, K)
+  const int K = w.size(2);
+  TORCH_CHECK(w.size(0) == Ci, "conv_transpose weight/input mismatch");
+  const long Lo = (Li - 1) * stride + K;
+  auto y = at::empty({(long)N, (long)Co, Lo}, x.options());
+  // view w (Ci, Co, K) as a conv weight (Co_conv=Ci, Cig=Co, K)
+  conv1d_dx_into(/*dy=*/x, /*w=*/w, /*dx=*/y, stride, /*padl=*/0,
+                 /*groups=*/1, /*dilation=*/1);
+  if (bias.has_value() && bias->defined()) {
+    y.add_(bias->to(y.scalar_type()).view({1, (long)Co, 1}));
+  }
+  return y;
+}
+
+std::vector<at::Tensor> conv_transpose1d_bwd(const at::Tensor& dy,
+                                             const at::Tensor& x,
+                                             const at::Tensor& w,
+                                             long stride, bool has_bias) {
+  // dX = strided conv forward of dY with weight (Ci, Co, K) as-is
+  auto dx = conv1d_fwd(dy, w, c10::nullopt, stride, /*padl=*/0,
+                       /*padr=*/0, /*groups=*/1, /*dilation=*/1);
+  // dW[ci][co][k]: conv weight gradient with (dy:=x, x:=dY)
+  auto dws = conv1d_bwd(/*dy=*/x, /*x=*/dy, /*w=*/w, stride, /*padl=*/0,
+                        /*padr=*/0, /*groups=*/1, /*dilation=*/1,
+                        /*has_bias=*/false);
+  auto dw = dws[1];
+  at::Tensor db;
+  if (has_bias) db = channel_sum(dy).to(w.scalar_type());
   return {dx, dw, db};
 }

@@ -222,3 +222,26 @@ def test_dense_conv_mfma_bf16(dev, Ci, Co, K, pad, dil):
     _cmp(xg.grad, xc.grad, 1e-1, 2e-3, msg="conv mfma dx")
     _cmp(wg.grad, wc.grad, 1.0, 5e-3, msg="conv mfma dw")
     _cmp(bg.grad, bc.grad, 1.0, 5e-3, msg="conv mfma db")
+
+
+@pytest.mark.parametrize("Ci,Co,K,stride", [(128, 64, 7, 4), (8, 8, 7, 4)])
+def test_conv_transpose1d_gpu(dev, Ci, Co, K, stride):
+    torch.manual_seed(5)
+    x32 = torch.randn(3, Ci, 128)
+    w32 = torch.randn(Ci, Co, K) * 0.2
+    b32 = torch.randn(Co) * 0.1
+    xg = x32.to(dev).requires_grad_(True)
+    wg = w32.to(dev).requires_grad_(True)
+    bg = b32.to(dev).requires_grad_(True)
+    y = ops.conv_transpose1d(xg, wg, bg, stride=stride)
+    xc = x32.clone().requires_grad_(True)
+    wc = w32.clone().requires_grad_(True)
+    bc = b32.clone().requires_grad_(True)
+    y_ref = ops.conv_transpose1d(xc, wc, bc, stride=stride)
+    _cmp(y, y_ref, 1e-4, msg="convT fwd")
+    g32 = torch.randn_like(y_ref)
+    y.backward(g32.to(dev))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, 1e-4, msg="convT dx")
+    _cmp(wg.grad, wc.grad, 1e-3, 1e-3, msg="convT dw")
+    _cmp(bg.grad, bc.grad, 1e-3, 1e-3, msg="convT db")

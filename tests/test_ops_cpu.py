@@ -134,3 +134,22 @@ def test_pooled_attention_matches_manual():
     attn = (q / math.sqrt(E)).transpose(-1, -2) @ k
     ref = (attn.softmax(-1) @ v.transpose(-1, -2)).transpose(-1, -2)
     assert torch.allclose(out, ref, atol=1e-6)
+
+
+@pytest.mark.parametrize("Ci,Co,K,stride", [(8, 4, 7, 4), (16, 8, 7, 4),
+                                            (4, 4, 3, 2)])
+def test_conv_transpose1d_matches_torch(Ci, Co, K, stride):
+    x = torch.randn(2, Ci, 64, requires_grad=True)
+    w = (torch.randn(Ci, Co, K) * 0.2).requires_grad_(True)
+    b = torch.randn(Co, requires_grad=True)
+    y = ops.conv_transpose1d(x, w, b, stride=stride)
+    y_ref = F.conv_transpose1d(x, w, b, stride=stride)
+    assert torch.allclose(y, y_ref, atol=1e-5)
+    g = torch.randn_like(y_ref)
+    y.backward(g)
+    gx, gw, gb = x.grad.clone(), w.grad.clone(), b.grad.clone()
+    x.grad = w.grad = b.grad = None
+    y_ref.backward(g)
+    assert torch.allclose(gx, x.grad, atol=1e-5)
+    assert torch.allclose(gw, w.grad, atol=1e-4)
+    assert torch.allclose(gb, b.grad, atol=1e-4)
