@@ -434,10 +434,11 @@ void conv1d_dx_into(const at::Tensor& dy, const at::Tensor& w,
       });
 }
 
-std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
-                                   const at::Tensor& w, long stride,
-                                   long padl, long padr, long groups,
-                                   long dilation, bool has_bias) {
+std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
+                                     const at::Tensor& x,
+                                     const at::Tensor& w, long stride,
+                                     long padl, long padr, long groups,
+                                     long dilation, bool has_bias) {
   const int N = x.size(0), Ci = x.size(1);
   const long L = x.size(2);
   const int Co = w.size(0), K = w.size(2);
@@ -445,9 +446,6 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   const int Cig = Ci / groups;
   const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
-
-  auto dx = at::empty_like(x);
-  conv1d_dx_into(dy, w, dx, stride, padl, groups, dilation);
 
   // stride-1 convs (dense AND grouped): dw is K shifted plain GEMMs — run
   // them on the matrix cores via rocBLAS strided-batched bmm over zero-copy
@@ -488,7 +486,7 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
     if (has_bias) {
       db = channel_sum(dy).to(w.scalar_type());
     }
-    return {dx, dw, db};
+    return {dw, db};
   }
 
   auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));
@@ -526,7 +524,18 @@ std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dw = dw32.to(w.scalar_type());
   at::Tensor db;
   if (has_bias) db = db32.to(w.scalar_type());
-  return {dx, dw, db};
+  return {dw, db};
+}
+
+std::vector<at::Tensor> conv1d_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                   const at::Tensor& w, long stride,
+                                   long padl, long padr, long groups,
+                                   long dilation, bool has_bias) {
+  auto dx = at::empty_like(x);
+  conv1d_dx_into(dy, w, dx, stride, padl, groups, dilation);
+  auto dwdb = conv1d_dw_db(dy, x, w, stride, padl, padr, groups, dilation,
+                           has_bias);
+  return {dx, dwdb[0], dwdb[1]};
 }
 
 // ---------------------------------------------------------------------------
@@ -566,10 +575,10 @@ std::vector<at::Tensor> conv_transpose1d_bwd(const at::Tensor& dy,
   auto dx = conv1d_fwd(dy, w, c10::nullopt, stride, /*padl=*/0,
                        /*padr=*/0, /*groups=*/1, /*dilation=*/1);
   // dW[ci][co][k]: conv weight gradient with (dy:=x, x:=dY)
-  auto dws = conv1d_bwd(/*dy=*/x, /*x=*/dy, /*w=*/w, stride, /*padl=*/0,
-                        /*padr=*/0, /*groups=*/1, /*dilation=*/1,
-                        /*has_bias=*/false);
-  auto dw = dws[1];
+  auto dws = conv1d_dw_db(/*dy=*/x, /*x=*/dy, /*w=*/w, stride, /*padl=*/0,
+                          /*padr=*/0, /*groups=*/1, /*dilation=*/1,
+                          /*has_bias=*/false);
+  auto dw = dws[0];
   at::Tensor db;
   if (has_bias) db = channel_sum(dy).to(w.scalar_type());
   return {dx, dw, db};
