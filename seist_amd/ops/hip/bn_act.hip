@@ -43,8 +43,8 @@ __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
   __syncthreads();
   s2 = sa::block_reduce_sum(s2, red);
   if (threadIdx.x == 0) {
-    part[((long)split * C + c) * 2 + 0] = s;
-    part[((long)split * C + c) * 2 + 1] = s2;
+    part[((long)c * nsplit + split) * 2 + 0] = s;
+    part[((long)c * nsplit + split) * 2 + 1] = s2;
   }
 }
 
@@ -60,8 +60,8 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
   if (c >= C) return;
   float s = 0.0f, s2 = 0.0f;
   for (int j = 0; j < nsplit; ++j) {
-    s += part[((long)j * C + c) * 2 + 0];
-    s2 += part[((long)j * C + c) * 2 + 1];
+    s += part[((long)c * nsplit + j) * 2 + 0];
+    s2 += part[((long)c * nsplit + j) * 2 + 1];
   }
   const float m = s / NL;
   const float var = fmaxf(s2 / NL - m * m, 0.0f);
@@ -125,8 +125,8 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
   __syncthreads();
   s2 = sa::block_reduce_sum(s2, red);
   if (threadIdx.x == 0) {
-    part[((long)split * C + c) * 2 + 0] = s1;
-    part[((long)split * C + c) * 2 + 1] = s2;
+    part[((long)c * nsplit + split) * 2 + 0] = s1;
+    part[((long)c * nsplit + split) * 2 + 1] = s2;
   }
 }
 
@@ -138,8 +138,8 @@ __global__ void bn_part_reduce_kernel(const float* __restrict__ part,
   if (c >= C) return;
   float s1 = 0.0f, s2 = 0.0f;
   for (int j = 0; j < nsplit; ++j) {
-    s1 += part[((long)j * C + c) * 2 + 0];
-    s2 += part[((long)j * C + c) * 2 + 1];
+    s1 += part[((long)c * nsplit + j) * 2 + 0];
+    s2 += part[((long)c * nsplit + j) * 2 + 1];
   }
   out[c * 2 + 0] = s1;
   out[c * 2 + 1] = s2;
@@ -202,7 +202,7 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
   const bool has_running = running_mean.has_value() && running_mean->defined();
   if (training) {
     const int nsplit = pick_nsplit(N, C);
-    auto part = at::empty({nsplit, C, 2}, opts);
+    auto part = at::empty({C, nsplit, 2}, opts);
     mean = at::empty({C}, opts);
     invstd = at::empty({C}, opts);
     TORCH_CHECK(!has_running || running_mean->scalar_type() == at::kFloat,
@@ -259,7 +259,7 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto b32 = beta.to(at::kFloat).contiguous();
 
   const int nsplit = pick_nsplit(N, C);
-  auto part = at::empty({nsplit, C, 2}, opts);
+  auto part = at::empty({C, nsplit, 2}, opts);
   auto sums = at::empty({C, 2}, opts);  // dbeta, dgamma
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),

@@ -61,7 +61,7 @@ __global__ void channel_sum_kernel(const scalar_t* __restrict__ in,
     for (long l = threadIdx.x; l < L; l += kBlock) s += (float)r[l];
   }
   s = sa::block_reduce_sum(s, red);
-  if (threadIdx.x == 0) part[(long)split * C + c] = s;
+  if (threadIdx.x == 0) part[(long)c * nsplit + split] = s;
 }
 
 __global__ void part_sum_kernel(const float* __restrict__ part,
@@ -69,7 +69,7 @@ __global__ void part_sum_kernel(const float* __restrict__ part,
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   float s = 0.0f;
-  for (int j = 0; j < nsplit; ++j) s += part[(long)j * C + c];
+  for (int j = 0; j < nsplit; ++j) s += part[(long)c * nsplit + j];
   out[c] = s;
 }
 
@@ -150,7 +150,7 @@ at::Tensor channel_sum(const at::Tensor& in) {
   const long L = in.size(2);
   const int nsplit = std::max(1, std::min<int>(
       (int)N, 2048 / std::max(C, 1)));
-  auto part = at::empty({nsplit, C}, in.options().dtype(at::kFloat));
+  auto part = at::empty({C, nsplit}, in.options().dtype(at::kFloat));
   auto out = at::empty({C}, in.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
