@@ -173,19 +173,44 @@ def _pick_phase(outputs: torch.Tensor, prob_threshold: float,
 def _detect_event(outputs: torch.Tensor, prob_threshold: float,
                   topk: int) -> torch.Tensor:
     """Batch event detection -> (N, 2*topk) [on,off] pairs, longest first,
-    padded with [1, 0]."""
+    padded with [1, 0].
+
+    Run detection is one vectorised pass over the whole (N, L) batch
+    (the reference calls obspy's trigger_onset per trace in a Python loop,
+    postprocess.py:129); only the tiny per-row top-k selection stays
+    sequential."""
     batch = outputs.detach().float().cpu().numpy()
-    detections = []
-    for row in batch:
-        pairs = trigger_onset(row, prob_threshold, prob_threshold)
-        pairs.sort(key=lambda v: v[1] - v[0], reverse=True)
-        pairs = pairs[:topk]
-        if len(pairs) < topk:
-            pairs = pairs + [[1, 0]] * (topk - len(pairs))
-        detections.append(pairs)
-    return torch.tensor(
-        np.array(detections, dtype=np.int64).reshape(len(detections), -1),
-        dtype=torch.long, device=outputs.device)
+    N, L = batch.shape
+    above = batch > prob_threshold
+    d = np.diff(above.astype(np.int8), axis=1)
+    srow, scol = np.where(d == 1)
+    erow, ecol = np.where(d == -1)
+    # row-edge runs
+    first = np.where(above[:, 0])[0]
+    last = np.where(above[:, -1])[0]
+    srow = np.concatenate([srow, first])
+    scol = np.concatenate([scol + 1, np.zeros(len(first), dtype=scol.dtype)])
+    erow = np.concatenate([erow, last])
+    ecol = np.concatenate([ecol, np.full(len(last), L - 1,
+                                         dtype=ecol.dtype)])
+    s_order = np.lexsort((scol, srow))
+    e_order = np.lexsort((ecol, erow))
+    srow, scol = srow[s_order], scol[s_order]
+    ecol = ecol[e_order]
+    # starts and ends are paired in order within each row
+    counts = np.bincount(srow, minlength=N)
+    offsets = np.concatenate([[0], np.cumsum(counts)])
+    lengths = ecol - scol
+    out = np.tile(np.array([1, 0], dtype=np.int64), (N, topk))
+    for i in range(N):
+        lo, hi = offsets[i], offsets[i + 1]
+        if hi <= lo:
+            continue
+        order = np.argsort(-lengths[lo:hi], kind="stable")[:topk]
+        for j, oi in enumerate(order):
+            out[i, 2 * j] = scol[lo + oi]
+            out[i, 2 * j + 1] = ecol[lo + oi]
+    return torch.tensor(out, dtype=torch.long, device=outputs.device)
 
 
 def process_outputs(args: argparse.Namespace,

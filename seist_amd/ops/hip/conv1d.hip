@@ -223,10 +223,9 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
 // One block per (co, cig-pair, n-split): dy chunk staged in LDS once and
 // reused for every (cig, k); KT is the compile-time tap bound so the
 // accumulators stay in registers.
-constexpr int kCigT = 2;
 constexpr int kMaxK = 24;
 
-template <typename scalar_t, bool HAS_BIAS, int KT>
+template <typename scalar_t, bool HAS_BIAS, int KT, int kCigT>
 __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
                                  const scalar_t* __restrict__ x,
                                  float* __restrict__ dw,
@@ -494,16 +493,17 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
   if (has_bias) db32 = at::zeros({Co}, w.options().dtype(at::kFloat));
   {
     TORCH_CHECK(K <= kMaxK, "conv1d_dw: kernel taps > ", kMaxK);
+    const int cig_t = (K <= 4) ? 8 : (K <= 8) ? 4 : 2;
     const int nsplit = std::max(1, std::min<int>(
-        (int)N, 4096 / (Co * sa::ceil_div(Cig, kCigT))));
-    dim3 grid(Co, sa::ceil_div(Cig, kCigT), nsplit);
+        (int)N, 4096 / (Co * sa::ceil_div(Cig, cig_t))));
+    dim3 grid(Co, sa::ceil_div(Cig, cig_t), nsplit);
     AT_DISPATCH_FLOATING_TYPES_AND2(
         at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
         "conv1d_dw", [&] {
-          auto launch_kt = [&](auto kt, auto hb) {
+          auto launch_kt = [&](auto kt, auto ct, auto hb) {
             hipLaunchKernelGGL(
                 (conv1d_dw_kernel<scalar_t, decltype(hb)::value,
-                                  decltype(kt)::value>),
+                                  decltype(kt)::value, decltype(ct)::value>),
                 grid, dim3(kBlock), 0, stream.stream(),
                 dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
                 dw32.data_ptr<float>(),
@@ -511,14 +511,19 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
                 N, Ci, Co, L, Lo, K, (int)stride, (int)padl,
                 (int)dilation, (int)groups, nsplit);
           };
-          auto launch_hb = [&](auto kt) {
-            if (has_bias) launch_kt(kt, std::true_type{});
-            else launch_kt(kt, std::false_type{});
+          auto launch_hb = [&](auto kt, auto ct) {
+            if (has_bias) launch_kt(kt, ct, std::true_type{});
+            else launch_kt(kt, ct, std::false_type{});
           };
-          if (K <= 4) launch_hb(std::integral_constant<int, 4>{});
-          else if (K <= 8) launch_hb(std::integral_constant<int, 8>{});
-          else if (K <= 12) launch_hb(std::integral_constant<int, 12>{});
-          else launch_hb(std::integral_constant<int, kMaxK>{});
+          // accumulator budget ~32 fp32: wider cig tiles at small K
+          if (K <= 4) launch_hb(std::integral_constant<int, 4>{},
+                                std::integral_constant<int, 8>{});
+          else if (K <= 8) launch_hb(std::integral_constant<int, 8>{},
+                                     std::integral_constant<int, 4>{});
+          else if (K <= 12) launch_hb(std::integral_constant<int, 12>{},
+                                      std::integral_constant<int, 2>{});
+          else launch_hb(std::integral_constant<int, kMaxK>{},
+                         std::integral_constant<int, 2>{});
         });
   }
   auto dw = dw32.to(w.scalar_type());
