@@ -80,12 +80,20 @@ __global__ void dwconv_dx_kernel(const scalar_t* __restrict__ dy,
   const int s = (STRIDE > 0) ? STRIDE : stride;
   float acc = 0.0f;
   const scalar_t* dyr = dy + ((long)n * Co + ci) * Lo;
-  for (int k = 0; k < K; ++k) {
-    const long num = li + padl - (long)k * dil;
-    if (num < 0) continue;
-    if (STRIDE != 1 && (num % s)) continue;
-    const long lo = (STRIDE == 1) ? num : num / s;
-    if (lo < Lo) acc += w_lds[k] * (float)dyr[lo];
+  if (STRIDE != 1 && dil == 1) {
+    // phase decomposition: only k == (li+padl) mod s hits a valid lo
+    for (int k = (int)((li + padl) % s); k < K; k += s) {
+      const long lo = (li + padl - k) / s;
+      if (lo >= 0 && lo < Lo) acc += w_lds[k] * (float)dyr[lo];
+    }
+  } else {
+    for (int k = 0; k < K; ++k) {
+      const long num = li + padl - (long)k * dil;
+      if (num < 0) continue;
+      if (STRIDE != 1 && (num % s)) continue;
+      const long lo = (STRIDE == 1) ? num : num / s;
+      if (lo < Lo) acc += w_lds[k] * (float)dyr[lo];
+    }
   }
   dx[((long)n * Ci + ci) * L + li] = (scalar_t)acc;
 }
@@ -192,12 +200,17 @@ __global__ void convgemm_dx_kernel(const scalar_t* __restrict__ dy,
     }
     __syncthreads();
     if (li < L) {
+      // phase decomposition: with dil==1 only taps k == (li+padl) mod s
+      // divide evenly — iterate those directly (no per-tap modulo)
+      const int kstart = (dil == 1 && stride > 1)
+                             ? (int)((li + padl) % stride) : 0;
+      const int kstep = (dil == 1 && stride > 1) ? stride : 1;
       for (int jo = 0; jo < jn; ++jo) {
         const scalar_t* dyr = dy + ((long)n * Co + g * Cog + j0_ + jo) * Lo;
-        for (int k = 0; k < K; ++k) {
+        for (int k = kstart; k < K; k += kstep) {
           const long num = li + padl - (long)k * dil;
           if (num < 0) continue;
-          if (stride > 1 && (num % stride)) continue;
+          if (kstep == 1 && stride > 1 && (num % stride)) continue;
           const long lo = (stride > 1) ? num / stride : num;
           if (lo >= Lo) continue;
           const float dyv = (float)dyr[lo];
