@@ -16,6 +16,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from ._blocks import run_bn, run_conv
 from ._registry import register_model
 
@@ -108,7 +109,9 @@ class AttentionLayer(nn.Module):
         q = torch.matmul(x, self.Wt).unsqueeze(2)    # (N,L,1,d)
         k = torch.matmul(x, self.Wx).unsqueeze(1)    # (N,1,L,d)
         h = torch.tanh(q + k + self.bh)              # (N,L,L,d)
-        e = (torch.matmul(h, self.Wa) + self.ba).squeeze(-1)
+        # (h @ Wa) is a K=d->1 GEMV over N*L*L rows — rocBLAS runs it ~100x
+        # off the memory roofline; an elementwise multiply-reduce is fast
+        e = (h * self.Wa.view(1, 1, 1, -1)).sum(-1) + self.ba
         e = torch.exp(e - torch.max(e, dim=-1, keepdim=True).values)
         if self.attn_width is not None:
             mask = (torch.ones(e.shape[-2:], dtype=torch.bool, device=e.device)
@@ -209,7 +212,7 @@ class UpSamplingBlock(nn.Module):
                 lambda g: g.data + bias_l1_alpha * torch.sign(self.conv.bias.data))
 
     def forward(self, x):
-        x = self.upsampling(x)[:, :, : self.out_samples]
+        x = ops.upsample2x(x)[:, :, : self.out_samples]
         return run_conv(self.conv, x, *self.conv_padding_same).relu()
 
 

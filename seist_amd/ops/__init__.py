@@ -61,5 +61,6 @@ from .functional import (  # noqa: E402,F401
     interp_linear,
     pointwise_conv,
     pooled_attention,
+    upsample2x,
 )
 from .adam import FusedAdam  # noqa: E402,F401

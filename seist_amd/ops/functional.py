@@ -444,3 +444,28 @@ def conv_transpose1d(x: torch.Tensor, weight: torch.Tensor,
     """ConvTranspose1d, weight (Ci, Co, K), no padding (PhaseNet's usage)."""
     return _ConvTranspose1d.apply(x.contiguous(), weight.contiguous(), bias,
                                   stride)
+
+
+# ---------------------------------------------------------------------------
+# nearest 2x upsample (EQTransformer decoders)
+# ---------------------------------------------------------------------------
+
+
+class _Upsample2x(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        if use_native(x) and hasattr(ext(), "upsample2x_fwd"):
+            return ext().upsample2x_fwd(x)
+        return F.interpolate(x, scale_factor=2, mode="nearest")
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous()
+        if dy.is_cuda and has_ext() and hasattr(ext(), "upsample2x_bwd"):
+            return ext().upsample2x_bwd(dy)
+        return dy.view(dy.size(0), dy.size(1), -1, 2).sum(-1)
+
+
+def upsample2x(x: torch.Tensor) -> torch.Tensor:
+    """Nearest-neighbour x2 upsample along the last dim."""
+    return _Upsample2x.apply(x.contiguous())
