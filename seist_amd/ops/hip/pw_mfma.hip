@@ -316,11 +316,138 @@ void conv_mfma_kernel(const sa_bf16* __restrict__ x,
 
 }  // namespace
 
+
+// ---------------------------------------------------------------------------
+// fp32 variant on the exact f32 MFMA (v_mfma_f32_16x16x4_f32 — bitwise an
+// fmaf chain, no TF32 on gfx950; ~2.4x a VALU GEMM per the CDNA4 guide).
+// Fragment maps: A[i = lane&15][k = lane>>4], B[k = lane>>4][n = lane&15],
+// C/D as the bf16 shapes. Same shifted-row staging as the bf16 kernel.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int kXPitchF = 132;  // fp32 X row pitch (128 + 4)
+constexpr int kWPitchF = 36;   // fp32 W row pitch (32 + 4)
+
+template <bool IS_DX, bool HAS_BIAS>
+__global__ __launch_bounds__(kBlock)
+void conv_mfma_f32_kernel(const float* __restrict__ x,
+                          const float* __restrict__ w,
+                          const float* __restrict__ bias,
+                          float* __restrict__ y,
+                          int N, int Cin, int Cout, long Lin, long Lout,
+                          int K, int padl, int dil) {
+  __shared__ float w_s[kCoT * kWPitchF];
+  __shared__ float x_s[kKT * kXPitchF];
+
+  const int n = blockIdx.y;
+  const int m0 = blockIdx.z * kCoT;
+  const long l0 = (long)blockIdx.x * kLT;
+  const int KK = Cin * K;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+  const int frag_m = lane & 15;
+  const int ksub = lane >> 4;  // 0..3 fragment k offset
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc3 = {0.f, 0.f, 0.f, 0.f};
+
+  const float* xb = x + (long)n * Cin * Lin;
+
+  for (int k0 = 0; k0 < KK; k0 += kKT) {
+    __syncthreads();
+    for (int idx = tid; idx < kCoT * kKT; idx += kBlock) {
+      const int m = idx / kKT;
+      const int kk = idx - m * kKT;
+      const int mg = m0 + m;
+      const int kkg = k0 + kk;
+      float v = 0.0f;
+      if (mg < Cout && kkg < KK) {
+        if (IS_DX) {
+          const int co = kkg / K;
+          const int k = kkg - co * K;
+          v = w[((long)co * Cout + mg) * K + k];
+        } else {
+          v = w[(long)mg * KK + kkg];
+        }
+      }
+      w_s[m * kWPitchF + kk] = v;
+    }
+    for (int idx = tid; idx < kKT * (kLT / 4); idx += kBlock) {
+      const int kk = idx / (kLT / 4);
+      const int c4 = idx - kk * (kLT / 4);
+      const int kkg = k0 + kk;
+      float4 v = {0.f, 0.f, 0.f, 0.f};
+      if (kkg < KK) {
+        const int c = kkg / K;
+        const int k = kkg - c * K;
+        const long shift = IS_DX ? (long)padl - (long)k * dil
+                                 : (long)k * dil - (long)padl;
+        const long lg = l0 + c4 * 4 + shift;
+        const float* row = xb + (long)c * Lin;
+        if (lg >= 0 && lg + 4 <= Lin) {
+          v = *(const float4*)(row + lg);
+        } else {
+          float t[4];
+          for (int j = 0; j < 4; ++j) {
+            const long lj = lg + j;
+            t[j] = (lj >= 0 && lj < Lin) ? row[lj] : 0.f;
+          }
+          v = make_float4(t[0], t[1], t[2], t[3]);
+        }
+      }
+      *(float4*)(x_s + kk * kXPitchF + c4 * 4) = v;
+    }
+    __syncthreads();
+
+    const float* wrow = w_s + (wr * 16 + frag_m) * kWPitchF;
+    const int lb = wc * 64 + frag_m;
+#pragma unroll
+    for (int kk4 = 0; kk4 < kKT / 4; ++kk4) {
+      const float a = wrow[kk4 * 4 + ksub];
+      const float* xcol = x_s + (kk4 * 4 + ksub) * kXPitchF + lb;
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, xcol[0], acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, xcol[16], acc1, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, xcol[32], acc2, 0, 0, 0);
+      acc3 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, xcol[48], acc3, 0, 0, 0);
+    }
+  }
+
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int nrep = 0; nrep < 4; ++nrep) {
+    const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
+                      : (nrep == 2) ? acc2 : acc3;
+    const long lg = l0 + wc * 64 + nrep * 16 + d_col;
+    if (lg >= Lout) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mg = m0 + wr * 16 + d_row0 + r;
+      if (mg < Cout) {
+        float v = acc[r];
+        if (HAS_BIAS) v += bias[mg];
+        y[((long)n * Cout + mg) * Lout + lg] = v;
+      }
+    }
+  }
+}
+
+}  // namespace
+
 // Dense stride-1 conv fwd/dx on MFMA; returns false if not applicable.
 bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
                long padl, long dilation, bool is_dx) {
-  if (x.scalar_type() != at::ScalarType::BFloat16) return false;
+  const bool is_bf16 = x.scalar_type() == at::ScalarType::BFloat16;
+  const bool is_f32 = x.scalar_type() == at::ScalarType::Float;
+  if (!is_bf16 && !is_f32) return false;
   const int N = x.size(0), Cin = x.size(1);
   const long Lin = x.size(2);
   const int Cout = y.size(1);
@@ -334,6 +461,29 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
 
   dim3 grid(sa::ceil_div(Lout, kLT), N, sa::ceil_div(Cout, kCoT));
   auto stream = at::hip::getCurrentHIPStream();
+
+  if (is_f32) {
+    const float* xp = x.data_ptr<float>();
+    const float* wp = w.data_ptr<float>();
+    const float* bp = has_bias ? bct.data_ptr<float>() : nullptr;
+    float* yp = y.data_ptr<float>();
+    auto launch = [&](auto dxp_, auto hb_) {
+      hipLaunchKernelGGL((conv_mfma_f32_kernel<decltype(dxp_)::value,
+                                               decltype(hb_)::value>),
+                         grid, dim3(kBlock), 0, stream.stream(), xp, wp, bp,
+                         yp, N, Cin, Cout, Lin, Lout, K, (int)padl,
+                         (int)dilation);
+    };
+    if (is_dx) {
+      if (has_bias) launch(std::true_type{}, std::true_type{});
+      else launch(std::true_type{}, std::false_type{});
+    } else {
+      if (has_bias) launch(std::false_type{}, std::true_type{});
+      else launch(std::false_type{}, std::false_type{});
+    }
+    return true;
+  }
+
   const sa_bf16* xp = (const sa_bf16*)x.data_ptr();
   const sa_bf16* wp = (const sa_bf16*)w.data_ptr();
   const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
