@@ -280,6 +280,16 @@ def train_worker(args, device) -> str:
         logger.info("model.load_state_dict")
 
     if pdist.is_main_process():
+        # back up the model source next to the run artifacts
+        # (reference train.py:288-291)
+        try:
+            import inspect
+            import shutil
+            src = inspect.getfile(model.__class__)
+            shutil.copy2(src, get_safe_path(
+                os.path.join(log_dir, "model_backup.py")))
+        except Exception as e:
+            logger.warning(f"model source backup skipped: {e}")
         logger.info(f"Model parameters: {count_parameters(model)}")
 
     if args.precision == "bf16":

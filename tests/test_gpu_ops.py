@@ -245,3 +245,31 @@ def test_conv_transpose1d_gpu(dev, Ci, Co, K, stride):
     _cmp(xg.grad, xc.grad, 1e-4, msg="convT dx")
     _cmp(wg.grad, wc.grad, 1e-3, 1e-3, msg="convT dw")
     _cmp(bg.grad, bc.grad, 1e-3, 1e-3, msg="convT db")
+
+
+def test_droppath_add_gpu(dev):
+    torch.manual_seed(1)
+    x32 = torch.randn(16, 4, 32)
+    y32 = torch.randn(16, 4, 32)
+    xg = x32.to(dev).requires_grad_(True)
+    yg = y32.to(dev).requires_grad_(True)
+    out = ops.droppath_add(xg, yg, 0.0, training=True)  # scale path, no mask
+    _cmp(out, x32 + y32, 1e-6, msg="droppath add")
+    g = torch.randn(16, 4, 32)
+    out.backward(g.to(dev))
+    _cmp(xg.grad, g, 1e-6, msg="droppath dx")
+    _cmp(yg.grad, g, 1e-6, msg="droppath dy")
+
+
+def test_upsample2x_gpu(dev):
+    x32 = torch.randn(4, 8, 64)
+    xg = x32.to(dev).requires_grad_(True)
+    y = ops.upsample2x(xg)
+    import torch.nn.functional as F
+    y_ref = F.interpolate(x32, scale_factor=2, mode="nearest")
+    _cmp(y, y_ref, 1e-6, msg="up2 fwd")
+    g = torch.randn_like(y_ref)
+    y.backward(g.to(dev))
+    xc = x32.clone().requires_grad_(True)
+    F.interpolate(xc, scale_factor=2, mode="nearest").backward(g)
+    _cmp(xg.grad, xc.grad, 1e-6, msg="up2 bwd")

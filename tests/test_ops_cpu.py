@@ -153,3 +153,37 @@ def test_conv_transpose1d_matches_torch(Ci, Co, K, stride):
     assert torch.allclose(gx, x.grad, atol=1e-5)
     assert torch.allclose(gw, w.grad, atol=1e-4)
     assert torch.allclose(gb, b.grad, atol=1e-4)
+
+
+def test_droppath_add_eval_is_plain_add():
+    x = torch.randn(4, 3, 16)
+    y = torch.randn(4, 3, 16)
+    out = ops.droppath_add(x, y, 0.5, training=False)
+    assert torch.allclose(out, x + y)
+
+
+def test_droppath_add_train_masks_rows():
+    torch.manual_seed(0)
+    x = torch.zeros(64, 2, 8, requires_grad=True)
+    y = torch.ones(64, 2, 8)
+    out = ops.droppath_add(x, y, 0.5, training=True)
+    rows = out.flatten(1).sum(1)
+    keep_scale = 1.0 / 0.5
+    # each row is either fully dropped or fully kept (scaled)
+    assert set(rows.round(decimals=4).unique().tolist()) <= {0.0, 16.0 * keep_scale}
+    # grad of x passes through unchanged
+    out.sum().backward()
+    assert torch.allclose(x.grad, torch.ones_like(x))
+
+
+def test_upsample2x_matches_interpolate():
+    x = torch.randn(2, 3, 37, requires_grad=True)
+    y = ops.upsample2x(x)
+    y_ref = F.interpolate(x, scale_factor=2, mode="nearest")
+    assert torch.equal(y, y_ref)
+    g = torch.randn_like(y)
+    y.backward(g)
+    gx = x.grad.clone()
+    x.grad = None
+    y_ref.backward(g)
+    assert torch.allclose(gx, x.grad)
