@@ -55,15 +55,9 @@ __global__ void dwconv_fwd_kernel(const scalar_t* __restrict__ x,
   const long li0 = lo * stride - padl;
   float acc = HAS_BIAS ? (float)bias[co] : 0.0f;
   const scalar_t* xr = x + ((long)n * Ci + co) * L;  // depthwise: ci == co
-  if (li0 >= 0 && li0 + (long)(K - 1) * dil < L) {
-    for (int k = 0; k < K; ++k) {
-      acc += w_lds[k] * (float)xr[li0 + (long)k * dil];
-    }
-  } else {
-    for (int k = 0; k < K; ++k) {
-      const long li = li0 + (long)k * dil;
-      if (li >= 0 && li < L) acc += w_lds[k] * (float)xr[li];
-    }
+  for (int k = 0; k < K; ++k) {
+    const long li = li0 + (long)k * dil;
+    if (li >= 0 && li < L) acc += w_lds[k] * (float)xr[li];
   }
   y[((long)n * Co + co) * Lo + lo] = (scalar_t)acc;
 }
@@ -284,26 +278,16 @@ __global__ void conv1d_dw_kernel(const scalar_t* __restrict__ dy,
         const float dyv = dy_s[threadIdx.x];
         if (HAS_BIAS && blockIdx.y == 0) bacc += dyv;
         const long li0 = lo * stride - padl;
-        const bool interior =
-            (li0 >= 0) && (li0 + (long)(KT - 1) * dil < L);
 #pragma unroll
         for (int c = 0; c < kCigT; ++c) {
           if (c >= cig_n) break;
           const scalar_t* xr = xr0 + (long)c * L;
-          if (interior) {
 #pragma unroll
-            for (int k = 0; k < KT; ++k) {
-              if (k >= K) break;
-              acc[c][k] += dyv * (float)xr[li0 + (long)k * dil];
-            }
-          } else {
-#pragma unroll
-            for (int k = 0; k < KT; ++k) {
-              if (k >= K) break;
-              const long li = li0 + (long)k * dil;
-              if (li >= 0 && li < L) {
-                acc[c][k] += dyv * (float)xr[li];
-              }
+          for (int k = 0; k < KT; ++k) {
+            if (k >= K) break;
+            const long li = li0 + (long)k * dil;
+            if (li >= 0 && li < L) {
+              acc[c][k] += dyv * (float)xr[li];
             }
           }
         }
