@@ -58,6 +58,9 @@ def main():
                    choices=["bf16", "fp32"])
     p.add_argument("--graph", type=int, default=1,
                    help="capture the step in a hipGraph (1=try, 0=off)")
+    p.add_argument("--mode", type=str, default="train",
+                   choices=["train", "infer"],
+                   help="train = full step; infer = forward-only (serving)")
     args = p.parse_args()
 
     distributed = pdist.init_distributed_mode()
@@ -93,18 +96,25 @@ def main():
     if tgt_trans is not None:
         t = tgt_trans(t)
 
-    def step():
-        replica.zero_grad()
-        out = model(x)
-        if isinstance(out, (list, tuple)):
-            out = [o.float() for o in out]
-        else:
-            out = out.float()
-        loss = loss_fn(out, t)
-        loss.backward()
-        replica.allreduce()
-        optimizer.step()
-        return loss
+    if args.mode == "infer":
+        model = model.eval()
+
+        @torch.no_grad()
+        def step():
+            return model(x)
+    else:
+        def step():
+            replica.zero_grad()
+            out = model(x)
+            if isinstance(out, (list, tuple)):
+                out = [o.float() for o in out]
+            else:
+                out = out.float()
+            loss = loss_fn(out, t)
+            loss.backward()
+            replica.allreduce()
+            optimizer.step()
+            return loss
 
     # warmup (also materializes optimizer state + adam packing)
     for _ in range(args.warmup):
@@ -174,6 +184,7 @@ def main():
                 "global_batch": global_batch,
                 "seq_len": args.in_samples,
                 "parallelism": f"dp{world}",
+                "mode": args.mode,
                 "graph": graphed,
             },
         }), flush=True)
