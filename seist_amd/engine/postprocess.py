@@ -142,22 +142,45 @@ def _batch_candidate_peaks(batch: np.ndarray, mph: float):
 def _pick_phase(outputs: torch.Tensor, prob_threshold: float,
                 min_peak_dist: int, topk: int,
                 padding_value: int) -> torch.Tensor:
-    """Batch phase picking -> (N, topk) sample indices, padded."""
-    batch = outputs.detach().float().cpu().numpy()
-    N, L = batch.shape
-    cand = _batch_candidate_peaks(batch, prob_threshold)
-    # exclude first/last sample like detect_peaks
-    cand[:, 0] = False
-    cand[:, -1] = False
+    """Batch phase picking -> (N, topk) sample indices, padded.
+
+    On GPU the rising-edge candidate mask is computed on-device and only
+    the sparse (row, col, value) candidate lists cross D2H — the reference
+    copies every full probability trace to the host per step
+    (postprocess.py:181)."""
+    x = outputs.detach().float()
+    N, L = x.shape
+    if x.is_cuda:
+        dx = x[:, 1:] - x[:, :-1]
+        cand = torch.zeros_like(x, dtype=torch.bool)
+        cand[:, 1:-1] = (dx[:, 1:] <= 0) & (dx[:, :-1] > 0)
+        cand &= x >= prob_threshold
+        cand[:, 0] = False
+        cand[:, -1] = False
+        nz = cand.nonzero()
+        rows = nz[:, 0].cpu().numpy()
+        cols = nz[:, 1].cpu().numpy()
+        vals = x[cand].cpu().numpy()
+    else:
+        batch = x.numpy()
+        cand = _batch_candidate_peaks(batch, prob_threshold)
+        cand[:, 0] = False
+        cand[:, -1] = False
+        rows, cols = np.where(cand)
+        vals = batch[rows, cols]
     out = np.full((N, topk), padding_value, dtype=np.int64)
-    rows, cols = np.where(cand)
+    # candidate counts per row (nonzero is row-major sorted)
+    counts = np.bincount(rows, minlength=N)
+    offsets = np.concatenate([[0], np.cumsum(counts)])
     for i in range(N):
-        ind = cols[rows == i]
-        if ind.size == 0:
+        lo, hi = offsets[i], offsets[i + 1]
+        if hi <= lo:
             continue
+        ind = cols[lo:hi]
         if min_peak_dist > 1:
-            x = batch[i]
-            ind = ind[np.argsort(x[ind])][::-1][:topk]
+            v = vals[lo:hi]
+            order = np.argsort(v)[::-1][:topk]
+            ind = ind[order]
             idel = np.zeros(ind.size, dtype=bool)
             for j in range(ind.size):
                 if not idel[j]:
@@ -179,15 +202,27 @@ def _detect_event(outputs: torch.Tensor, prob_threshold: float,
     (the reference calls obspy's trigger_onset per trace in a Python loop,
     postprocess.py:129); only the tiny per-row top-k selection stays
     sequential."""
-    batch = outputs.detach().float().cpu().numpy()
-    N, L = batch.shape
-    above = batch > prob_threshold
-    d = np.diff(above.astype(np.int8), axis=1)
-    srow, scol = np.where(d == 1)
-    erow, ecol = np.where(d == -1)
-    # row-edge runs
-    first = np.where(above[:, 0])[0]
-    last = np.where(above[:, -1])[0]
+    x = outputs.detach().float()
+    N, L = x.shape
+    if x.is_cuda:
+        above_t = x > prob_threshold
+        d_t = above_t[:, 1:].char() - above_t[:, :-1].char()
+        s_nz = (d_t == 1).nonzero()
+        e_nz = (d_t == -1).nonzero()
+        srow = s_nz[:, 0].cpu().numpy()
+        scol = s_nz[:, 1].cpu().numpy()
+        erow = e_nz[:, 0].cpu().numpy()
+        ecol = e_nz[:, 1].cpu().numpy()
+        first = above_t[:, 0].nonzero().flatten().cpu().numpy()
+        last = above_t[:, -1].nonzero().flatten().cpu().numpy()
+    else:
+        batch = x.numpy()
+        above = batch > prob_threshold
+        d = np.diff(above.astype(np.int8), axis=1)
+        srow, scol = np.where(d == 1)
+        erow, ecol = np.where(d == -1)
+        first = np.where(above[:, 0])[0]
+        last = np.where(above[:, -1])[0]
     srow = np.concatenate([srow, first])
     scol = np.concatenate([scol + 1, np.zeros(len(first), dtype=scol.dtype)])
     erow = np.concatenate([erow, last])

@@ -143,3 +143,15 @@ def test_other_model_families_gpu_step(dev, name, C, L):
     loss.backward()
     opt.step()
     assert torch.isfinite(loss).item()
+
+
+def test_postprocess_gpu_matches_cpu(dev):
+    from seist_amd.engine.postprocess import _detect_event, _pick_phase
+    torch.manual_seed(2)
+    out = torch.rand(64, 2048)
+    a = _pick_phase(out.to(dev), 0.6, 50, 3, -7).cpu()
+    b = _pick_phase(out, 0.6, 50, 3, -7)
+    assert torch.equal(a, b)
+    a = _detect_event(out.to(dev), 0.8, 2).cpu()
+    b = _detect_event(out, 0.8, 2)
+    assert torch.equal(a, b)
