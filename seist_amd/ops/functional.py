@@ -353,8 +353,11 @@ def pooled_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     path is selected when available.
     """
     E = q.size(2)
-    if use_native(q) and hasattr(ext(), "pooled_attn_fwd") and not training:
-        return ext().pooled_attn_fwd(q, k, v)
+    if (use_native(q) and hasattr(ext(), "pooled_attn_fwd") and not training
+            and not torch.is_grad_enabled() and E in (8, 16, 32)
+            and k.size(3) <= 256):
+        return ext().pooled_attn_fwd(q.contiguous(), k.contiguous(),
+                                     v.contiguous())
     attn = torch.matmul(q.transpose(-1, -2), k) * (1.0 / math.sqrt(E))
     attn = attn.softmax(dim=-1)
     if attn_dropout > 0.0 and training:

@@ -41,6 +41,8 @@ at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
 at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len);
 at::Tensor upsample2x_fwd(const at::Tensor& x);
 at::Tensor upsample2x_bwd(const at::Tensor& dy);
+at::Tensor pooled_attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                           const at::Tensor& v);
 at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len);
 
 at::Tensor adam_pack(std::vector<at::Tensor> params,
@@ -76,6 +78,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("interp_linear_bwd", &interp_linear_bwd, "linear interp backward");
   m.def("upsample2x_fwd", &upsample2x_fwd, "nearest 2x upsample forward");
   m.def("upsample2x_bwd", &upsample2x_bwd, "nearest 2x upsample backward");
+  m.def("pooled_attn_fwd", &pooled_attn_fwd,
+        "fused pooled-KV attention forward (inference)");
   m.def("adam_pack", &adam_pack, "pack fused-adam chunk metadata");
   m.def("sum_batch", &sum_batch, "batch-axis sum to fp32");
   m.def("channel_sum", &channel_sum, "per-channel sum to fp32");

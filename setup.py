@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "reductions.hip"),
         os.path.join(HIP_DIR, "pw_mfma.hip"),
         os.path.join(HIP_DIR, "rowscale.hip"),
+        os.path.join(HIP_DIR, "attention.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -273,3 +273,23 @@ def test_upsample2x_gpu(dev):
     xc = x32.clone().requires_grad_(True)
     F.interpolate(xc, scale_factor=2, mode="nearest").backward(g)
     _cmp(xg.grad, xc.grad, 1e-6, msg="up2 bwd")
+
+
+@pytest.mark.parametrize("E,Lq,Lk,H", [(8, 1024, 128, 2), (16, 256, 128, 4),
+                                       (32, 128, 128, 3), (8, 100, 96, 1)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_pooled_attn_fused_inference(dev, E, Lq, Lk, H, dtype):
+    """Fused online-softmax attention vs the bmm+softmax composite."""
+    torch.manual_seed(6)
+    N = 3
+    q32 = torch.randn(N, H, E, Lq)
+    k32 = torch.randn(N, H, E, Lk)
+    v32 = torch.randn(N, H, E, Lk)
+    with torch.no_grad():
+        out = ops.pooled_attention(q32.to(dev, dtype), k32.to(dev, dtype),
+                                   v32.to(dev, dtype))
+        ref = ops.pooled_attention(q32.to(dtype).float(),
+                                   k32.to(dtype).float(),
+                                   v32.to(dtype).float())
+    atol = 1e-4 if dtype == torch.float32 else 3e-2
+    _cmp(out, ref, atol, msg="pooled attn")
