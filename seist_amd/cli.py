@@ -52,6 +52,9 @@ def get_args(argv=None):
     parser.add_argument("--trace-step-time", type=bool_, default=False,
                         help="log a data/h2d/fwd/loss/bwd/opt/comm/"
                              "postprocess/metrics step-time breakdown")
+    parser.add_argument("--train-metrics-interval", type=int, default=1,
+                        help="compute train metrics/postprocess every Nth "
+                             "step (1 = reference parity)")
     parser.add_argument("--sync-metrics-per-step", type=bool_, default=False,
                         help="reference-parity per-step metric collectives "
                              "(default: per-epoch)")

@@ -135,6 +135,14 @@ def train(args, tasks, model, optimizer, scheduler, loss_fn, train_loader,
         average_meters["loss"].update(loss.item(), step_batch_size)
         train_loss_per_step.append(loss.item())
 
+        interval = max(1, getattr(args, "train_metrics_interval", 1))
+        if step % interval != 0:
+            timer.step()
+            if step % args.log_step == 0 and pdist.is_main_process():
+                logger.info(progress.get_str(
+                    batch_idx=step, name=f"{args.model_name}_train"))
+            continue
+
         with timer.phase("postprocess"):
             outputs_for_metrics = (outs_trans_for_res(outputs)
                                    if outs_trans_for_res is not None
