@@ -19,23 +19,22 @@ def convert_to_bf16(model: nn.Module) -> nn.Module:
     """Cast conv/linear parameters to bf16; keep norms and recurrent cells
     in fp32 (their params are small, their numerics are touchy).
 
-    Models containing LSTM/LayerNorm interleaved with dense layers
-    (EQTransformer, MagNet) stay fp32 — matching the reference's numerical
-    regime — because MIOpen RNNs and mixed-dtype LayerNorm do not take
-    bf16 activations cleanly."""
-    from ..utils.logger import logger
-    if any(isinstance(m, (nn.LSTM, nn.GRU, nn.LayerNorm))
-           for m in model.modules()):
-        logger.warning(
-            "Model contains LSTM/LayerNorm; keeping fp32 numerics.")
-        return model
+    Models containing LSTM/LayerNorm (EQTransformer, MagNet) get the
+    SELECTIVE policy: only Conv1d/ConvTranspose1d go bf16 — the big
+    encoder/decoder tensors run on the bf16 matrix cores while the tiny
+    recurrent/attention stage (L=64) keeps the reference's fp32 numerics.
+    The conv path casts activations to the weight dtype and the models
+    cast back to fp32 at their LSTM/LayerNorm boundaries."""
+    selective = any(isinstance(m, (nn.LSTM, nn.GRU, nn.LayerNorm))
+                    for m in model.modules())
     for m in model.modules():
-        if isinstance(m, _BF16_TYPES):
+        if isinstance(m, (nn.Conv1d, nn.ConvTranspose1d)):
+            m.to(torch.bfloat16)
+        elif not selective and isinstance(m, nn.Linear):
             m.to(torch.bfloat16)
         elif isinstance(m, _FP32_TYPES):
             m.to(torch.float32)
-        else:
-            # bare nn.Parameter holders (e.g. EQT AttentionLayer weights)
+        elif not selective:
             for p in m.parameters(recurse=False):
                 p.data = p.data.to(torch.bfloat16)
     return model

@@ -80,7 +80,8 @@ class BiLSTMBlock(nn.Module):
         self.bn = nn.BatchNorm1d(out_channels)
 
     def forward(self, x):
-        x, _ = self.bilstm(x.permute(0, 2, 1))
+        # MIOpen LSTM keeps fp32 weights; cast bf16 encoder activations up
+        x, _ = self.bilstm(x.permute(0, 2, 1).float())
         x = self.dropout(x).permute(0, 2, 1)
         x = run_conv(self.conv, x)
         return run_bn(self.bn, x)
@@ -150,6 +151,7 @@ class TransformerLayer(nn.Module):
         self.ln1 = nn.LayerNorm(io_channels)
 
     def forward(self, x):
+        x = x.float()  # attention/LayerNorm stage stays fp32 (tiny, L=64)
         x1, w = self.attn(x)
         x2 = (x1 + x).permute(0, 2, 1)
         x2 = self.ln0(x2)
@@ -260,7 +262,7 @@ class Decoder(nn.Module):
                                   padding=5)
 
     def forward(self, x):
-        x = x.permute(0, 2, 1)
+        x = x.permute(0, 2, 1).float()  # decoder LSTM stays fp32
         x, _ = self.lstm(x)
         x = self.lstm_dropout(x).permute(0, 2, 1)
         x, _ = self.transformer(x)

@@ -44,7 +44,7 @@ class MagNet(nn.Module):
 
     def forward(self, x):
         x = self.conv_layers(x)
-        hs, (h, c) = self.lstm(x.transpose(-1, -2))
+        hs, (h, c) = self.lstm(x.transpose(-1, -2).float())
         h = h.transpose(0, 1).flatten(1)
         return self.lin(h)
 
