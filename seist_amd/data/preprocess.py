@@ -565,6 +565,28 @@ class SeismicDataset(Dataset):
             logger.info(f"Data augmentation: dataset size -> "
                         f"{self._dataset_size * 2}")
 
+        def _flat(names):
+            for n in names:
+                if isinstance(n, (tuple, list)):
+                    yield from _flat(n)
+                else:
+                    yield n
+
+        # Regression/classification labels ("value"/"onehot" io items) have
+        # no target on a noise-replaced window: the reference crashes at
+        # collate ([] vs [1]) unless the user remembers
+        # --generate-noise-rate 0. Force it off here, mirroring the
+        # reference's own p_position_ratio -> generate_noise_rate=0 rule
+        # (reference training/preprocess.py:125-131).
+        generate_noise_rate = args.generate_noise_rate
+        scalar_labels = [n for n in _flat(self._label_names)
+                         if Config.get_type(n) in ("value", "onehot")]
+        if scalar_labels and generate_noise_rate > 0:
+            logger.warning(
+                f"labels {scalar_labels} are undefined on generated noise; "
+                f"`generate_noise_rate` -> 0.0")
+            generate_noise_rate = 0.0
+
         self._preprocessor = DataPreprocessor(
             data_channels=self._dataset.channels(),
             sampling_rate=self._dataset.sampling_rate(),
@@ -581,7 +603,7 @@ class SeismicDataset(Dataset):
             pre_emphasis_rate=args.pre_emphasis_rate,
             pre_emphasis_ratio=args.pre_emphasis_ratio,
             max_event_num=args.max_event_num,
-            generate_noise_rate=args.generate_noise_rate,
+            generate_noise_rate=generate_noise_rate,
             shift_event_rate=args.shift_event_rate,
             mask_percent=args.mask_percent,
             noise_percent=args.noise_percent,

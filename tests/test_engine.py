@@ -42,6 +42,19 @@ def test_seist_regression_task_cpu(tmp_path):
     assert len(csvs) == 1
 
 
+def test_regression_task_with_noise_augmentation_cpu(tmp_path):
+    """Value-type labels + augmentation: generate_noise_rate must be forced
+    off (a noise-replaced window has no magnitude/baz/dis target — the
+    collate would see [] vs [1])."""
+    args = _args(tmp_path, ["--model-name", "seist_s_dis",
+                            "--augmentation", "true",
+                            "--generate-noise-rate", "0.9"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) == 1
+
+
 def test_resume_from_checkpoint(tmp_path):
     args = _args(tmp_path, ["--model-name", "phasenet", "--mode", "train"])
     args.distributed = False
