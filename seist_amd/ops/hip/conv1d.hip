@@ -28,6 +28,10 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
                long padl, long dilation, bool is_dx);
 at::Tensor channel_sum(const at::Tensor& in);
+c10::optional<at::Tensor> dw_mfma_try(const at::Tensor& dy,
+                                      const at::Tensor& x, long stride,
+                                      long padl, long groups, long dilation,
+                                      int K);
 
 namespace {
 
@@ -499,6 +503,13 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
       db = channel_sum(dy).to(w.scalar_type());
     }
     return {dw, db};
+  }
+
+  // grouped/depthwise stride-1 convs: matrix-core weight gradient
+  if (auto dwm = dw_mfma_try(dy, x, stride, padl, groups, dilation, K)) {
+    at::Tensor db;
+    if (has_bias) db = channel_sum(dy).to(w.scalar_type());
+    return {dwm->to(w.scalar_type()), db};
   }
 
   auto dw32 = at::zeros_like(w, w.options().dtype(at::kFloat));

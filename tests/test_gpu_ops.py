@@ -91,6 +91,51 @@ def test_conv1d_fwd_bwd(dev, groups, k, stride, dil, Ci, Co):
     _cmp(wg.grad, wc.grad, 1e-3, 1e-3, msg="conv dw")
 
 
+@pytest.mark.parametrize("groups,k,C,L,stride", [
+    (16, 13, 16, 4096, 1),  # depthwise stage conv (MFMA dw diagonal tile)
+    (8, 9, 8, 1000, 1),     # depthwise, C < 16 (padded tile) + odd-L tail
+    (2, 7, 32, 1024, 1),    # groups=2 stage conv (Cog=16)
+    (2, 5, 16, 520, 1),     # groups=2, Cog=8 (two groups inside one tile)
+    (4, 3, 16, 256, 1),     # Cog=4
+    (3, 19, 3, 8192, 2),    # stride-2 stem depthwise, K=19 (KT=24 path)
+    (16, 11, 16, 4096, 2),  # stride-2 stage depthwise
+    (16, 7, 16, 2048, 2),   # stride-2, K<=8 path
+])
+def test_conv1d_grouped_bf16(dev, groups, k, C, L, stride):
+    """bf16 grouped/depthwise convs: the dw path runs the MFMA diagonal-tile
+    kernel (ops/hip/dw_mfma.hip). Reference = fp32 CPU on the same
+    bf16-quantized inputs."""
+    torch.manual_seed(3)
+    N = 3
+    padl, padr = (k - 1) // 2, (k - 1) - (k - 1) // 2
+    if stride > 1:
+        from seist_amd.ops.functional import auto_pad_lr
+        padl, padr = auto_pad_lr(L, k, stride)
+    x32 = torch.randn(N, C, L).to(torch.bfloat16).float()
+    w32 = (torch.randn(C, C // groups, k) * 0.2).to(torch.bfloat16).float()
+    b32 = (torch.randn(C) * 0.1).to(torch.bfloat16).float()
+
+    xg = x32.to(dev, torch.bfloat16).requires_grad_(True)
+    wg = w32.to(dev, torch.bfloat16).requires_grad_(True)
+    bg = b32.to(dev, torch.bfloat16).requires_grad_(True)
+    y = ops.conv1d(xg, wg, bg, stride=stride, padding=(padl, padr),
+                   groups=groups)
+
+    xc = x32.clone().requires_grad_(True)
+    wc = w32.clone().requires_grad_(True)
+    bc = b32.clone().requires_grad_(True)
+    y_ref = ops.conv1d(xc, wc, bc, stride=stride, padding=(padl, padr),
+                       groups=groups)
+    _cmp(y, y_ref, 5e-2, 1e-2, msg="grouped bf16 fwd")
+
+    g32 = torch.randn_like(y_ref).to(torch.bfloat16).float()
+    y.backward(g32.to(dev, torch.bfloat16))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, 1e-1, 1e-2, msg="grouped bf16 dx")
+    _cmp(wg.grad, wc.grad, 0.0, 3e-2, msg="grouped bf16 dw")
+    _cmp(bg.grad, bc.grad, 0.0, 3e-2, msg="grouped bf16 db")
+
+
 def test_conv1d_with_bias(dev):
     x32 = torch.randn(2, 8, 256)
     w32 = torch.randn(4, 8, 7) * 0.2
