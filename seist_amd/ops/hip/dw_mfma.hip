@@ -45,19 +45,35 @@ constexpr int x_elems(int KT, int S) {
   return (8 + (KT - 1) + S * 63 + 1 + 7) & ~7;
 }
 
+// tile -> (row ch0, col ch0): tp == 0 is the diagonal mode (Cog <= 16, the
+// group mask lives inside the tile); tp >= 1 enumerates the (tp x tp) tile
+// pairs of each group's diagonal block (Cog = 16*tp > 16).
+__device__ __forceinline__ void tile_channels(int bx, int tp, int Cog,
+                                              int* ch0r, int* ch0c) {
+  if (tp == 0) {
+    *ch0r = *ch0c = bx * 16;
+  } else {
+    const int g = bx / (tp * tp);
+    const int rem = bx - g * (tp * tp);
+    *ch0r = g * Cog + (rem / tp) * 16;
+    *ch0c = g * Cog + (rem % tp) * 16;
+  }
+}
+
 template <int KT, int S>
 __global__ __launch_bounds__(256)
 void dw_mfma_kernel(const sa_bf16* __restrict__ dy,
                     const sa_bf16* __restrict__ x,
                     float* __restrict__ partial,  // [gridx*zsplit][KT*256]
                     int N, int C, long L, long Lo, int K, int padl,
-                    long nwork, int zsplit) {
+                    long nwork, int zsplit, int tp, int Cog) {
   constexpr int kXElems = x_elems(KT, S);
   constexpr int kXPitch = kXElems + 3;
   __shared__ sa_bf16 x_s[16 * kXPitch];
   __shared__ float red[4][256];
 
-  const int ch0 = blockIdx.x * 16;
+  int ch0, ch0c;
+  tile_channels(blockIdx.x, tp, Cog, &ch0, &ch0c);
   const long per = (nwork + zsplit - 1) / zsplit;
   const long w0 = (long)blockIdx.y * per;
   const long w1 = min(nwork, w0 + per);
@@ -88,7 +104,7 @@ void dw_mfma_kernel(const sa_bf16* __restrict__ dy,
       const int r = idx / (kXElems / 8);
       const int c8 = idx - r * (kXElems / 8);
       const long g = s0 + c8 * 8;
-      const int cg = ch0 + r;
+      const int cg = ch0c + r;
       bf16x8 v = {};
       if (cg < C) {
         const sa_bf16* xr = x + ((long)n * C + cg) * L;
@@ -169,24 +185,28 @@ constexpr int kZChunk = 128;
 
 template <int KT>
 __global__ void dw_mfma_reduce_kernel(const float* __restrict__ partial,
-                                      float* __restrict__ dw,
-                                      int C, int K, int Cog, int zsplit) {
+                                      float* __restrict__ dw, int C, int K,
+                                      int Cog, int zsplit, int tp) {
   const int tile = blockIdx.x;
   const int k = blockIdx.y;  // < K
   const long z0 = (long)blockIdx.z * kZChunk;
   const long z1 = min((long)zsplit, z0 + kZChunk);
   const int t = threadIdx.x;
   const int row = t >> 4, col = t & 15;
-  const int co = tile * 16 + row;
-  const int ci = tile * 16 + col;
-  // valid pair: same group (ch0 is a multiple of Cog since Cog divides 16)
-  const bool ok = co < C && ci < C && (row / Cog) == (col / Cog);
+  int ch0r, ch0c;
+  tile_channels(tile, tp, Cog, &ch0r, &ch0c);
+  const int co = ch0r + row;
+  const int ci = ch0c + col;
+  // valid pair: same group (in diagonal mode ch0 is a multiple of Cog
+  // because Cog divides 16; in tile-pair mode the pairs are in-group by
+  // construction and this check is always true)
+  const bool ok = co < C && ci < C && (co / Cog) == (ci / Cog);
   if (!ok) return;
   const float* base =
       partial + (long)tile * zsplit * (KT * 256) + k * 256 + t;
   float v = 0.f;
   for (long z = z0; z < z1; ++z) v += base[z * (KT * 256)];
-  atomicAdd(&dw[((long)co * Cog + (col % Cog)) * K + k], v);
+  atomicAdd(&dw[((long)co * Cog + (ci % Cog)) * K + k], v);
 }
 
 }  // namespace
@@ -206,11 +226,13 @@ c10::optional<at::Tensor> dw_mfma_try(const at::Tensor& dy,
   if (x.scalar_type() != at::kBFloat16 || dy.scalar_type() != at::kBFloat16)
     return c10::nullopt;
   if ((stride != 1 && stride != 2) || dilation != 1 || Ci != Co ||
-      Cog != Cig || Cog > 16 || 16 % Cog != 0 || K > 24 || K < 1)
+      Cog != Cig || K > 24 || K < 1)
     return c10::nullopt;
+  if (Cog > 16 ? (Cog % 16 != 0) : (16 % Cog != 0)) return c10::nullopt;
 
   const int C = Ci;
-  const int gridx = (C + 15) / 16;
+  const int tp = (Cog > 16) ? Cog / 16 : 0;  // 0 = diagonal-tile mode
+  const int gridx = tp ? (int)groups * tp * tp : (C + 15) / 16;
   const long lochunks = (Lo + 63) / 64;
   const long nwork = (long)N * lochunks;
   // enough blocks to fill the chip on big calls, but never more splits than
@@ -232,12 +254,12 @@ c10::optional<at::Tensor> dw_mfma_try(const at::Tensor& dy,
         (dw_mfma_kernel<decltype(kt)::value, decltype(st)::value>), grid,
         dim3(256), 0, stream.stream(), (const sa_bf16*)dy.data_ptr(),
         (const sa_bf16*)x.data_ptr(), partial.data_ptr<float>(), N, C, L, Lo,
-        K, (int)padl, nwork, zsplit);
+        K, (int)padl, nwork, zsplit, tp, Cog);
     dim3 rgrid(gridx, K, (zsplit + kZChunk - 1) / kZChunk);
     hipLaunchKernelGGL((dw_mfma_reduce_kernel<decltype(kt)::value>), rgrid,
                        dim3(256), 0, stream.stream(),
                        partial.data_ptr<float>(), dw.data_ptr<float>(), C, K,
-                       Cog, zsplit);
+                       Cog, zsplit, tp);
   };
   auto launch_s = [&](auto kt) {
     if (stride == 1) launch(kt, std::integral_constant<int, 1>{});

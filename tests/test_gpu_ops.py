@@ -100,6 +100,8 @@ def test_conv1d_fwd_bwd(dev, groups, k, stride, dil, Ci, Co):
     (3, 19, 3, 8192, 2),    # stride-2 stem depthwise, K=19 (KT=24 path)
     (16, 11, 16, 4096, 2),  # stride-2 stage depthwise
     (16, 7, 16, 2048, 2),   # stride-2, K<=8 path
+    (2, 5, 64, 512, 1),     # Cog=32 > 16: block-diagonal tile pairs
+    (2, 7, 96, 300, 1),     # Cog=48, odd L
 ])
 def test_conv1d_grouped_bf16(dev, groups, k, C, L, stride):
     """bf16 grouped/depthwise convs: the dw path runs the MFMA diagonal-tile
