@@ -34,14 +34,26 @@ __global__ void sum_batch_kernel(const scalar_t* __restrict__ in,
   part[(long)blockIdx.y * M + j] = s;
 }
 
+// 64 columns x 4 split-slices per block: M is often small (Co*Ci of one
+// conv), so column-only parallelism leaves the chip idle — the 4 z-slices
+// quadruple the block count and the per-thread serial depth drops to
+// nsplit/4.
 __global__ void sum_batch_final_kernel(const float* __restrict__ part,
                                        float* __restrict__ out,
                                        int nsplit, long M) {
-  const long j = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (j >= M) return;
+  __shared__ float red[4][64];
+  const int jt = threadIdx.x & 63;
+  const int zs = threadIdx.x >> 6;
+  const long j = (long)blockIdx.x * 64 + jt;
   float s = 0.0f;
-  for (int i = 0; i < nsplit; ++i) s += part[(long)i * M + j];
-  out[j] = s;
+  if (j < M) {
+    for (int i = zs; i < nsplit; i += 4) s += part[(long)i * M + j];
+  }
+  red[zs][jt] = s;
+  __syncthreads();
+  if (threadIdx.x < 64 && j < M) {
+    out[j] = red[0][jt] + red[1][jt] + red[2][jt] + red[3][jt];
+  }
 }
 
 template <typename scalar_t>
@@ -125,7 +137,7 @@ at::Tensor sum_batch(const at::Tensor& in) {
   // target ~512 blocks total, nsplit bounded so the final pass stays tiny
   const long col_blocks = std::max<long>(M / kBlock, 1);
   const int nsplit = std::max(1, (int)std::min<long>(
-      std::min<long>(B, 64), 512 / col_blocks));
+      std::min<long>(B, 128), 1024 / col_blocks));
   auto part = at::empty({nsplit, M}, in.options().dtype(at::kFloat));
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, in.scalar_type(),
@@ -137,7 +149,7 @@ at::Tensor sum_batch(const at::Tensor& in) {
                            part.data_ptr<float>(), B, M);
       });
   hipLaunchKernelGGL(sum_batch_final_kernel,
-                     dim3(sa::ceil_div(M, kBlock)), dim3(kBlock), 0,
+                     dim3(sa::ceil_div(M, 64)), dim3(kBlock), 0,
                      stream.stream(), part.data_ptr<float>(),
                      out.data_ptr<float>(), nsplit, M);
   return out;
