@@ -32,6 +32,9 @@ c10::optional<at::Tensor> dw_mfma_try(const at::Tensor& dy,
                                       const at::Tensor& x, long stride,
                                       long padl, long groups, long dilation,
                                       int K);
+bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
+                   const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                   long padl, long dilation, long groups, bool is_dx);
 
 namespace {
 
@@ -370,7 +373,14 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
   const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
 
-  // dense stride-1 convs go to the matrix cores
+  // stride-1 convs (dense AND grouped) go to the matrix cores: the
+  // tap-gather kernel stages each channel row once; the im2col kernel
+  // remains as the fp32 fallback
+  if (stride == 1
+      && conv_tap_mfma(x, w, bias, y, padl, dilation, groups,
+                       /*is_dx=*/false)) {
+    return y;
+  }
   if (groups == 1 && stride == 1
       && conv_mfma(x, w, bias, y, padl, dilation, /*is_dx=*/false)) {
     return y;
@@ -438,6 +448,11 @@ void conv1d_dx_into(const at::Tensor& dy, const at::Tensor& w,
   const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
 
+  if (stride == 1
+      && conv_tap_mfma(dy, w, c10::nullopt, dx, padl, dilation, groups,
+                       /*is_dx=*/true)) {
+    return;
+  }
   if (groups == 1 && stride == 1
       && conv_mfma(dy, w, c10::nullopt, dx, padl, dilation, /*is_dx=*/true)) {
     return;

@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "reductions.hip"),
         os.path.join(HIP_DIR, "pw_mfma.hip"),
         os.path.join(HIP_DIR, "dw_mfma.hip"),
+        os.path.join(HIP_DIR, "conv_tap.hip"),
         os.path.join(HIP_DIR, "rowscale.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
     ],

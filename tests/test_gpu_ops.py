@@ -91,25 +91,30 @@ def test_conv1d_fwd_bwd(dev, groups, k, stride, dil, Ci, Co):
     _cmp(wg.grad, wc.grad, 1e-3, 1e-3, msg="conv dw")
 
 
-@pytest.mark.parametrize("groups,k,C,L,stride", [
-    (16, 13, 16, 4096, 1),  # depthwise stage conv (MFMA dw diagonal tile)
-    (8, 9, 8, 1000, 1),     # depthwise, C < 16 (padded tile) + odd-L tail
-    (2, 7, 32, 1024, 1),    # groups=2 stage conv (Cog=16)
-    (2, 5, 16, 520, 1),     # groups=2, Cog=8 (two groups inside one tile)
-    (4, 3, 16, 256, 1),     # Cog=4
-    (3, 19, 3, 8192, 2),    # stride-2 stem depthwise, K=19 (KT=24 path)
-    (16, 11, 16, 4096, 2),  # stride-2 stage depthwise
-    (16, 7, 16, 2048, 2),   # stride-2, K<=8 path
-    (2, 5, 64, 512, 1),     # Cog=32 > 16: block-diagonal tile pairs
-    (2, 7, 96, 300, 1),     # Cog=48, odd L
+@pytest.mark.parametrize("groups,k,C,L,stride,dil", [
+    (16, 13, 16, 4096, 1, 1),  # depthwise stage conv (MFMA dw diag tile)
+    (8, 9, 8, 1000, 1, 1),     # depthwise, C < 16 (padded tile), odd L
+    (2, 7, 32, 1024, 1, 1),    # groups=2 stage conv (Cog=16)
+    (2, 5, 16, 520, 1, 1),     # groups=2, Cog=8 (two groups in one tile)
+    (4, 3, 16, 256, 1, 1),     # Cog=4
+    (3, 19, 3, 8192, 2, 1),    # stride-2 stem depthwise, K=19 (KT=24)
+    (16, 11, 16, 4096, 2, 1),  # stride-2 stage depthwise
+    (16, 7, 16, 2048, 2, 1),   # stride-2, K<=8 path
+    (2, 5, 64, 512, 1, 1),     # Cog=32 > 16: block-diagonal tile pairs
+    (2, 7, 96, 300, 1, 1),     # Cog=48, odd L
+    (1, 7, 32, 1000, 1, 1),    # dense bf16: tap-gather MFMA fwd/dx
+    (1, 11, 48, 777, 1, 1),    # dense, odd length
+    (1, 6, 20, 512, 1, 64),    # dilated causal shape (dist-PT), bf16
 ])
-def test_conv1d_grouped_bf16(dev, groups, k, C, L, stride):
-    """bf16 grouped/depthwise convs: the dw path runs the MFMA diagonal-tile
-    kernel (ops/hip/dw_mfma.hip). Reference = fp32 CPU on the same
+def test_conv1d_grouped_bf16(dev, groups, k, C, L, stride, dil):
+    """bf16 convs through the MFMA paths: tap-gather fwd/dx
+    (ops/hip/conv_tap.hip) and the diagonal-tile weight gradient
+    (ops/hip/dw_mfma.hip). Reference = fp32 CPU on the same
     bf16-quantized inputs."""
     torch.manual_seed(3)
     N = 3
-    padl, padr = (k - 1) // 2, (k - 1) - (k - 1) // 2
+    padl = (k - 1) * dil // 2
+    padr = (k - 1) * dil - padl
     if stride > 1:
         from seist_amd.ops.functional import auto_pad_lr
         padl, padr = auto_pad_lr(L, k, stride)
@@ -121,13 +126,13 @@ def test_conv1d_grouped_bf16(dev, groups, k, C, L, stride):
     wg = w32.to(dev, torch.bfloat16).requires_grad_(True)
     bg = b32.to(dev, torch.bfloat16).requires_grad_(True)
     y = ops.conv1d(xg, wg, bg, stride=stride, padding=(padl, padr),
-                   groups=groups)
+                   groups=groups, dilation=dil)
 
     xc = x32.clone().requires_grad_(True)
     wc = w32.clone().requires_grad_(True)
     bc = b32.clone().requires_grad_(True)
     y_ref = ops.conv1d(xc, wc, bc, stride=stride, padding=(padl, padr),
-                       groups=groups)
+                       groups=groups, dilation=dil)
     _cmp(y, y_ref, 5e-2, 1e-2, msg="grouped bf16 fwd")
 
     g32 = torch.randn_like(y_ref).to(torch.bfloat16).float()
