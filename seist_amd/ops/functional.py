@@ -340,6 +340,28 @@ def interp_linear(x: torch.Tensor, out_len: int) -> torch.Tensor:
 # ---------------------------------------------------------------------------
 
 
+class _PooledAttnTrain(torch.autograd.Function):
+    """Fused training attention: forward saves the per-query softmax stats
+    and a bit-packed dropout mask (32x smaller than the (Lq x Lk)
+    probability tensor the composite materializes); backward is a
+    flash-style recompute split into a per-query kernel (dQ + correction
+    term) and a per-key kernel (dK, dV)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, p):
+        out, stats, mask = ext().pooled_attn_train_fwd(q, k, v, p)
+        ctx.save_for_backward(q, k, v, out, stats, mask)
+        ctx.p = p
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, stats, mask = ctx.saved_tensors
+        dq, dk, dv = ext().pooled_attn_bwd(q, k, v, out, dout.contiguous(),
+                                           stats, mask, ctx.p)
+        return dq, dk, dv, None
+
+
 def pooled_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                      attn_dropout: float = 0.0,
                      training: bool = False) -> torch.Tensor:
@@ -349,15 +371,19 @@ def pooled_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     the aggregated sequence (Lk == Lq / aggr_ratio; 128 at every stage for
     the published configs). Returns (N, H, E, Lq).
 
-    Runs as batched GEMM + fused softmax; a fully fused single-kernel HIP
-    path is selected when available.
+    A fully fused single-kernel HIP path serves inference; training runs
+    the fused stats+mask forward with a flash-style backward. The batched
+    GEMM + softmax composite remains the CPU/fallback path.
     """
     E = q.size(2)
-    if (use_native(q) and hasattr(ext(), "pooled_attn_fwd") and not training
-            and not torch.is_grad_enabled() and E in (8, 16, 32)
-            and k.size(3) <= 256):
+    fusable = (use_native(q) and E in (8, 16, 32) and k.size(3) <= 256
+               and E * k.size(3) <= 4096)
+    if fusable and not training and not torch.is_grad_enabled():
         return ext().pooled_attn_fwd(q.contiguous(), k.contiguous(),
                                      v.contiguous())
+    if fusable and training and hasattr(ext(), "pooled_attn_train_fwd"):
+        return _PooledAttnTrain.apply(q.contiguous(), k.contiguous(),
+                                      v.contiguous(), float(attn_dropout))
     attn = torch.matmul(q.transpose(-1, -2), k) * (1.0 / math.sqrt(E))
     attn = attn.softmax(dim=-1)
     if attn_dropout > 0.0 and training:

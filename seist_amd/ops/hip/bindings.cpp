@@ -41,6 +41,16 @@ at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
 at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len);
 at::Tensor upsample2x_fwd(const at::Tensor& x);
 at::Tensor upsample2x_bwd(const at::Tensor& dy);
+std::vector<at::Tensor> pooled_attn_train_fwd(const at::Tensor& q,
+                                              const at::Tensor& k,
+                                              const at::Tensor& v, double p);
+std::vector<at::Tensor> pooled_attn_bwd(const at::Tensor& q,
+                                        const at::Tensor& k,
+                                        const at::Tensor& v,
+                                        const at::Tensor& out,
+                                        const at::Tensor& dout,
+                                        const at::Tensor& stats,
+                                        const at::Tensor& mask, double p);
 at::Tensor pooled_attn_fwd(const at::Tensor& q, const at::Tensor& k,
                            const at::Tensor& v);
 at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len);
@@ -80,6 +90,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("upsample2x_bwd", &upsample2x_bwd, "nearest 2x upsample backward");
   m.def("pooled_attn_fwd", &pooled_attn_fwd,
         "fused pooled-KV attention forward (inference)");
+  m.def("pooled_attn_train_fwd", &pooled_attn_train_fwd,
+        "fused pooled-KV attention training forward (stats + packed mask)");
+  m.def("pooled_attn_bwd", &pooled_attn_bwd,
+        "fused pooled-KV attention backward (dq, dk, dv)");
   m.def("adam_pack", &adam_pack, "pack fused-adam chunk metadata");
   m.def("sum_batch", &sum_batch, "batch-axis sum to fp32");
   m.def("channel_sum", &channel_sum, "per-channel sum to fp32");

@@ -345,3 +345,87 @@ def test_pooled_attn_fused_inference(dev, E, Lq, Lk, H, dtype):
                                    v32.to(dtype).float())
     atol = 1e-4 if dtype == torch.float32 else 3e-2
     _cmp(out, ref, atol, msg="pooled attn")
+
+
+@pytest.mark.parametrize("dtype,atol", [(torch.float32, 1e-4),
+                                        (torch.bfloat16, 3e-2)])
+@pytest.mark.parametrize("E,Lq,Lk", [(8, 1024, 128), (16, 256, 128),
+                                     (32, 128, 128), (8, 1000, 96)])
+def test_pooled_attention_train_fused(dev, dtype, atol, E, Lq, Lk):
+    """Fused training attention (ops/hip/attention.hip): p=0 must match the
+    bmm+softmax composite exactly; gradients compared against autograd of
+    the composite."""
+    import math
+    torch.manual_seed(5)
+    N, H = 2, 3
+    q32 = torch.randn(N, H, E, Lq).to(dtype).float()
+    k32 = torch.randn(N, H, E, Lk).to(dtype).float()
+    v32 = torch.randn(N, H, E, Lk).to(dtype).float()
+
+    from seist_amd.ops import functional as Fn
+    qg = q32.to(dev, dtype).requires_grad_(True)
+    kg = k32.to(dev, dtype).requires_grad_(True)
+    vg = v32.to(dev, dtype).requires_grad_(True)
+    out = Fn.pooled_attention(qg, kg, vg, attn_dropout=0.0, training=True)
+
+    qc = q32.clone().requires_grad_(True)
+    kc = k32.clone().requires_grad_(True)
+    vc = v32.clone().requires_grad_(True)
+    attn = torch.matmul(qc.transpose(-1, -2), kc) * (1.0 / math.sqrt(E))
+    ref = torch.matmul(attn.softmax(-1), vc.transpose(-1, -2)).transpose(-1, -2)
+    _cmp(out, ref, atol, 1e-3, msg="attn train fwd p=0")
+
+    g32 = torch.randn_like(ref).to(dtype).float()
+    out.backward(g32.to(dev, dtype))
+    ref.backward(g32)
+    _cmp(qg.grad, qc.grad, atol * 4, 1e-2, msg="attn dq")
+    _cmp(kg.grad, kc.grad, atol * 8, 1e-2, msg="attn dk")
+    _cmp(vg.grad, vc.grad, atol * 8, 1e-2, msg="attn dv")
+
+
+def test_pooled_attention_dropout_mask(dev):
+    """p>0: the fused path must equal the composite evaluated with the
+    very mask the kernel drew (saved bit-packed), forward and backward."""
+    import math
+    from seist_amd import ops as ops_pkg
+    from seist_amd.ops import ext
+    torch.manual_seed(6)
+    N, H, E, Lq, Lk, p = 2, 2, 16, 512, 128, 0.2
+    q = torch.randn(N, H, E, Lq, device=dev)
+    k = torch.randn(N, H, E, Lk, device=dev)
+    v = torch.randn(N, H, E, Lk, device=dev)
+    out, stats, mask = ext().pooled_attn_train_fwd(q, k, v, p)
+    # unpack mask bits -> (N, H, Lq, Lk) float
+    W = (Lk + 31) // 32
+    words = mask.view(N, H, Lq, W).unsqueeze(-1)          # int32
+    shifts = torch.arange(32, device=dev).view(1, 1, 1, 1, 32)
+    bits = ((words >> shifts) & 1).reshape(N, H, Lq, W * 32)[..., :Lk]
+    keep = bits.float()
+    assert 0.6 < keep.mean().item() < 0.95, "dropout rate off"
+
+    qc = q.clone().requires_grad_(True)
+    kc = k.clone().requires_grad_(True)
+    vc = v.clone().requires_grad_(True)
+    attn = torch.matmul(qc.transpose(-1, -2), kc) * (1.0 / math.sqrt(E))
+    attn = attn.softmax(-1) * keep / (1.0 - p)
+    ref = torch.matmul(attn, vc.transpose(-1, -2)).transpose(-1, -2)
+    _cmp(out, ref, 1e-4, 1e-4, msg="attn dropout fwd")
+
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    dq, dk, dv = ext().pooled_attn_bwd(q, k, v, out, g.contiguous(),
+                                       stats, mask, p)
+    _cmp(dq, qc.grad, 1e-3, 1e-3, msg="attn dropout dq")
+    _cmp(dk, kc.grad, 1e-3, 1e-3, msg="attn dropout dk")
+    _cmp(dv, vc.grad, 1e-3, 1e-3, msg="attn dropout dv")
+
+
+def test_pooled_attention_mask_varies(dev):
+    """the device-side seed must advance between calls (fresh masks)."""
+    from seist_amd.ops import ext
+    q = torch.randn(1, 1, 8, 256, device=dev)
+    k = torch.randn(1, 1, 8, 128, device=dev)
+    v = torch.randn(1, 1, 8, 128, device=dev)
+    _, _, m1 = ext().pooled_attn_train_fwd(q, k, v, 0.3)
+    _, _, m2 = ext().pooled_attn_train_fwd(q, k, v, 0.3)
+    assert not torch.equal(m1, m2), "dropout mask frozen across calls"
