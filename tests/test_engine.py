@@ -77,6 +77,25 @@ def test_early_stopping(tmp_path):
     main_worker(args, torch.device("cpu"))
 
 
+def test_ditingmotion_train_test_cpu(tmp_path):
+    # 2-channel [z, dz] input, dual clarity+polarity heads, onehot labels
+    args = _args(tmp_path, ["--model-name", "ditingmotion",
+                            "--in-samples", "256"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) == 1
+
+
+def test_magnet_train_test_cpu(tmp_path):
+    # conv+BiLSTM regression with MousaviLoss (mag, log-var) head
+    args = _args(tmp_path, ["--model-name", "magnet"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) == 1
+
+
 def test_scaled_activation_heads_cpu(tmp_path):
     # baz head uses cos/sin transforms end to end
     args = _args(tmp_path, ["--model-name", "baz_network", "--mode",
