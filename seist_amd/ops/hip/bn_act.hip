@@ -48,6 +48,9 @@ __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
   }
 }
 
+// one wave per channel: the per-channel partial reduction runs across the
+// 64 lanes instead of as one serial dependent-load loop (the thread-per-
+// channel version was latency-bound at ~12 us for KB-scale reads)
 __global__ void bn_finalize_kernel(const float* __restrict__ part,
                                    int nsplit,
                                    float* __restrict__ mean,
@@ -56,13 +59,18 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
                                    float* __restrict__ running_var,
                                    int C, long NL, float momentum,
                                    float eps) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & (sa::kWave - 1);
+  const int c = blockIdx.x * (blockDim.x / sa::kWave)
+                + threadIdx.x / sa::kWave;
   if (c >= C) return;
   float s = 0.0f, s2 = 0.0f;
-  for (int j = 0; j < nsplit; ++j) {
+  for (int j = lane; j < nsplit; j += sa::kWave) {
     s += part[((long)c * nsplit + j) * 2 + 0];
     s2 += part[((long)c * nsplit + j) * 2 + 1];
   }
+  s = sa::warp_reduce_sum(s);
+  s2 = sa::warp_reduce_sum(s2);
+  if (lane != 0) return;
   const float m = s / NL;
   const float var = fmaxf(s2 / NL - m * m, 0.0f);
   mean[c] = m;
@@ -130,17 +138,22 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
   }
 }
 
-// reduce (C, nsplit, 2) partials -> (C, 2)
+// reduce (C, nsplit, 2) partials -> (C, 2); one wave per channel
 __global__ void bn_part_reduce_kernel(const float* __restrict__ part,
                                       float* __restrict__ out,
                                       int C, int nsplit) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & (sa::kWave - 1);
+  const int c = blockIdx.x * (blockDim.x / sa::kWave)
+                + threadIdx.x / sa::kWave;
   if (c >= C) return;
   float s1 = 0.0f, s2 = 0.0f;
-  for (int j = 0; j < nsplit; ++j) {
+  for (int j = lane; j < nsplit; j += sa::kWave) {
     s1 += part[((long)c * nsplit + j) * 2 + 0];
     s2 += part[((long)c * nsplit + j) * 2 + 1];
   }
+  s1 = sa::warp_reduce_sum(s1);
+  s2 = sa::warp_reduce_sum(s2);
+  if (lane != 0) return;
   out[c * 2 + 0] = s1;
   out[c * 2 + 1] = s2;
 }
@@ -215,7 +228,7 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
                              stream.stream(), x.data_ptr<scalar_t>(),
                              part.data_ptr<float>(), C, NL, L);
         });
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 256)),
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)),
                        dim3(256), 0, stream.stream(),
                        part.data_ptr<float>(), nsplit, mean.data_ptr<float>(),
                        invstd.data_ptr<float>(),
@@ -272,7 +285,7 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
                            b32.data_ptr<float>(), part.data_ptr<float>(),
                            C, NL, L, (int)act);
       });
-  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 256)),
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
                      dim3(256), 0, stream.stream(), part.data_ptr<float>(),
                      sums.data_ptr<float>(), C, nsplit);
 

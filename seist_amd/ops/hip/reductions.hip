@@ -76,13 +76,19 @@ __global__ void channel_sum_kernel(const scalar_t* __restrict__ in,
   if (threadIdx.x == 0) part[(long)c * nsplit + split] = s;
 }
 
+// one wave per channel (a thread-per-channel serial loop is latency-bound)
 __global__ void part_sum_kernel(const float* __restrict__ part,
                                 float* __restrict__ out, int C, int nsplit) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & (sa::kWave - 1);
+  const int c = blockIdx.x * (blockDim.x / sa::kWave)
+                + threadIdx.x / sa::kWave;
   if (c >= C) return;
   float s = 0.0f;
-  for (int j = 0; j < nsplit; ++j) s += part[(long)c * nsplit + j];
-  out[c] = s;
+  for (int j = lane; j < nsplit; j += sa::kWave) {
+    s += part[(long)c * nsplit + j];
+  }
+  s = sa::warp_reduce_sum(s);
+  if (lane == 0) out[c] = s;
 }
 
 template <typename scalar_t>
@@ -173,7 +179,7 @@ at::Tensor channel_sum(const at::Tensor& in) {
                            in.data_ptr<scalar_t>(), part.data_ptr<float>(),
                            C, N, L);
       });
-  hipLaunchKernelGGL(part_sum_kernel, dim3(sa::ceil_div(C, 256)), dim3(256),
+  hipLaunchKernelGGL(part_sum_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
                      0, stream.stream(), part.data_ptr<float>(),
                      out.data_ptr<float>(), C, nsplit);
   return out;
