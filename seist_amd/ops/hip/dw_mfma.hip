@@ -181,7 +181,7 @@ void dw_mfma_kernel(const sa_bf16* __restrict__ dy,
 // Parallel over (tile, tap, z-chunk): each block sums <= kZChunk partial
 // rows (coalesced: lane t strides the z axis) and lands one atomicAdd per
 // valid (co, ci) pair — at most 8 adds per output value.
-constexpr int kZChunk = 128;
+constexpr int kZChunk = 32;
 
 template <int KT>
 __global__ void dw_mfma_reduce_kernel(const float* __restrict__ partial,

@@ -41,18 +41,21 @@ __global__ void sum_batch_kernel(const scalar_t* __restrict__ in,
 __global__ void sum_batch_final_kernel(const float* __restrict__ part,
                                        float* __restrict__ out,
                                        int nsplit, long M) {
-  __shared__ float red[4][64];
-  const int jt = threadIdx.x & 63;
-  const int zs = threadIdx.x >> 6;
-  const long j = (long)blockIdx.x * 64 + jt;
+  __shared__ float red[8][32];
+  const int jt = threadIdx.x & 31;
+  const int zs = threadIdx.x >> 5;
+  const long j = (long)blockIdx.x * 32 + jt;
   float s = 0.0f;
   if (j < M) {
-    for (int i = zs; i < nsplit; i += 4) s += part[(long)i * M + j];
+    for (int i = zs; i < nsplit; i += 8) s += part[(long)i * M + j];
   }
   red[zs][jt] = s;
   __syncthreads();
-  if (threadIdx.x < 64 && j < M) {
-    out[j] = red[0][jt] + red[1][jt] + red[2][jt] + red[3][jt];
+  if (threadIdx.x < 32 && j < M) {
+    float t = 0.0f;
+#pragma unroll
+    for (int z = 0; z < 8; ++z) t += red[z][jt];
+    out[j] = t;
   }
 }
 
@@ -155,7 +158,7 @@ at::Tensor sum_batch(const at::Tensor& in) {
                            part.data_ptr<float>(), B, M);
       });
   hipLaunchKernelGGL(sum_batch_final_kernel,
-                     dim3(sa::ceil_div(M, 64)), dim3(kBlock), 0,
+                     dim3(sa::ceil_div(M, 32)), dim3(kBlock), 0,
                      stream.stream(), part.data_ptr<float>(),
                      out.data_ptr<float>(), nsplit, M);
   return out;
