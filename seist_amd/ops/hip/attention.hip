@@ -302,10 +302,13 @@ void pooled_attn_bwd_q_kernel(const scalar_t* __restrict__ q,
 }
 
 // per-key: one block per (n, h); thread j owns key column j and streams the
-// queries through LDS chunks.
+// queries through 64-wide LDS chunks. When Lk <= 128 the block splits into
+// TWO query stripes (threads 128..255 process the chunk's second half with
+// their own accumulators, merged through LDS at the end) — otherwise half
+// the block would idle through the whole Lq loop.
 //   dV[e,j] = sum_lq Pdrop[lq,j] * dOut[e,lq]
 //   dK[e,j] = scale * sum_lq dS[lq,j] * Q[e,lq]
-template <typename scalar_t, int E, bool DROP>
+template <typename scalar_t, int E, bool DROP, int NS>
 __global__ __launch_bounds__(kBlock)
 void pooled_attn_bwd_kv_kernel(const scalar_t* __restrict__ q,
                                const scalar_t* __restrict__ k,
@@ -318,14 +321,17 @@ void pooled_attn_bwd_kv_kernel(const scalar_t* __restrict__ q,
                                scalar_t* __restrict__ dv,
                                long Lq, int Lk, int W, float scale,
                                float inv_keep) {
-  __shared__ float q_s[E * kLqChunk];
-  __shared__ float do_s[E * kLqChunk];
-  __shared__ float md_s[kLqChunk * 2];
-  __shared__ float delta_s[kLqChunk];
-  __shared__ unsigned mask_s[kLqChunk * 8];
+  constexpr int kChunk = 2 * kLqChunk;  // staged queries per iteration
+  __shared__ float q_s[E * kChunk];
+  __shared__ float do_s[E * kChunk];
+  __shared__ float md_s[kChunk * 2];
+  __shared__ float delta_s[kChunk];
+  __shared__ unsigned mask_s[kChunk * 8];
+  __shared__ float red[(NS > 1) ? 2 * E * 128 : 1];
 
   const long nh = blockIdx.x;
-  const int j = threadIdx.x;
+  const int j = (NS > 1) ? (threadIdx.x & 127) : threadIdx.x;
+  const int stripe = (NS > 1) ? (threadIdx.x >> 7) : 0;
   const scalar_t* qb = q + nh * (long)E * Lq;
   const scalar_t* db = dout + nh * (long)E * Lq;
 
@@ -340,16 +346,16 @@ void pooled_attn_bwd_kv_kernel(const scalar_t* __restrict__ q,
     }
   }
 
-  for (long lq0 = 0; lq0 < Lq; lq0 += kLqChunk) {
-    const int cn = (int)min((long)kLqChunk, Lq - lq0);
+  for (long lq0 = 0; lq0 < Lq; lq0 += kChunk) {
+    const int cn = (int)min((long)kChunk, Lq - lq0);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < E * kLqChunk; idx += kBlock) {
-      const int e = idx / kLqChunk;
-      const int i = idx - e * kLqChunk;
+    for (int idx = threadIdx.x; idx < E * kChunk; idx += kBlock) {
+      const int e = idx / kChunk;
+      const int i = idx - e * kChunk;
       const float qv = (i < cn) ? (float)qb[(long)e * Lq + lq0 + i] : 0.0f;
       const float dv_ = (i < cn) ? (float)db[(long)e * Lq + lq0 + i] : 0.0f;
-      q_s[e * kLqChunk + i] = qv;
-      do_s[e * kLqChunk + i] = dv_;
+      q_s[e * kChunk + i] = qv;
+      do_s[e * kChunk + i] = dv_;
     }
     for (int idx = threadIdx.x; idx < cn; idx += kBlock) {
       const long row = nh * Lq + lq0 + idx;
@@ -364,14 +370,16 @@ void pooled_attn_bwd_kv_kernel(const scalar_t* __restrict__ q,
     }
     __syncthreads();
     if (j >= Lk) continue;
-    for (int i = 0; i < cn; ++i) {
+    const int i0 = stripe * (cn > kLqChunk || NS == 1 ? kLqChunk : cn);
+    const int i1 = (NS > 1 && stripe == 0) ? min(cn, kLqChunk) : cn;
+    for (int i = i0; i < i1; ++i) {
       float s = 0.0f;
 #pragma unroll
-      for (int e = 0; e < E; ++e) s += q_s[e * kLqChunk + i] * kr[e];
+      for (int e = 0; e < E; ++e) s += q_s[e * kChunk + i] * kr[e];
       const float P = __expf(s * scale - md_s[i * 2 + 0]) / md_s[i * 2 + 1];
       float dP = 0.0f;
 #pragma unroll
-      for (int e = 0; e < E; ++e) dP += do_s[e * kLqChunk + i] * vr[e];
+      for (int e = 0; e < E; ++e) dP += do_s[e * kChunk + i] * vr[e];
       float Pd = P;
       if (DROP) {
         const bool keep = (mask_s[i * 8 + (j >> 5)] >> (j & 31)) & 1u;
@@ -381,8 +389,29 @@ void pooled_attn_bwd_kv_kernel(const scalar_t* __restrict__ q,
       const float dS = P * (dP - delta_s[i]);
 #pragma unroll
       for (int e = 0; e < E; ++e) {
-        dv_acc[e] += Pd * do_s[e * kLqChunk + i];
-        dk_acc[e] += dS * q_s[e * kLqChunk + i];
+        dv_acc[e] += Pd * do_s[e * kChunk + i];
+        dk_acc[e] += dS * q_s[e * kChunk + i];
+      }
+    }
+  }
+
+  if (NS > 1) {
+    // stripe 1 parks its accumulators in LDS; stripe 0 merges
+    __syncthreads();
+    if (stripe == 1 && j < Lk) {
+#pragma unroll
+      for (int e = 0; e < E; ++e) {
+        red[e * 128 + j] = dk_acc[e];
+        red[(E + e) * 128 + j] = dv_acc[e];
+      }
+    }
+    __syncthreads();
+    if (stripe == 1) return;
+    if (j < Lk) {
+#pragma unroll
+      for (int e = 0; e < E; ++e) {
+        dk_acc[e] += red[e * 128 + j];
+        dv_acc[e] += red[(E + e) * 128 + j];
       }
     }
   }
@@ -501,16 +530,21 @@ std::vector<at::Tensor> pooled_attn_bwd(const at::Tensor& q,
               drop ? (const unsigned*)mask.data_ptr<int>() : nullptr,
               dq.data_ptr<scalar_t>(), delta.data_ptr<float>(), Lq, Lk, W,
               scale, inv_keep);
-          hipLaunchKernelGGL(
-              (pooled_attn_bwd_kv_kernel<scalar_t, decltype(e_)::value,
-                                         decltype(d_)::value>),
-              dim3(N * H), dim3(kBlock), 0, stream.stream(),
-              q.data_ptr<scalar_t>(), k.data_ptr<scalar_t>(),
-              v.data_ptr<scalar_t>(), dout.data_ptr<scalar_t>(),
-              stats.data_ptr<float>(),
-              drop ? (const unsigned*)mask.data_ptr<int>() : nullptr,
-              delta.data_ptr<float>(), dk.data_ptr<scalar_t>(),
-              dv.data_ptr<scalar_t>(), Lq, Lk, W, scale, inv_keep);
+          auto launch_kv = [&](auto ns_) {
+            hipLaunchKernelGGL(
+                (pooled_attn_bwd_kv_kernel<scalar_t, decltype(e_)::value,
+                                           decltype(d_)::value,
+                                           decltype(ns_)::value>),
+                dim3(N * H), dim3(kBlock), 0, stream.stream(),
+                q.data_ptr<scalar_t>(), k.data_ptr<scalar_t>(),
+                v.data_ptr<scalar_t>(), dout.data_ptr<scalar_t>(),
+                stats.data_ptr<float>(),
+                drop ? (const unsigned*)mask.data_ptr<int>() : nullptr,
+                delta.data_ptr<float>(), dk.data_ptr<scalar_t>(),
+                dv.data_ptr<scalar_t>(), Lq, Lk, W, scale, inv_keep);
+          };
+          if (Lk <= 128) launch_kv(std::integral_constant<int, 2>{});
+          else launch_kv(std::integral_constant<int, 1>{});
         };
         auto launch_e = [&](auto e_) {
           if (drop) launch(e_, std::true_type{});
