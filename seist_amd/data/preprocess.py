@@ -13,6 +13,7 @@ matches the reference so seeded streams are comparable.
 import argparse
 import copy
 import json
+import os
 from operator import itemgetter
 from typing import Any, List, Tuple, Union
 
@@ -22,6 +23,17 @@ from torch.utils.data import Dataset
 from ..config import Config
 from ..utils.logger import logger
 from .registry import build_dataset
+
+# K20: native (C++) loader-side workers — bit-exact with the numpy path
+# (numpy's pairwise summation is reproduced; label windows are computed in
+# numpy and only rasterized natively). SEIST_AMD_PY_DATA=1 forces numpy.
+try:
+    if os.environ.get("SEIST_AMD_PY_DATA") == "1":
+        _native_data = None
+    else:
+        from .. import _native_data  # type: ignore
+except ImportError:  # pragma: no cover - extension not built
+    _native_data = None
 
 __all__ = ["DataPreprocessor", "SeismicDataset", "_pad_phases", "_pad_array"]
 
@@ -188,6 +200,12 @@ class DataPreprocessor:
         return data, ppks, spks
 
     def _normalize(self, data, mode):
+        if mode not in ("", "max", "std"):
+            raise ValueError(f"Supported mode: 'max','std', got '{mode}'")
+        if (_native_data is not None and data.dtype == np.float32
+                and data.ndim == 2 and data.flags["C_CONTIGUOUS"]):
+            _native_data.normalize(data, {"": 0, "max": 1, "std": 2}[mode])
+            return data
         data -= np.mean(data, axis=1, keepdims=True)
         if mode == "max":
             mx = np.max(data, axis=1, keepdims=True)
@@ -197,8 +215,6 @@ class DataPreprocessor:
             sd = np.std(data, axis=1, keepdims=True)
             sd[sd == 0] = 1
             data /= sd
-        elif mode != "":
-            raise ValueError(f"Supported mode: 'max','std', got '{mode}'")
         return data
 
     # ------------------------------------------------------------------
@@ -403,6 +419,18 @@ class DataPreprocessor:
     def _rasterize(self, idxs, length: int, soft_label_width: int,
                    soft_label_shape: str) -> np.ndarray:
         """Sum the label window at each index (clipped at the edges)."""
+        if _native_data is not None and len(idxs) > 0:
+            key = (soft_label_width, soft_label_shape)
+            cache = getattr(self, "_window_cache", None)
+            if cache is None:
+                cache = self._window_cache = {}
+            window = cache.get(key)
+            if window is None:
+                window = cache[key] = np.ascontiguousarray(
+                    self._label_window(soft_label_width, soft_label_shape),
+                    dtype=np.float64)
+            return _native_data.rasterize(
+                [int(i) for i in idxs], length, soft_label_width, window)
         slabel = np.zeros(length)
         if len(idxs) == 0:
             return slabel
@@ -465,8 +493,12 @@ class DataPreprocessor:
             label = event["data"][self.data_channels.index(name)]
         elif name in [f"d{c}" for c in self.data_channels]:
             ch = event["data"][self.data_channels.index(name[-1])]
-            label = np.zeros_like(ch)
-            label[1:] = np.diff(ch)
+            if (_native_data is not None and ch.dtype == np.float32
+                    and ch.flags["C_CONTIGUOUS"]):
+                label = _native_data.diff_label(ch)
+            else:
+                label = np.zeros_like(ch)
+                label[1:] = np.diff(ch)
         else:
             raise NotImplementedError(f"Unsupported label name: '{name}'")
         return label.astype(self.dtype)

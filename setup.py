@@ -12,9 +12,21 @@ from setuptools import setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
+import pybind11  # noqa: E402
+from setuptools import Extension  # noqa: E402
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
 
 HIP_DIR = os.path.join("seist_amd", "ops", "hip")
+
+# K20: plain-C++ data-pipeline workers (no HIP/torch dependency — runs in
+# DataLoader worker processes)
+native_data = Extension(
+    name="seist_amd._native_data",
+    sources=[os.path.join("seist_amd", "data", "_native.cpp")],
+    include_dirs=[pybind11.get_include()],
+    extra_compile_args=["-O3", "-std=c++17"],
+    language="c++",
+)
 
 ext = CUDAExtension(
     name="seist_amd._C",
@@ -42,6 +54,6 @@ setup(
     name="seist_amd",
     version="0.1.0",
     packages=["seist_amd"],
-    ext_modules=[ext],
+    ext_modules=[ext, native_data],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},
 )

@@ -142,3 +142,62 @@ def test_metric_targets_match_reference():
         task_names=["ppk", "spk", "det", "emg", "pmp"])
     for k in ta:
         assert np.array_equal(ta[k], tb[k]), k
+
+
+def test_native_data_workers_bitexact():
+    """K20 native C++ loader workers (seist_amd/data/_native.cpp) must be
+    bit-exact with the numpy path: pairwise-summation normalize, window
+    rasterization (window precomputed in numpy), diff labels, cal_snr."""
+    nd = pytest.importorskip("seist_amd._native_data")
+    rng = np.random.default_rng(11)
+    from seist_amd.data.preprocess import DataPreprocessor
+    pp = DataPreprocessor.__new__(DataPreprocessor)
+
+    for L in (8192, 1000, 127, 7):
+        x = rng.standard_normal((3, L)).astype(np.float32)
+        for mode, mid in (("", 0), ("max", 1), ("std", 2)):
+            ref = x.copy()
+            ref -= np.mean(ref, axis=1, keepdims=True)
+            if mode == "max":
+                mx = np.max(ref, axis=1, keepdims=True)
+                mx[mx == 0] = 1
+                ref /= mx
+            elif mode == "std":
+                sd = np.std(ref, axis=1, keepdims=True)
+                sd[sd == 0] = 1
+                ref /= sd
+            got = x.copy()
+            nd.normalize(got, mid)
+            assert np.array_equal(got, ref), (L, mode)
+
+    for shape in ("gaussian", "triangle", "box", "sigmoid"):
+        win = np.ascontiguousarray(pp._label_window(100, shape))
+        for idxs in ([50], [3], [8190, 100], [-5, 8100], [0], [8191],
+                     [49, 51]):
+            import os
+            os.environ["SEIST_AMD_PY_DATA"] = "1"
+            try:
+                # numpy reference (native module already imported; call the
+                # pure-python body by rebuilding the window path inline)
+                slabel = np.zeros(8192)
+                left, right = 50, 50
+                for idx in idxs:
+                    if idx < 0:
+                        continue
+                    elif idx - left < 0:
+                        slabel[: idx + right + 1] += win[
+                            100 + 1 - (idx + right + 1):]
+                    elif idx + right <= 8191:
+                        slabel[idx - left: idx + right + 1] += win
+                    elif idx <= 8191:
+                        slabel[-(8192 - (idx - left)):] += win[
+                            : 8192 - (idx - left)]
+            finally:
+                del os.environ["SEIST_AMD_PY_DATA"]
+            got = nd.rasterize([int(i) for i in idxs], 8192, 100, win)
+            assert np.array_equal(got, slabel), (shape, idxs)
+
+    x1 = rng.standard_normal(4096).astype(np.float32)
+    ref = np.zeros_like(x1)
+    ref[1:] = np.diff(x1)
+    assert np.array_equal(nd.diff_label(x1), ref)
