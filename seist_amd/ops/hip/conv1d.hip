@@ -48,6 +48,11 @@ constexpr int kJT = 32;         // output-channel chunk of the conv-GEMM
 // coalesced guarded loads, so the K-tap inner loop issues LDS reads instead
 // of K global loads per output element (the global-load version was ~3x off
 // roofline on the stems — VMEM instruction-issue bound, not bandwidth).
+// 4 outputs per thread (1024-wide l tile): at 256-wide tiles the grid was
+// ~128k sub-microsecond blocks and the step was dispatch-rate bound, not
+// bandwidth bound.
+constexpr int kDwTile = 4;
+
 template <typename scalar_t, bool HAS_BIAS>
 __global__ void dwconv_fwd_kernel(const scalar_t* __restrict__ x,
                                   const scalar_t* __restrict__ w,
@@ -59,9 +64,9 @@ __global__ void dwconv_fwd_kernel(const scalar_t* __restrict__ x,
   float* x_s = w_lds + K;
   const int n = blockIdx.y;
   const int co = blockIdx.z;
-  const long lo0 = (long)blockIdx.x * kBlock;
+  const long lo0 = (long)blockIdx.x * (kBlock * kDwTile);
   const long s0 = lo0 * stride - padl;
-  const int ext = kBlock * stride + (K - 1) * dil + 1;
+  const int ext = kBlock * kDwTile * stride + (K - 1) * dil + 1;
   const scalar_t* xr = x + ((long)n * Ci + co) * L;  // depthwise: ci == co
   for (int idx = threadIdx.x; idx < K; idx += kBlock) {
     w_lds[idx] = (float)w[(long)co * K + idx];
@@ -71,14 +76,19 @@ __global__ void dwconv_fwd_kernel(const scalar_t* __restrict__ x,
     x_s[idx] = (g >= 0 && g < L) ? (float)xr[g] : 0.0f;
   }
   __syncthreads();
-  const long lo = lo0 + threadIdx.x;
-  if (lo >= Lo) return;
-  float acc = HAS_BIAS ? (float)bias[co] : 0.0f;
-  const int base = threadIdx.x * stride;
-  for (int k = 0; k < K; ++k) {
-    acc += w_lds[k] * x_s[base + k * dil];
+  scalar_t* yr = y + ((long)n * Co + co) * Lo;
+  const float b0 = HAS_BIAS ? (float)bias[co] : 0.0f;
+#pragma unroll
+  for (int t = 0; t < kDwTile; ++t) {
+    const long lo = lo0 + t * kBlock + threadIdx.x;
+    if (lo >= Lo) break;
+    float acc = b0;
+    const int base = (t * kBlock + threadIdx.x) * stride;
+    for (int k = 0; k < K; ++k) {
+      acc += w_lds[k] * x_s[base + k * dil];
+    }
+    yr[lo] = (scalar_t)acc;
   }
-  y[((long)n * Co + co) * Lo + lo] = (scalar_t)acc;
 }
 
 template <typename scalar_t, int STRIDE>
@@ -91,12 +101,12 @@ __global__ void dwconv_dx_kernel(const scalar_t* __restrict__ dy,
   float* dy_s = w_lds + K;
   const int n = blockIdx.y;
   const int ci = blockIdx.z;
-  const long li0 = (long)blockIdx.x * kBlock;
+  const long li0 = (long)blockIdx.x * (kBlock * kDwTile);
   const int s = (STRIDE > 0) ? STRIDE : stride;
-  // lo range touched by li in [li0, li0+kBlock): ceil((li+padl-(K-1)dil)/s)
-  // .. (li+padl)/s — stage with one element of slack on each side
+  // lo range touched by li in [li0, li0+kBlock*kDwTile): stage with one
+  // element of slack on each side
   const long lob = (li0 + padl - (long)(K - 1) * dil) / (long)s - 1;
-  const int ext = kBlock / s + ((K - 1) * dil) / s + 4;
+  const int ext = (kBlock * kDwTile) / s + ((K - 1) * dil) / s + 4;
   const scalar_t* dyr = dy + ((long)n * Co + ci) * Lo;
   for (int idx = threadIdx.x; idx < K; idx += kBlock) {
     w_lds[idx] = (float)w[(long)ci * K + idx];
@@ -106,25 +116,29 @@ __global__ void dwconv_dx_kernel(const scalar_t* __restrict__ dy,
     dy_s[idx] = (g >= 0 && g < Lo) ? (float)dyr[g] : 0.0f;
   }
   __syncthreads();
-  const long li = li0 + threadIdx.x;
-  if (li >= L) return;
-  float acc = 0.0f;
-  if (STRIDE != 1 && dil == 1) {
-    // phase decomposition: only k == (li+padl) mod s hits a valid lo
-    for (int k = (int)((li + padl) % s); k < K; k += s) {
-      const long lo = (li + padl - k) / s;
-      if (lo >= 0 && lo < Lo) acc += w_lds[k] * dy_s[lo - lob];
+  scalar_t* dxr = dx + ((long)n * Ci + ci) * L;
+#pragma unroll
+  for (int t = 0; t < kDwTile; ++t) {
+    const long li = li0 + t * kBlock + threadIdx.x;
+    if (li >= L) break;
+    float acc = 0.0f;
+    if (STRIDE != 1 && dil == 1) {
+      // phase decomposition: only k == (li+padl) mod s hits a valid lo
+      for (int k = (int)((li + padl) % s); k < K; k += s) {
+        const long lo = (li + padl - k) / s;
+        if (lo >= 0 && lo < Lo) acc += w_lds[k] * dy_s[lo - lob];
+      }
+    } else {
+      for (int k = 0; k < K; ++k) {
+        const long num = li + padl - (long)k * dil;
+        if (num < 0) continue;
+        if (STRIDE != 1 && (num % s)) continue;
+        const long lo = (STRIDE == 1) ? num : num / s;
+        if (lo < Lo) acc += w_lds[k] * dy_s[lo - lob];
+      }
     }
-  } else {
-    for (int k = 0; k < K; ++k) {
-      const long num = li + padl - (long)k * dil;
-      if (num < 0) continue;
-      if (STRIDE != 1 && (num % s)) continue;
-      const long lo = (STRIDE == 1) ? num : num / s;
-      if (lo < Lo) acc += w_lds[k] * dy_s[lo - lob];
-    }
+    dxr[li] = (scalar_t)acc;
   }
-  dx[((long)n * Ci + ci) * L + li] = (scalar_t)acc;
 }
 
 // ---------------- dense / grouped conv-GEMM ----------------
@@ -391,9 +405,9 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
       "conv1d_fwd", [&] {
         const scalar_t* bp = has_bias ? bct.data_ptr<scalar_t>() : nullptr;
         if (Cog == 1 && Cig == 1) {  // true depthwise (groups == Ci == Co)
-          dim3 grid(sa::ceil_div(Lo, kBlock), N, Co);
+          dim3 grid(sa::ceil_div(Lo, (long)kBlock * kDwTile), N, Co);
           const size_t lds = sizeof(float) *
-              (K + kBlock * stride + (K - 1) * dilation + 1);
+              (K + kBlock * kDwTile * stride + (K - 1) * dilation + 1);
           TORCH_CHECK(lds <= 64 * 1024, "depthwise LDS window too large");
           if (has_bias) {
             hipLaunchKernelGGL((dwconv_fwd_kernel<scalar_t, true>), grid,
@@ -461,9 +475,10 @@ void conv1d_dx_into(const at::Tensor& dy, const at::Tensor& w,
       at::ScalarType::BFloat16, at::ScalarType::Half, dx.scalar_type(),
       "conv1d_dx", [&] {
         if (Cog == 1 && Cig == 1) {
-          dim3 grid(sa::ceil_div(L, kBlock), N, Ci);
+          dim3 grid(sa::ceil_div(L, (long)kBlock * kDwTile), N, Ci);
           const size_t lds = sizeof(float) *
-              (K + kBlock / stride + ((K - 1) * dilation) / stride + 4);
+              (K + (kBlock * kDwTile) / stride
+               + ((K - 1) * dilation) / stride + 4);
           TORCH_CHECK(lds <= 64 * 1024, "depthwise LDS window too large");
           auto launch_dw = [&](auto st) {
             hipLaunchKernelGGL((dwconv_dx_kernel<scalar_t,
