@@ -82,6 +82,10 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
   }
 }
 
+// 4 elements per thread: elementwise grids of 128k one-wave blocks were
+// dispatch-rate bound, not bandwidth bound
+constexpr int kEwTile = 4;
+
 template <typename scalar_t>
 __global__ void bn_apply_kernel(const scalar_t* __restrict__ x,
                                 scalar_t* __restrict__ y,
@@ -90,12 +94,16 @@ __global__ void bn_apply_kernel(const scalar_t* __restrict__ x,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
                                 int C, long L, long total, int act) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= total) return;
-  const int c = (int)((i / L) % C);
-  const float xh = ((float)x[i] - mean[c]) * invstd[c];
-  const float pre = xh * gamma[c] + beta[c];
-  y[i] = (scalar_t)sa::act_fwd(pre, act);
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i =
+        ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= total) return;
+    const int c = (int)((i / L) % C);
+    const float xh = ((float)x[i] - mean[c]) * invstd[c];
+    const float pre = xh * gamma[c] + beta[c];
+    y[i] = (scalar_t)sa::act_fwd(pre, act);
+  }
 }
 
 template <typename scalar_t>
@@ -168,22 +176,26 @@ __global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
                                  const float* __restrict__ beta,
                                  const float* __restrict__ sums,  // dbeta,dgamma
                                  int C, long L, long total, long NL, int act) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= total) return;
-  const int c = (int)((i / L) % C);
-  const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
-  const float xh = ((float)x[i] - m) * is;
-  float d = (float)dy[i];
-  if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
-  float v;
-  if (TRAINING) {
-    const float dbeta = sums[c * 2 + 0];
-    const float dgamma = sums[c * 2 + 1];
-    v = (g * is / (float)NL) * ((float)NL * d - dbeta - xh * dgamma);
-  } else {
-    v = d * g * is;
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i =
+        ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= total) return;
+    const int c = (int)((i / L) % C);
+    const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
+    const float xh = ((float)x[i] - m) * is;
+    float d = (float)dy[i];
+    if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
+    float v;
+    if (TRAINING) {
+      const float dbeta = sums[c * 2 + 0];
+      const float dgamma = sums[c * 2 + 1];
+      v = (g * is / (float)NL) * ((float)NL * d - dbeta - xh * dgamma);
+    } else {
+      v = d * g * is;
+    }
+    dx[i] = (scalar_t)v;
   }
-  dx[i] = (scalar_t)v;
 }
 
 int pick_nsplit(long N, int C) {
@@ -246,7 +258,8 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "bn_apply", [&] {
         hipLaunchKernelGGL((bn_apply_kernel<scalar_t>),
-                           dim3(sa::ceil_div(total, kBlock)), dim3(kBlock), 0,
+                           dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                           dim3(kBlock), 0,
                            stream.stream(), x.data_ptr<scalar_t>(),
                            y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
                            invstd.data_ptr<float>(), g32.data_ptr<float>(),
@@ -295,8 +308,9 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
       "bn_bwd_dx", [&] {
         if (training) {
           hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, true>),
-                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
-                             0, stream.stream(), dy.data_ptr<scalar_t>(),
+                             dim3(sa::ceil_div(total,
+                                               (long)kBlock * kEwTile)),
+                             dim3(kBlock), 0, stream.stream(), dy.data_ptr<scalar_t>(),
                              x.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                              mean.data_ptr<float>(), invstd.data_ptr<float>(),
                              g32.data_ptr<float>(), b32.data_ptr<float>(),
@@ -304,8 +318,9 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
                              (int)act);
         } else {
           hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, false>),
-                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
-                             0, stream.stream(), dy.data_ptr<scalar_t>(),
+                             dim3(sa::ceil_div(total,
+                                               (long)kBlock * kEwTile)),
+                             dim3(kBlock), 0, stream.stream(), dy.data_ptr<scalar_t>(),
                              x.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                              mean.data_ptr<float>(), invstd.data_ptr<float>(),
                              g32.data_ptr<float>(), b32.data_ptr<float>(),

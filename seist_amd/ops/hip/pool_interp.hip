@@ -13,13 +13,16 @@
 namespace {
 
 constexpr int kBlock = 256;
+constexpr int kEwTile = 4;  // elements per thread (dispatch-rate relief)
 
 template <typename scalar_t>
 __global__ void avgmax_fwd_kernel(const scalar_t* __restrict__ x,
                                   scalar_t* __restrict__ y,
                                   int* __restrict__ argmax,
                                   long L, long Lo, int k, long rows) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+  const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
   if (i >= rows * Lo) return;
   const long row = i / Lo;
   const long lo = i - row * Lo;
@@ -39,6 +42,7 @@ __global__ void avgmax_fwd_kernel(const scalar_t* __restrict__ x,
   }
   y[i] = (scalar_t)(s / (float)(lo1 - lo0) + mx);
   argmax[i] = mi;
+  }
 }
 
 template <typename scalar_t>
@@ -46,25 +50,30 @@ __global__ void avgmax_bwd_kernel(const scalar_t* __restrict__ dy,
                                   const int* __restrict__ argmax,
                                   scalar_t* __restrict__ dx,
                                   long L, long Lo, int k, long rows) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= rows * L) return;
-  const long row = i / L;
-  const long li = i - row * L;
-  const long lo = li / k;
-  const long lo0 = lo * k;
-  const long lo1 = min(lo0 + (long)k, L);
-  const long oi = row * Lo + lo;
-  const float g = (float)dy[oi];
-  float v = g / (float)(lo1 - lo0);
-  if (argmax[oi] == (int)li) v += g;
-  dx[i] = (scalar_t)v;
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= rows * L) return;
+    const long row = i / L;
+    const long li = i - row * L;
+    const long lo = li / k;
+    const long lo0 = lo * k;
+    const long lo1 = min(lo0 + (long)k, L);
+    const long oi = row * Lo + lo;
+    const float g = (float)dy[oi];
+    float v = g / (float)(lo1 - lo0);
+    if (argmax[oi] == (int)li) v += g;
+    dx[i] = (scalar_t)v;
+  }
 }
 
 template <typename scalar_t>
 __global__ void interp_fwd_kernel(const scalar_t* __restrict__ x,
                                   scalar_t* __restrict__ y,
                                   long Li, long Lo, float scale, long rows) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+  const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
   if (i >= rows * Lo) return;
   const long row = i / Lo;
   const long lo = i - row * Lo;
@@ -76,6 +85,7 @@ __global__ void interp_fwd_kernel(const scalar_t* __restrict__ x,
   const float w1 = src - (float)l0;
   const scalar_t* xr = x + row * Li;
   y[i] = (scalar_t)((1.0f - w1) * (float)xr[l0] + w1 * (float)xr[l1]);
+  }
 }
 
 // Gather formulation: each dx element sums the dy window that maps onto
@@ -179,7 +189,8 @@ std::vector<at::Tensor> avgmax_pool_fwd(const at::Tensor& x, long k) {
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "avgmax_fwd", [&] {
         hipLaunchKernelGGL((avgmax_fwd_kernel<scalar_t>),
-                           dim3(sa::ceil_div(rows * Lo, kBlock)),
+                           dim3(sa::ceil_div(rows * Lo,
+                                             (long)kBlock * kEwTile)),
                            dim3(kBlock), 0, stream.stream(),
                            x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
                            idx.data_ptr<int>(), L, Lo, (int)k, rows);
@@ -197,7 +208,8 @@ at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
       at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
       "avgmax_bwd", [&] {
         hipLaunchKernelGGL((avgmax_bwd_kernel<scalar_t>),
-                           dim3(sa::ceil_div(rows * in_len, kBlock)),
+                           dim3(sa::ceil_div(rows * in_len,
+                                             (long)kBlock * kEwTile)),
                            dim3(kBlock), 0, stream.stream(),
                            dy.data_ptr<scalar_t>(), argmax.data_ptr<int>(),
                            dx.data_ptr<scalar_t>(), in_len, Lo, (int)k, rows);
@@ -216,7 +228,8 @@ at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len) {
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "interp_fwd", [&] {
         hipLaunchKernelGGL((interp_fwd_kernel<scalar_t>),
-                           dim3(sa::ceil_div(rows * out_len, kBlock)),
+                           dim3(sa::ceil_div(rows * out_len,
+                                             (long)kBlock * kEwTile)),
                            dim3(kBlock), 0, stream.stream(),
                            x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
                            Li, out_len, scale, rows);
@@ -266,26 +279,32 @@ template <typename scalar_t>
 __global__ void up2_fwd_kernel(const scalar_t* __restrict__ x,
                                scalar_t* __restrict__ y,
                                long Li, long rows) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= rows * Li) return;
-  const long row = i / Li;
-  const long li = i - row * Li;
-  const scalar_t v = x[i];
-  scalar_t* yr = y + row * Li * 2 + li * 2;
-  yr[0] = v;
-  yr[1] = v;
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= rows * Li) return;
+    const long row = i / Li;
+    const long li = i - row * Li;
+    const scalar_t v = x[i];
+    scalar_t* yr = y + row * Li * 2 + li * 2;
+    yr[0] = v;
+    yr[1] = v;
+  }
 }
 
 template <typename scalar_t>
 __global__ void up2_bwd_kernel(const scalar_t* __restrict__ dy,
                                scalar_t* __restrict__ dx,
                                long Li, long rows) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= rows * Li) return;
-  const long row = i / Li;
-  const long li = i - row * Li;
-  const scalar_t* dyr = dy + row * Li * 2 + li * 2;
-  dx[i] = (scalar_t)((float)dyr[0] + (float)dyr[1]);
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= rows * Li) return;
+    const long row = i / Li;
+    const long li = i - row * Li;
+    const scalar_t* dyr = dy + row * Li * 2 + li * 2;
+    dx[i] = (scalar_t)((float)dyr[0] + (float)dyr[1]);
+  }
 }
 
 }  // namespace
@@ -300,7 +319,8 @@ at::Tensor upsample2x_fwd(const at::Tensor& x) {
       at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
       "up2_fwd", [&] {
         hipLaunchKernelGGL((up2_fwd_kernel<scalar_t>),
-                           dim3(sa::ceil_div(rows * Li, kBlock)),
+                           dim3(sa::ceil_div(rows * Li,
+                                             (long)kBlock * kEwTile)),
                            dim3(kBlock), 0, stream.stream(),
                            x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
                            Li, rows);
@@ -319,7 +339,8 @@ at::Tensor upsample2x_bwd(const at::Tensor& dy) {
       at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
       "up2_bwd", [&] {
         hipLaunchKernelGGL((up2_bwd_kernel<scalar_t>),
-                           dim3(sa::ceil_div(rows * Li, kBlock)),
+                           dim3(sa::ceil_div(rows * Li,
+                                             (long)kBlock * kEwTile)),
                            dim3(kBlock), 0, stream.stream(),
                            dy.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                            Li, rows);

@@ -12,6 +12,7 @@
 namespace {
 
 constexpr int kBlock = 256;
+constexpr int kEwTile = 4;  // elements per thread (dispatch-rate relief)
 
 template <typename scalar_t, bool HAS_MASK>
 __global__ void row_scale_add_kernel(const scalar_t* __restrict__ x,
@@ -20,11 +21,14 @@ __global__ void row_scale_add_kernel(const scalar_t* __restrict__ x,
                                      scalar_t* __restrict__ z,
                                      float scale, long row_elems,
                                      long total) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= total) return;
-  float m = scale;
-  if (HAS_MASK) m *= mask[i / row_elems];
-  z[i] = (scalar_t)((float)x[i] + m * (float)y[i]);
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= total) return;
+    float m = scale;
+    if (HAS_MASK) m *= mask[i / row_elems];
+    z[i] = (scalar_t)((float)x[i] + m * (float)y[i]);
+  }
 }
 
 template <typename scalar_t, bool HAS_MASK>
@@ -32,11 +36,14 @@ __global__ void row_scale_kernel(const scalar_t* __restrict__ y,
                                  const float* __restrict__ mask,
                                  scalar_t* __restrict__ z,
                                  float scale, long row_elems, long total) {
-  const long i = (long)blockIdx.x * kBlock + threadIdx.x;
-  if (i >= total) return;
-  float m = scale;
-  if (HAS_MASK) m *= mask[i / row_elems];
-  z[i] = (scalar_t)(m * (float)y[i]);
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= total) return;
+    float m = scale;
+    if (HAS_MASK) m *= mask[i / row_elems];
+    z[i] = (scalar_t)(m * (float)y[i]);
+  }
 }
 
 }  // namespace
@@ -56,14 +63,16 @@ at::Tensor row_scale_add(const at::Tensor& x, const at::Tensor& y,
       "row_scale_add", [&] {
         if (has_mask) {
           hipLaunchKernelGGL((row_scale_add_kernel<scalar_t, true>),
-                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
+                             dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                             dim3(kBlock),
                              0, stream.stream(), x.data_ptr<scalar_t>(),
                              y.data_ptr<scalar_t>(), mask->data_ptr<float>(),
                              z.data_ptr<scalar_t>(), (float)scale, row_elems,
                              total);
         } else {
           hipLaunchKernelGGL((row_scale_add_kernel<scalar_t, false>),
-                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
+                             dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                             dim3(kBlock),
                              0, stream.stream(), x.data_ptr<scalar_t>(),
                              y.data_ptr<scalar_t>(), nullptr,
                              z.data_ptr<scalar_t>(), (float)scale, row_elems,
@@ -86,13 +95,15 @@ at::Tensor row_scale(const at::Tensor& y,
       "row_scale", [&] {
         if (has_mask) {
           hipLaunchKernelGGL((row_scale_kernel<scalar_t, true>),
-                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
+                             dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                             dim3(kBlock),
                              0, stream.stream(), y.data_ptr<scalar_t>(),
                              mask->data_ptr<float>(), z.data_ptr<scalar_t>(),
                              (float)scale, row_elems, total);
         } else {
           hipLaunchKernelGGL((row_scale_kernel<scalar_t, false>),
-                             dim3(sa::ceil_div(total, kBlock)), dim3(kBlock),
+                             dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                             dim3(kBlock),
                              0, stream.stream(), y.data_ptr<scalar_t>(),
                              nullptr, z.data_ptr<scalar_t>(),
                              (float)scale, row_elems, total);
