@@ -139,11 +139,13 @@ __global__ void interp_bwd_lds_kernel(const scalar_t* __restrict__ dy,
                                       long rows) {
   extern __shared__ float dy_s[];
   const long row = blockIdx.y;
-  const long li0 = (long)blockIdx.x * kBlock;
+  const long li0 = (long)blockIdx.x * (kBlock * kEwTile);
   const float inv = 1.0f / scale;
   long blo = (long)floorf(((float)li0 - 0.5f) * inv - 0.5f) - 1;
   blo = max(blo, (long)0);
-  long bhi = (long)ceilf(((float)(li0 + kBlock - 1) + 1.5f) * inv - 0.5f) + 1;
+  long bhi = (long)ceilf(
+                 ((float)(li0 + kBlock * kEwTile - 1) + 1.5f) * inv - 0.5f)
+             + 1;
   bhi = min(bhi, Lo - 1);
   const int ext = (int)(bhi - blo + 1);
   const scalar_t* dyr = dy + row * Lo;
@@ -151,27 +153,30 @@ __global__ void interp_bwd_lds_kernel(const scalar_t* __restrict__ dy,
     dy_s[idx] = (float)dyr[blo + idx];
   }
   __syncthreads();
-  const long li = li0 + threadIdx.x;
-  if (li >= Li) return;
-  long lo_lo = (long)floorf(((float)li - 1.0f + 0.5f) * inv - 0.5f) - 1;
-  long lo_hi = (long)ceilf(((float)li + 1.0f + 0.5f) * inv - 0.5f) + 1;
-  if (li == 0) lo_lo = 0;
-  lo_lo = max(lo_lo, blo);
-  lo_hi = min(lo_hi, bhi);
-  float acc = 0.0f;
-  for (long lo = lo_lo; lo <= lo_hi; ++lo) {
-    float src = ((float)lo + 0.5f) * scale - 0.5f;
-    src = fmaxf(src, 0.0f);
-    long l0 = (long)src;
-    l0 = min(l0, Li - 1);
-    const long l1 = min(l0 + 1, Li - 1);
-    const float w1 = src - (float)l0;
-    float wgt = 0.0f;
-    if (l0 == li) wgt += 1.0f - w1;
-    if (l1 == li && w1 != 0.0f) wgt += w1;
-    acc += wgt * dy_s[lo - blo];
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long li = li0 + t * kBlock + threadIdx.x;
+    if (li >= Li) break;
+    long lo_lo = (long)floorf(((float)li - 1.0f + 0.5f) * inv - 0.5f) - 1;
+    long lo_hi = (long)ceilf(((float)li + 1.0f + 0.5f) * inv - 0.5f) + 1;
+    if (li == 0) lo_lo = 0;
+    lo_lo = max(lo_lo, blo);
+    lo_hi = min(lo_hi, bhi);
+    float acc = 0.0f;
+    for (long lo = lo_lo; lo <= lo_hi; ++lo) {
+      float src = ((float)lo + 0.5f) * scale - 0.5f;
+      src = fmaxf(src, 0.0f);
+      long l0 = (long)src;
+      l0 = min(l0, Li - 1);
+      const long l1 = min(l0 + 1, Li - 1);
+      const float w1 = src - (float)l0;
+      float wgt = 0.0f;
+      if (l0 == li) wgt += 1.0f - w1;
+      if (l1 == li && w1 != 0.0f) wgt += w1;
+      acc += wgt * dy_s[lo - blo];
+    }
+    dx[row * Li + li] = (scalar_t)acc;
   }
-  dx[row * Li + li] = (scalar_t)acc;
 }
 
 }  // namespace
@@ -248,9 +253,9 @@ at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len) {
       "interp_bwd", [&] {
         // dy window of one 256-li block, with slack for the edge clamps
         const float inv = 1.0f / scale;
-        const long ext_bound = (long)((kBlock + 4) * inv) + 8;
+        const long ext_bound = (long)((kBlock * kEwTile + 4) * inv) + 8;
         if (ext_bound <= 12288 && rows <= 65535) {
-          dim3 grid(sa::ceil_div(in_len, (long)kBlock), rows);
+          dim3 grid(sa::ceil_div(in_len, (long)kBlock * kEwTile), rows);
           hipLaunchKernelGGL((interp_bwd_lds_kernel<scalar_t>), grid,
                              dim3(kBlock), sizeof(float) * ext_bound,
                              stream.stream(), dy.data_ptr<scalar_t>(),
