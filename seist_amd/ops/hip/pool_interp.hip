@@ -157,6 +157,15 @@ __global__ void interp_bwd_lds_kernel(const scalar_t* __restrict__ dy,
   for (int t = 0; t < kEwTile; ++t) {
     const long li = li0 + t * kBlock + threadIdx.x;
     if (li >= Li) break;
+    if (Lo == 2 * Li && li > 0 && li < Li - 1) {
+      // exact 2x upsample (every power-of-two head stage): closed 4-tap
+      // form; dy[2li]/dy[2li+1] carry 0.75, the outer neighbours 0.25
+      const long b = 2 * li - blo;
+      const float acc2 = 0.25f * dy_s[b - 1] + 0.75f * dy_s[b]
+                         + 0.75f * dy_s[b + 1] + 0.25f * dy_s[b + 2];
+      dx[row * Li + li] = (scalar_t)acc2;
+      continue;
+    }
     long lo_lo = (long)floorf(((float)li - 1.0f + 0.5f) * inv - 0.5f) - 1;
     long lo_hi = (long)ceilf(((float)li + 1.0f + 0.5f) * inv - 0.5f) + 1;
     if (li == 0) lo_lo = 0;
