@@ -237,6 +237,15 @@ def train_worker(args, device) -> str:
                      if args.use_tensorboard else None)
     if pdist.is_main_process():
         os.makedirs(checkpoint_save_dir, exist_ok=True)
+        if args.use_tensorboard:
+            # convenience launcher next to the run dir (reference
+            # train.py:193-194 writes the same helper)
+            helper = os.path.join(log_dir, "run_tensorboard.sh")
+            with open(helper, "w") as f:
+                f.write("#!/bin/bash\ntensorboard --logdir "
+                        f"{os.path.join(log_dir, 'tensorboard')} "
+                        "--port ${1:-6006}\n")
+            os.chmod(helper, 0o755)
 
     model_inputs, model_labels, model_tasks = Config.get_model_config_(
         args.model_name, "inputs", "labels", "eval")
