@@ -131,3 +131,22 @@ def test_torchrun_bench_cpu(tmp_path):
     line = [l for l in res.stdout.splitlines() if l.startswith("{")][-1]
     out = _json.loads(line)
     assert out["n_gpus"] == 2 and out["value"] > 0
+
+
+def test_demo_predict_end_to_end(tmp_path):
+    """Train one tiny epoch, then run demo_predict.py against the saved
+    checkpoint (the README deployment recipe)."""
+    import subprocess
+    import sys
+    args = _args(tmp_path, ["--model-name", "seist_s_dpk", "--mode", "train"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    ckpts = sorted(glob.glob(str(tmp_path / "*" / "checkpoints" / "*.pth")))
+    assert ckpts
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = subprocess.run(
+        [sys.executable, "demo_predict.py", "--checkpoint", ckpts[-1],
+         "--model-name", "seist_s_dpk", "--save-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=600, cwd=root)
+    assert res.returncode == 0, res.stderr[-1500:]
+    assert glob.glob(str(tmp_path / "*demo_prediction*.png"))

@@ -146,3 +146,28 @@ def test_sym3_eig_analytic():
     # eigen equation A v = lambda v
     err = (torch.matmul(cov, vecs) - vals.transpose(1, 2) * vecs).abs().max()
     assert err < 1e-4
+
+
+def test_seist_activation_checkpointing_matches():
+    """use_checkpoint=True (reference seist.py:841-847 parity) must not
+    change forward or gradients."""
+    import torch
+    from seist_amd.models import create_model
+    torch.manual_seed(0)
+    m1 = create_model("seist_s_dpk", in_channels=3, in_samples=512)
+    m2 = create_model("seist_s_dpk", in_channels=3, in_samples=512,
+                      use_checkpoint=True)
+    m2.load_state_dict(m1.state_dict())
+    m1.train(); m2.train()
+    # droppath/dropout randomness: evaluate in eval mode for exactness
+    m1.eval(); m2.eval()
+    x = torch.randn(2, 3, 512)
+    y1 = m1(x)
+    y2 = m2(x)
+    assert torch.allclose(y1, y2, atol=1e-6)
+
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    m1(x1).sum().backward()
+    m2(x2).sum().backward()
+    assert torch.allclose(x1.grad, x2.grad, atol=1e-6)
