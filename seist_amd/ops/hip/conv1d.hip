@@ -35,6 +35,10 @@ c10::optional<at::Tensor> dw_mfma_try(const at::Tensor& dy,
 bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
                    const c10::optional<at::Tensor>& bias, at::Tensor& y,
                    long padl, long dilation, long groups, bool is_dx);
+bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
+                     const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                     long stride, long padl, long dilation, long groups,
+                     bool is_dx);
 
 namespace {
 
@@ -395,6 +399,10 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
                        /*is_dx=*/false)) {
     return y;
   }
+  if (conv_tap_s_mfma(x, w, bias, y, stride, padl, dilation, groups,
+                      /*is_dx=*/false)) {
+    return y;
+  }
   if (groups == 1 && stride == 1
       && conv_mfma(x, w, bias, y, padl, dilation, /*is_dx=*/false)) {
     return y;
@@ -465,6 +473,10 @@ void conv1d_dx_into(const at::Tensor& dy, const at::Tensor& w,
   if (stride == 1
       && conv_tap_mfma(dy, w, c10::nullopt, dx, padl, dilation, groups,
                        /*is_dx=*/true)) {
+    return;
+  }
+  if (conv_tap_s_mfma(dy, w, c10::nullopt, dx, stride, padl, dilation,
+                      groups, /*is_dx=*/true)) {
     return;
   }
   if (groups == 1 && stride == 1

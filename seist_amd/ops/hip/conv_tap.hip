@@ -239,3 +239,193 @@ bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
   }
   return true;
 }
+
+// ---------------------------------------------------------------------------
+// Strided variant (S in {2, 4}, dense, dilation 1): PhaseNet's stride-4
+// encoder convs and the transposed-conv gather. 64-wide l tile (the x
+// window is LT*S + K wide, so the 256-l tile of the stride-1 kernel would
+// not fit in LDS at S=4); 4 waves side by side along l, one 16x16 output
+// fragment each. Same staging discipline: each input channel row staged
+// once per (l-tile, ci-chunk); B fragments gathered at stride S (fwd) or
+// with the % S divisibility mask (dx).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int kLTS = 64;  // l per block in the strided kernel
+
+template <bool IS_DX, bool HAS_BIAS, int S>
+__global__ __launch_bounds__(kBlock)
+void conv_tap_s_kernel(const sa_bf16* __restrict__ x,
+                       const sa_bf16* __restrict__ w,
+                       const sa_bf16* __restrict__ bias,
+                       sa_bf16* __restrict__ y,
+                       int N, int Cin, int Cout, long Lin, long Lout,
+                       int K, int padl, int xext, int xpitch) {
+  extern __shared__ sa_bf16 smem[];
+  sa_bf16* w_s = smem;                         // [16][K][kCT] (c contiguous)
+  sa_bf16* x_s = smem + 16 * K * kCT;          // [kCT][xpitch]
+
+  const int n = blockIdx.y;
+  const int m0 = blockIdx.z * 16;
+  const long l0 = (long)blockIdx.x * kLTS;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int frag_m = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+
+  // fwd: x index = l*S - padl + k; staged from aligned s0
+  // dx:  dy index = (l + padl - k)/S where divisible; staged likewise
+  long s0;
+  int base;  // l0's first index relative to s0
+  long lob = 0;
+  if (IS_DX) {
+    lob = (l0 + padl - (long)(K - 1)) / (long)S - 1;
+    if (lob < 0) lob = 0;
+    s0 = 0;  // unused in dx mode
+    base = 0;
+  } else {
+    s0 = (l0 * S - padl) & ~7L;
+    base = (int)(l0 * S - padl - s0);
+  }
+
+  const sa_bf16* xb = x + (long)n * Cin * Lin;
+
+  for (int c0 = 0; c0 < Cin; c0 += kCT) {
+    const int cn = min(kCT, Cin - c0);
+    __syncthreads();
+    // ---- weight chunk [16 m][K][kCT c] ----
+    for (int idx = tid; idx < 16 * K * kCT; idx += kBlock) {
+      const int m = idx / (K * kCT);
+      const int r = idx - m * K * kCT;
+      const int k = r / kCT;
+      const int c = r - k * kCT;
+      const int mg = m0 + m;
+      const int cg = c0 + c;
+      float v = 0.0f;
+      if (mg < Cout && c < cn) {
+        const int co = IS_DX ? cg : mg;
+        const int ci = IS_DX ? mg : cg;
+        v = (float)w[((long)co * (IS_DX ? Cout : Cin) + ci) * K + k];
+      }
+      w_s[(m * K + k) * kCT + c] = (sa_bf16)v;
+    }
+    // ---- input rows [kCT c][xext], staged once (aligned b128) ----
+    for (int idx = tid; idx < kCT * (xext / 8); idx += kBlock) {
+      const int c = idx / (xext / 8);
+      const int e8 = idx - c * (xext / 8);
+      const long gl = (IS_DX ? lob : s0) + e8 * 8;
+      bf16x8 v = {};
+      if (c < cn) {
+        const sa_bf16* row = xb + (long)(c0 + c) * Lin;
+        if (gl >= 0 && gl + 8 <= Lin) {
+          v = *(const bf16x8*)(row + gl);
+        } else {
+          for (int j = 0; j < 8; ++j) {
+            const long lj = gl + j;
+            if (lj >= 0 && lj < Lin) v[j] = row[lj];
+          }
+        }
+      }
+      *(bf16x8*)(x_s + c * xpitch + e8 * 8) = v;
+    }
+    __syncthreads();
+
+    const int ll = wid * 16 + (lane & 15);  // this lane's l column in tile
+    for (int k = 0; k < K; ++k) {
+      const bf16x8 a =
+          *(const bf16x8*)(w_s + (frag_m * K + k) * kCT + kbase);
+      int idxl = -1;
+      if (IS_DX) {
+        const long num = l0 + ll + padl - k;
+        if (num >= 0 && num % S == 0) {
+          const long lo = num / S;
+          if (lo < Lin) idxl = (int)(lo - lob);
+        }
+      } else {
+        idxl = base + ll * S + k;
+      }
+      bf16x8 b = {};
+      if (idxl >= 0) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          b[j] = x_s[(kbase + j) * xpitch + idxl];
+        }
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: D col = l, row = out channel ----
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+  const long lg = l0 + wid * 16 + d_col;
+  if (lg >= Lout) return;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int mg = m0 + d_row0 + r;
+    if (mg < Cout) {
+      float v = acc[r];
+      if (HAS_BIAS) v += (float)bias[mg];
+      y[((long)n * Cout + mg) * Lout + lg] = (sa_bf16)v;
+    }
+  }
+}
+
+}  // namespace
+
+// strided dense path; returns false outside the envelope
+bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
+                     const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                     long stride, long padl, long dilation, long groups,
+                     bool is_dx) {
+  if (x.scalar_type() != at::kBFloat16 || w.scalar_type() != at::kBFloat16)
+    return false;
+  if (groups != 1 || dilation != 1 || (stride != 2 && stride != 4))
+    return false;
+  const int N = x.size(0), Cin = x.size(1);
+  const long Lin = x.size(2);
+  const int Cout = y.size(1);
+  const long Lout = y.size(2);
+  const int K = w.size(2);
+  if (K < 1 || K > 24) return false;
+  // fwd window: kLTS*S + K + slack; dx window: kLTS/S + K/S + slack
+  const int xext = is_dx
+      ? ((kLTS / (int)stride + K / (int)stride + 12 + 7) & ~7)
+      : ((kLTS * (int)stride + K + 16 + 7) & ~7);
+  const int xpitch = xext + 3;
+  const size_t lds = sizeof(sa_bf16) * (16 * K * kCT + kCT * xpitch);
+  if (lds > kMaxLds) return false;
+
+  dim3 grid(sa::ceil_div(Lout, (long)kLTS), N, (Cout + 15) / 16);
+  auto stream = at::hip::getCurrentHIPStream();
+  const bool has_bias = bias.has_value() && bias->defined();
+  at::Tensor bct;
+  if (has_bias) bct = bias->to(x.scalar_type()).contiguous();
+  const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
+
+  auto launch = [&](auto dx_t, auto hb_t, auto s_t) {
+    hipLaunchKernelGGL(
+        (conv_tap_s_kernel<decltype(dx_t)::value, decltype(hb_t)::value,
+                           decltype(s_t)::value>),
+        grid, dim3(kBlock), lds, stream.stream(),
+        (const sa_bf16*)x.data_ptr(), (const sa_bf16*)w.data_ptr(), bp,
+        (sa_bf16*)y.data_ptr(), N, Cin, Cout, Lin, Lout, K, (int)padl,
+        xext, xpitch);
+  };
+  auto launch_hb = [&](auto dx_t, auto s_t) {
+    if (has_bias) launch(dx_t, std::true_type{}, s_t);
+    else launch(dx_t, std::false_type{}, s_t);
+  };
+  auto launch_dx = [&](auto s_t) {
+    if (is_dx) launch_hb(std::true_type{}, s_t);
+    else launch_hb(std::false_type{}, s_t);
+  };
+  if (stride == 2) launch_dx(std::integral_constant<int, 2>{});
+  else launch_dx(std::integral_constant<int, 4>{});
+  return true;
+}

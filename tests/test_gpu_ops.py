@@ -111,6 +111,50 @@ def test_conv1d_grouped_bf16(dev, groups, k, C, L, stride, dil):
     (ops/hip/conv_tap.hip) and the diagonal-tile weight gradient
     (ops/hip/dw_mfma.hip). Reference = fp32 CPU on the same
     bf16-quantized inputs."""
+    _run_grouped_bf16_case(dev, groups, k, C, L, stride, dil)
+
+
+STRIDED_CASES = [
+    (8, 8, 7, 4, 8192),    # phasenet encoder conv (strided MFMA path)
+    (16, 16, 7, 4, 2048),
+    (32, 64, 7, 2, 1000),  # stride-2, Ci != Co, odd L
+    (8, 16, 7, 4, 515),    # tail chunk
+]
+
+
+@pytest.mark.parametrize("Ci,Co,k,stride,L", STRIDED_CASES)
+def test_conv1d_strided_dense_bf16(dev, Ci, Co, k, stride, L):
+    """bf16 strided dense convs: fwd/dx on the strided tap MFMA kernel
+    (ops/hip/conv_tap.hip conv_tap_s_kernel); used by PhaseNet's encoder
+    and the transposed-conv gather."""
+    torch.manual_seed(7)
+    N = 3
+    from seist_amd.ops.functional import auto_pad_lr
+    padl, padr = auto_pad_lr(L, k, stride)
+    x32 = torch.randn(N, Ci, L).to(torch.bfloat16).float()
+    w32 = (torch.randn(Co, Ci, k) * 0.2).to(torch.bfloat16).float()
+    b32 = (torch.randn(Co) * 0.1).to(torch.bfloat16).float()
+
+    xg = x32.to(dev, torch.bfloat16).requires_grad_(True)
+    wg = w32.to(dev, torch.bfloat16).requires_grad_(True)
+    bg = b32.to(dev, torch.bfloat16).requires_grad_(True)
+    y = ops.conv1d(xg, wg, bg, stride=stride, padding=(padl, padr))
+
+    xc = x32.clone().requires_grad_(True)
+    wc = w32.clone().requires_grad_(True)
+    bc = b32.clone().requires_grad_(True)
+    y_ref = ops.conv1d(xc, wc, bc, stride=stride, padding=(padl, padr))
+    _cmp(y, y_ref, 5e-2, 1e-2, msg="strided bf16 fwd")
+
+    g32 = torch.randn_like(y_ref).to(torch.bfloat16).float()
+    y.backward(g32.to(dev, torch.bfloat16))
+    y_ref.backward(g32)
+    _cmp(xg.grad, xc.grad, 1e-1, 1e-2, msg="strided bf16 dx")
+    _cmp(wg.grad, wc.grad, 0.0, 4e-2, msg="strided bf16 dw")
+    _cmp(bg.grad, bc.grad, 0.0, 4e-2, msg="strided bf16 db")
+
+
+def _run_grouped_bf16_case(dev, groups, k, C, L, stride, dil):
     torch.manual_seed(3)
     N = 3
     padl = (k - 1) * dil // 2
