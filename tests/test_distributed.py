@@ -85,10 +85,10 @@ def _w_metrics_sync(rank):
     assert d == 16
 
 
-def _w_train(rank, tmpdir):
+def _w_train(rank, tmpdir, mode="train"):
     from seist_amd.cli import get_args, main_worker
     args = get_args([
-        "--mode", "train", "--model-name", "phasenet",
+        "--mode", mode, "--model-name", "phasenet",
         "--dataset-name", "synthetic", "--dataset-size", "24",
         "--dataset-samples", "9000", "--batch-size", "2", "--epochs", "1",
         "--workers", "0", "--device", "cpu", "--use-tensorboard", "false",
@@ -117,3 +117,13 @@ def test_two_rank_training(tmp_path):
     _spawn(_w_train, 29514, str(tmp_path))
     import glob
     assert glob.glob(str(tmp_path / "*" / "checkpoints" / "*.pth"))
+
+
+def test_two_rank_train_test(tmp_path):
+    """train_test mode under 2 ranks: exercises the validate/test reduce
+    paths, the best-checkpoint broadcast into test_worker, and the rank-0
+    ResultSaver gating (exactly one CSV)."""
+    _spawn(_w_train, 29515, str(tmp_path), "train_test")
+    import glob
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) == 1
