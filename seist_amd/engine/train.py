@@ -32,7 +32,7 @@ from ..parallel import dist as pdist
 from ..parallel.ddp import wrap_distributed
 from ..utils.logger import logger
 from ..utils.meters import AverageMeter, ProgressMeter
-from ..utils.misc import count_parameters, get_safe_path, get_time_str, strftimedelta
+from ..utils.misc import count_parameters, get_safe_path, strftimedelta
 from .metrics import Metrics
 from .postprocess import process_outputs
 from .precision import cast_inputs, convert_to_bf16

@@ -12,7 +12,6 @@ convs are LDS-tiled HIP kernels, BatchNorm+GELU is fused, and attention runs
 as a fused pooled-KV kernel (see ``seist_amd.ops``).
 """
 
-import math
 from collections import OrderedDict
 from functools import partial
 
