@@ -17,6 +17,8 @@
 #include <cstdint>
 #include <vector>
 
+#include "_rng.h"
+
 namespace py = pybind11;
 
 namespace {
@@ -163,8 +165,40 @@ double cal_snr_native(py::array_t<float, py::array::c_style> data, long pat,
 
 }  // namespace
 
+// Python-visible wrapper for tests / future C++ augmentation chain
+class PyRandomState {
+ public:
+  explicit PyRandomState(uint32_t seed) : rs_(seed) {}
+  double random_sample() { return rs_.random_sample(); }
+  double uniform(double lo, double hi) { return rs_.uniform(lo, hi); }
+  long randint(long lo, long hi) { return rs_.randint(lo, hi); }
+  double gauss() { return rs_.gauss(); }
+  py::array_t<double> standard_normal(long n) {
+    py::array_t<double> out(n);
+    auto b = out.mutable_unchecked<1>();
+    for (long i = 0; i < n; ++i) b(i) = rs_.gauss();
+    return out;
+  }
+  std::vector<long> permutation(long n) { return rs_.permutation(n); }
+  std::vector<long> choice_no_replace(long n, long size) {
+    return rs_.choice_no_replace(n, size);
+  }
+
+ private:
+  seist_rng::RandomState rs_;
+};
+
 PYBIND11_MODULE(_native_data, m) {
   m.doc() = "seist_amd native data-pipeline workers (K20)";
+  py::class_<PyRandomState>(m, "RandomState")
+      .def(py::init<uint32_t>())
+      .def("random_sample", &PyRandomState::random_sample)
+      .def("uniform", &PyRandomState::uniform)
+      .def("randint", &PyRandomState::randint)
+      .def("gauss", &PyRandomState::gauss)
+      .def("standard_normal", &PyRandomState::standard_normal)
+      .def("permutation", &PyRandomState::permutation)
+      .def("choice_no_replace", &PyRandomState::choice_no_replace);
   m.def("normalize", &normalize, "in-place demean + max/std normalize",
         py::arg("data"), py::arg("mode"));
   m.def("rasterize", &rasterize, "sum label window at indices",

@@ -201,3 +201,32 @@ def test_native_data_workers_bitexact():
     ref = np.zeros_like(x1)
     ref[1:] = np.diff(x1)
     assert np.array_equal(nd.diff_label(x1), ref)
+
+
+def test_native_randomstate_bitexact():
+    """C++ MT19937 RandomState (data/_rng.h) must reproduce numpy's legacy
+    generator draw-for-draw — the prerequisite for porting the seeded
+    augmentation chain to the native workers."""
+    nd = pytest.importorskip("seist_amd._native_data")
+    for seed in (0, 42, 20260913):
+        np_rs = np.random.RandomState(seed)
+        rs = nd.RandomState(seed)
+        for i in range(120):
+            k = i % 5
+            if k == 0:
+                assert np_rs.random_sample() == rs.random_sample()
+            elif k == 1:
+                assert np_rs.uniform(1, 3) == rs.uniform(1, 3)
+            elif k == 2:
+                assert np_rs.randint(0, 8192) == rs.randint(0, 8192)
+            elif k == 3:
+                assert np_rs.randn() == rs.gauss()
+            else:
+                assert np.array_equal(np_rs.randn(5), rs.standard_normal(5))
+        np_rs = np.random.RandomState(seed)
+        rs = nd.RandomState(seed)
+        assert np.array_equal(np_rs.permutation(33), rs.permutation(33))
+        np_rs = np.random.RandomState(seed)
+        rs = nd.RandomState(seed)
+        assert np.array_equal(np_rs.choice(range(96), 20, replace=False),
+                              rs.choice_no_replace(96, 20))
