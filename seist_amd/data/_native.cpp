@@ -76,6 +76,18 @@ float pairwise_sum_f32(const float* a, ssize_t n) {
   return pairwise_sum_f32(a, n2) + pairwise_sum_f32(a + n2, n - n2);
 }
 
+// np.sum / np.mean iterate in 8192-element reduce buffers combined
+// sequentially; pairwise applies within each buffer (verified empirically
+// against numpy 2.2 — plain whole-array pairwise diverges for n > 8194)
+float npsum_f32(const float* a, ssize_t n) {
+  float s = 0.0f;
+  for (ssize_t off = 0; off < n; off += 8192) {
+    const float cs = pairwise_sum_f32(a + off, std::min<ssize_t>(8192, n - off));
+    s = (off == 0) ? cs : s + cs;
+  }
+  return s;
+}
+
 // demean + optional per-channel scaling, matching
 // DataPreprocessor._normalize (numpy semantics: mean/std computed in the
 // array dtype via pairwise summation; np.std is the biased estimator)
@@ -84,7 +96,7 @@ void normalize(py::array_t<float, py::array::c_style> data, int mode) {
   const ssize_t C = buf.shape(0), L = buf.shape(1);
   for (ssize_t c = 0; c < C; ++c) {
     float* row = buf.mutable_data(c, 0);
-    const float mean = pairwise_sum_f32(row, L) / (float)L;
+    const float mean = npsum_f32(row, L) / (float)L;
     for (ssize_t l = 0; l < L; ++l) row[l] -= mean;
     if (mode == 1) {  // "max"
       float mx = row[0];
@@ -94,13 +106,13 @@ void normalize(py::array_t<float, py::array::c_style> data, int mode) {
     } else if (mode == 2) {  // "std"
       // np.std: mean of squared deviations (row is already demeaned, but
       // numpy recomputes the mean of the demeaned row — reproduce that)
-      const float m2 = pairwise_sum_f32(row, L) / (float)L;
+      const float m2 = npsum_f32(row, L) / (float)L;
       std::vector<float> sq((size_t)L);
       for (ssize_t l = 0; l < L; ++l) {
         const float d = row[l] - m2;
         sq[(size_t)l] = d * d;
       }
-      float sd = std::sqrt(pairwise_sum_f32(sq.data(), L) / (float)L);
+      float sd = std::sqrt(npsum_f32(sq.data(), L) / (float)L);
       if (sd == 0.0f) sd = 1.0f;
       for (ssize_t l = 0; l < L; ++l) row[l] /= sd;
     }
@@ -165,7 +177,13 @@ double cal_snr_native(py::array_t<float, py::array::c_style> data, long pat,
 
 }  // namespace
 
-// Python-visible wrapper for tests / future C++ augmentation chain
+// full per-sample pipeline (implemented in _augment.cpp)
+py::tuple process_event(py::array_t<float, py::array::c_style> data,
+                        std::vector<long> ppks, std::vector<long> spks,
+                        std::vector<double> snr, bool augmentation,
+                        py::dict params, seist_rng::RandomState* rng);
+
+// Python-visible wrapper for tests / the C++ augmentation chain
 class PyRandomState {
  public:
   explicit PyRandomState(uint32_t seed) : rs_(seed) {}
@@ -183,6 +201,19 @@ class PyRandomState {
   std::vector<long> choice_no_replace(long n, long size) {
     return rs_.choice_no_replace(n, size);
   }
+  void set_state(const std::vector<uint32_t>& keys, int pos,
+                 bool has_gauss, double gauss) {
+    rs_.set_state(keys, pos, has_gauss, gauss);
+  }
+  py::tuple get_state() const {
+    std::vector<uint32_t> keys;
+    int pos;
+    bool hg;
+    double g;
+    rs_.get_state(&keys, &pos, &hg, &g);
+    return py::make_tuple(keys, pos, hg, g);
+  }
+  seist_rng::RandomState* raw() { return &rs_; }
 
  private:
   seist_rng::RandomState rs_;
@@ -198,7 +229,20 @@ PYBIND11_MODULE(_native_data, m) {
       .def("gauss", &PyRandomState::gauss)
       .def("standard_normal", &PyRandomState::standard_normal)
       .def("permutation", &PyRandomState::permutation)
-      .def("choice_no_replace", &PyRandomState::choice_no_replace);
+      .def("choice_no_replace", &PyRandomState::choice_no_replace)
+      .def("set_state", &PyRandomState::set_state)
+      .def("get_state", &PyRandomState::get_state);
+  m.def("process_event",
+        [](py::array_t<float, py::array::c_style> data,
+           std::vector<long> ppks, std::vector<long> spks,
+           std::vector<double> snr, bool augmentation, py::dict params,
+           PyRandomState& rng) {
+          return process_event(data, std::move(ppks), std::move(spks),
+                               std::move(snr), augmentation, params,
+                               rng.raw());
+        },
+        "full per-sample preprocessing pipeline (noise gate, pad, "
+        "augment, cut, normalize) — bit-exact with the numpy path");
   m.def("normalize", &normalize, "in-place demean + max/std normalize",
         py::arg("data"), py::arg("mode"));
   m.def("rasterize", &rasterize, "sum label window at indices",

@@ -32,6 +32,26 @@ class RandomState {
     gauss_ = 0.0;
   }
 
+  // adopt / export the full generator state (numpy get_state tuple:
+  // keys[624], pos, has_gauss, cached gauss) so the native stream can
+  // continue exactly where np.random left off
+  void set_state(const std::vector<uint32_t>& keys, int pos,
+                 bool has_gauss, double gauss) {
+    if (keys.size() != 624) throw std::invalid_argument("state: 624 keys");
+    for (int i = 0; i < 624; ++i) mt_[i] = keys[(size_t)i];
+    mti_ = pos;
+    has_gauss_ = has_gauss;
+    gauss_ = gauss;
+  }
+
+  void get_state(std::vector<uint32_t>* keys, int* pos, bool* has_gauss,
+                 double* gauss) const {
+    keys->assign(mt_, mt_ + 624);
+    *pos = mti_;
+    *has_gauss = has_gauss_;
+    *gauss = gauss_;
+  }
+
   // rk_random: one 32-bit MT19937 draw
   uint32_t next32() {
     if (mti_ >= 624) generate_block();

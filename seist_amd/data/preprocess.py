@@ -371,6 +371,63 @@ class DataPreprocessor:
     # main entry
     # ------------------------------------------------------------------
 
+    def _native_params(self):
+        params = getattr(self, "_native_params_cache", None)
+        if params is None:
+            params = self._native_params_cache = dict(
+                min_snr=float(self.min_snr), coda_ratio=float(self.coda_ratio),
+                p_position_ratio=float(self.p_position_ratio),
+                add_event_rate=float(self.add_event_rate),
+                add_noise_rate=float(self.add_noise_rate),
+                add_gap_rate=float(self.add_gap_rate),
+                drop_channel_rate=float(self.drop_channel_rate),
+                scale_amplitude_rate=float(self.scale_amplitude_rate),
+                pre_emphasis_rate=float(self.pre_emphasis_rate),
+                pre_emphasis_ratio=float(self.pre_emphasis_ratio),
+                generate_noise_rate=float(self.generate_noise_rate),
+                shift_event_rate=float(self.shift_event_rate),
+                max_event_num=int(self._max_event_num),
+                mask_percent=int(self.mask_percent),
+                noise_percent=int(self.noise_percent),
+                min_event_gap=int(self.min_event_gap),
+                in_samples=int(self.in_samples),
+                sampling_rate=int(self.sampling_rate),
+                norm_mode={"": 0, "max": 1, "std": 2}[self.norm_mode])
+        return params
+
+    def process_native(self, event: dict, augmentation: bool, rng) -> dict:
+        """Full pipeline in the C++ worker (data/_augment.cpp): bit-exact
+        with :meth:`process` when ``rng`` continues np.random's stream
+        (see SeismicDataset). Falls back to the numpy path — with the MT
+        state synced both ways — outside the native envelope."""
+        data = event["data"]
+        fixed_p = 0.0 <= self.p_position_ratio <= 1.0
+        ok = (isinstance(data, np.ndarray) and data.dtype == np.float32
+              and data.ndim == 2 and data.flags["C_CONTIGUOUS"]
+              and (fixed_p or data.shape[-1] >= self.in_samples))
+        if not ok:
+            keys, pos, hg, g = rng.get_state()
+            np.random.set_state(("MT19937", np.array(keys, dtype=np.uint32),
+                                 pos, int(hg), float(g)))
+            event = self.process(event, augmentation)
+            st = np.random.get_state()
+            rng.set_state([int(k) for k in st[1]], int(st[2]), bool(st[3]),
+                          float(st[4]))
+            return event
+
+
+        snr = np.atleast_1d(np.asarray(event["snr"], dtype=np.float64))
+        out, ppks, spks, cleared = _native_data.process_event(
+            data, [int(p) for p in event["ppks"]],
+            [int(s) for s in event["spks"]], [float(v) for v in snr],
+            bool(augmentation), self._native_params(), rng)
+        if cleared:
+            self._clear_dict_except(event, "data", "ppks", "spks")
+        event["data"] = out
+        event["ppks"] = list(ppks)
+        event["spks"] = list(spks)
+        return event
+
     def process(self, event: dict, augmentation: bool,
                 inplace: bool = True) -> dict:
         if not inplace:
@@ -659,11 +716,35 @@ class SeismicDataset(Dataset):
         return 2 * self._dataset_size if self._augmentation \
             else self._dataset_size
 
+    def __getstate__(self):
+        # the pybind RandomState is not picklable (spawn-mode workers);
+        # each process re-adopts np.random's state lazily
+        state = dict(self.__dict__)
+        state.pop("_native_rng_obj", None)
+        return state
+
+    def _native_rng(self):
+        """Per-process C++ RandomState continuing np.random's exact MT
+        stream (lazily adopted, so a DataLoader worker that forked with a
+        copied numpy state behaves identically to the Python path)."""
+        rng = getattr(self, "_native_rng_obj", None)
+        if rng is None:
+            st = np.random.get_state()
+            rng = _native_data.RandomState(0)
+            rng.set_state([int(k) for k in st[1]], int(st[2]), bool(st[3]),
+                          float(st[4]))
+            self._native_rng_obj = rng
+        return rng
+
     def __getitem__(self, idx: int):
         event, meta_data = self._dataset[idx % self._dataset_size]
-        event = self._preprocessor.process(
-            event=event,
-            augmentation=(self._augmentation and idx >= self._dataset_size))
+        augment = self._augmentation and idx >= self._dataset_size
+        if _native_data is not None:
+            event = self._preprocessor.process_native(
+                event=event, augmentation=augment, rng=self._native_rng())
+        else:
+            event = self._preprocessor.process(
+                event=event, augmentation=augment)
         inputs = self._preprocessor.get_inputs(
             event=event, input_names=self._input_names)
         loss_targets = self._preprocessor.get_targets_for_loss(

@@ -22,7 +22,8 @@ HIP_DIR = os.path.join("seist_amd", "ops", "hip")
 # DataLoader worker processes)
 native_data = Extension(
     name="seist_amd._native_data",
-    sources=[os.path.join("seist_amd", "data", "_native.cpp")],
+    sources=[os.path.join("seist_amd", "data", "_native.cpp"),
+             os.path.join("seist_amd", "data", "_augment.cpp")],
     include_dirs=[pybind11.get_include()],
     extra_compile_args=["-O3", "-std=c++17"],
     language="c++",

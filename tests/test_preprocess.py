@@ -230,3 +230,91 @@ def test_native_randomstate_bitexact():
         rs = nd.RandomState(seed)
         assert np.array_equal(np_rs.choice(range(96), 20, replace=False),
                               rs.choice_no_replace(96, 20))
+
+
+def test_native_process_event_bitexact():
+    """Full C++ pipeline (data/_augment.cpp process_event) vs the numpy
+    path: noise gate, phase padding, all augmentations in RNG order,
+    window cut and normalize — bit-exact on identical seeding."""
+    nd = pytest.importorskip("seist_amd._native_data")
+    import copy
+    from seist_amd.data.preprocess import DataPreprocessor
+
+    def make_pp(**over):
+        kw = dict(data_channels=["z", "n", "e"], sampling_rate=50,
+                  in_samples=8192, min_snr=-10.0, coda_ratio=1.4,
+                  norm_mode="std", p_position_ratio=-1.0,
+                  add_event_rate=0.3, add_noise_rate=0.5, add_gap_rate=0.4,
+                  drop_channel_rate=0.4, scale_amplitude_rate=0.4,
+                  pre_emphasis_rate=0.4, pre_emphasis_ratio=0.97,
+                  max_event_num=3, generate_noise_rate=0.2,
+                  shift_event_rate=0.4, mask_percent=0, noise_percent=0,
+                  min_event_gap_sec=0.2, soft_label_shape="gaussian",
+                  soft_label_width=100)
+        kw.update(over)
+        return DataPreprocessor(**kw)
+
+    for trial in range(40):
+        rng0 = np.random.default_rng(trial)
+        L = int(rng0.integers(8192, 16000))
+        data = (rng0.standard_normal((3, L)) * 0.3).astype(np.float32)
+        ppks, spks, base = [], [], 100
+        for _ in range(int(rng0.integers(1, 3))):
+            p = int(rng0.integers(base, base + 2000))
+            s = p + int(rng0.integers(50, 800))
+            ppks.append(p)
+            spks.append(s)
+            base = s + 500
+        over = {}
+        snr = [20.0] * 3 if trial % 5 else [-100.0] * 3
+        if trial % 7 == 0:
+            over = dict(mask_percent=30, noise_percent=20)
+        if trial % 11 == 0:
+            over = dict(p_position_ratio=0.3)
+            snr = [20.0] * 3
+        pp = make_pp(**over)
+        ev = {"data": data.copy(), "ppks": list(ppks), "spks": list(spks),
+              "snr": np.array(snr)}
+        np.random.seed(trial)
+        out = pp.process(copy.deepcopy(ev), augmentation=True)
+        R = nd.RandomState(trial)
+        d2, p2, s2, _ = nd.process_event(
+            data.copy(), list(ppks), list(spks), list(map(float, snr)),
+            True, pp._native_params(), R)
+        assert np.array_equal(out["data"].astype(np.float32), d2), trial
+        assert list(out["ppks"]) == list(p2), trial
+        assert list(out["spks"]) == list(s2), trial
+
+
+def test_native_dataset_pipeline_bitexact():
+    """SeismicDataset with the native loader path must produce the same
+    bytes as the Python path (the C++ RandomState continues np.random's
+    exact MT stream)."""
+    pytest.importorskip("seist_amd._native_data")
+    import seist_amd.data.preprocess as pre
+    from seist_amd.cli import get_args
+    from seist_amd.config import Config
+    from seist_amd.data.preprocess import SeismicDataset
+
+    args = get_args(["--mode", "train", "--model-name", "seist_m_dpk",
+                     "--dataset-name", "synthetic", "--dataset-size", "8",
+                     "--dataset-samples", "9000", "--augmentation", "true",
+                     "--device", "cpu"])
+    inp, lab, tasks = Config.get_model_config_(
+        "seist_m_dpk", "inputs", "labels", "eval")
+
+    def collect(native):
+        saved = pre._native_data
+        if not native:
+            pre._native_data = None
+        try:
+            np.random.seed(123)
+            ds = SeismicDataset(args=args, input_names=inp,
+                                label_names=lab, task_names=tasks,
+                                mode="train")
+            return [np.asarray(ds[i][0]) for i in (0, 3, 9, 12, 15)]
+        finally:
+            pre._native_data = saved
+
+    for x1, x2 in zip(collect(True), collect(False)):
+        assert np.array_equal(x1, x2)
