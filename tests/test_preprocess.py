@@ -318,3 +318,59 @@ def test_native_dataset_pipeline_bitexact():
 
     for x1, x2 in zip(collect(True), collect(False)):
         assert np.array_equal(x1, x2)
+
+
+def test_native_pipeline_fallback_resync():
+    """Traces shorter than in_samples take the numpy pad branch; the MT
+    state must sync native->numpy->native so the stream stays identical."""
+    pytest.importorskip("seist_amd._native_data")
+    import seist_amd.data.preprocess as pre
+    from seist_amd.cli import get_args
+    from seist_amd.config import Config
+    from seist_amd.data.preprocess import SeismicDataset
+
+    args = get_args(["--mode", "train", "--model-name", "seist_m_dpk",
+                     "--dataset-name", "synthetic", "--dataset-size", "8",
+                     "--dataset-samples", "5000", "--augmentation", "true",
+                     "--device", "cpu"])
+    inp, lab, tasks = Config.get_model_config_(
+        "seist_m_dpk", "inputs", "labels", "eval")
+
+    def collect(native):
+        saved = pre._native_data
+        if not native:
+            pre._native_data = None
+        try:
+            np.random.seed(7)
+            ds = SeismicDataset(args=args, input_names=inp,
+                                label_names=lab, task_names=tasks,
+                                mode="train")
+            return [np.asarray(ds[i][0], dtype=np.float64)
+                    for i in (0, 2, 9, 11)]
+        finally:
+            pre._native_data = saved
+
+    for x, y in zip(collect(True), collect(False)):
+        assert np.array_equal(x, y)
+
+
+def test_native_dataset_picklable():
+    """A live pybind RandomState must not break DataLoader spawn-mode
+    pickling (dropped in __getstate__, re-adopted lazily)."""
+    pytest.importorskip("seist_amd._native_data")
+    import pickle
+    from seist_amd.cli import get_args
+    from seist_amd.config import Config
+    from seist_amd.data.preprocess import SeismicDataset
+    args = get_args(["--mode", "train", "--model-name", "seist_m_dpk",
+                     "--dataset-name", "synthetic", "--dataset-size", "4",
+                     "--dataset-samples", "9000", "--augmentation", "true",
+                     "--device", "cpu"])
+    inp, lab, tasks = Config.get_model_config_(
+        "seist_m_dpk", "inputs", "labels", "eval")
+    np.random.seed(1)
+    ds = SeismicDataset(args=args, input_names=inp, label_names=lab,
+                        task_names=tasks, mode="train")
+    _ = ds[0]
+    ds2 = pickle.loads(pickle.dumps(ds))
+    _ = ds2[1]
