@@ -27,6 +27,43 @@ from .models import (
     get_model_list,
 )
 
+_ZNE = ["z", "n", "e"]
+
+
+def _bind(loss, inputs, labels, eval_tasks, tgt_loss=None, out_loss=None,
+          out_results=None):
+    """One model->task binding (the reference spells these dicts out
+    longhand; the keys and values are identical)."""
+    return {
+        "loss": loss,
+        "inputs": inputs,
+        "labels": labels,
+        "eval": eval_tasks,
+        "targets_transform_for_loss": tgt_loss,
+        "outputs_transform_for_loss": out_loss,
+        "outputs_transform_for_results": out_results,
+    }
+
+
+def _deg2trig(x):
+    rad = x * math.pi / 180
+    return rad.cos(), rad.sin()
+
+
+def _io(kind, metrics, num_classes=None):
+    item = {"type": kind, "metrics": list(metrics)}
+    if num_classes is not None:
+        item["num_classes"] = num_classes
+    return item
+
+
+# metric bundles shared by io items
+_WAVE_M = ("mean", "rmse", "mae")
+_PICK_M = ("precision", "recall", "f1", "mean", "rmse", "mae", "mape")
+_DET_M = ("precision", "recall", "f1")
+_IDX_M = ("mean", "rmse", "mae", "mape", "r2")
+_REG_M = ("mean", "rmse", "mae", "r2")
+
 
 class Config:
     _model_conf_keys = (
@@ -38,111 +75,39 @@ class Config:
     )
 
     models = {
-        # ------------------------------------------------------ PhaseNet
-        "phasenet": {
-            "loss": partial(CELoss, weight=[[1], [1], [1]]),
-            "inputs": [["z", "n", "e"]],
-            "labels": [["non", "ppk", "spk"]],
-            "eval": ["ppk", "spk"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
-        # ------------------------------------------------------ EQTransformer
-        "eqtransformer": {
-            "loss": partial(BCELoss, weight=[[0.5], [1], [1]]),
-            "inputs": [["z", "n", "e"]],
-            "labels": [["det", "ppk", "spk"]],
-            "eval": ["det", "ppk", "spk"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
-        # ------------------------------------------------------ MagNet
-        "magnet": {
-            "loss": MousaviLoss,
-            "inputs": [["z", "n", "e"]],
-            "labels": ["emg"],
-            "eval": ["emg"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": lambda x: x[:, 0].reshape(-1, 1),
-        },
-        # ------------------------------------------------------ BAZ Network
-        "baz_network": {
-            "loss": partial(CombinationLoss, losses=[MSELoss, MSELoss]),
-            "inputs": [["z", "n", "e"]],
-            "labels": ["baz"],
-            "eval": ["baz"],
-            "targets_transform_for_loss": lambda x: (
-                (x * math.pi / 180).cos(),
-                (x * math.pi / 180).sin(),
-            ),
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": lambda x: torch.atan2(x[1], x[0])
-            * 180
-            / math.pi,
-        },
-        # ------------------------------------------------------ DiTingMotion
-        "ditingmotion": {
-            "loss": partial(CombinationLoss, losses=[FocalLoss, FocalLoss]),
-            "inputs": [["z", "dz"]],
-            "labels": ["clr", "pmp"],
-            "eval": ["pmp"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": lambda xs: [x.softmax(-1) for x in xs],
-        },
-        # ------------------------------------------------------ SeisT dpk
-        "seist_.*?_dpk.*": {
-            "loss": partial(BCELoss, weight=[[0.5], [1], [1]]),
-            "inputs": [["z", "n", "e"]],
-            "labels": [["det", "ppk", "spk"]],
-            "eval": ["det", "ppk", "spk"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
-        # ------------------------------------------------------ SeisT pmp
-        "seist_.*?_pmp": {
-            "loss": partial(CELoss, weight=[1, 1]),
-            "inputs": [["z", "n", "e"]],
-            "labels": ["pmp"],
-            "eval": ["pmp"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
-        # ------------------------------------------------------ SeisT emg
-        "seist_.*?_emg": {
-            "loss": HuberLoss,
-            "inputs": [["z", "n", "e"]],
-            "labels": ["emg"],
-            "eval": ["emg"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
-        # ------------------------------------------------------ SeisT baz
-        "seist_.*?_baz": {
-            "loss": HuberLoss,
-            "inputs": [["z", "n", "e"]],
-            "labels": ["baz"],
-            "eval": ["baz"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
-        # ------------------------------------------------------ SeisT dis
-        "seist_.*?_dis": {
-            "loss": HuberLoss,
-            "inputs": [["z", "n", "e"]],
-            "labels": ["dis"],
-            "eval": ["dis"],
-            "targets_transform_for_loss": None,
-            "outputs_transform_for_loss": None,
-            "outputs_transform_for_results": None,
-        },
+        "phasenet": _bind(
+            partial(CELoss, weight=[[1], [1], [1]]),
+            inputs=[_ZNE], labels=[["non", "ppk", "spk"]],
+            eval_tasks=["ppk", "spk"]),
+        "eqtransformer": _bind(
+            partial(BCELoss, weight=[[0.5], [1], [1]]),
+            inputs=[_ZNE], labels=[["det", "ppk", "spk"]],
+            eval_tasks=["det", "ppk", "spk"]),
+        "magnet": _bind(
+            MousaviLoss, inputs=[_ZNE], labels=["emg"], eval_tasks=["emg"],
+            out_results=lambda x: x[:, 0].reshape(-1, 1)),
+        "baz_network": _bind(
+            partial(CombinationLoss, losses=[MSELoss, MSELoss]),
+            inputs=[_ZNE], labels=["baz"], eval_tasks=["baz"],
+            tgt_loss=_deg2trig,
+            out_results=lambda x: torch.atan2(x[1], x[0]) * 180 / math.pi),
+        "ditingmotion": _bind(
+            partial(CombinationLoss, losses=[FocalLoss, FocalLoss]),
+            inputs=[["z", "dz"]], labels=["clr", "pmp"], eval_tasks=["pmp"],
+            out_results=lambda xs: [x.softmax(-1) for x in xs]),
+        "seist_.*?_dpk.*": _bind(
+            partial(BCELoss, weight=[[0.5], [1], [1]]),
+            inputs=[_ZNE], labels=[["det", "ppk", "spk"]],
+            eval_tasks=["det", "ppk", "spk"]),
+        "seist_.*?_pmp": _bind(
+            partial(CELoss, weight=[1, 1]),
+            inputs=[_ZNE], labels=["pmp"], eval_tasks=["pmp"]),
+        "seist_.*?_emg": _bind(
+            HuberLoss, inputs=[_ZNE], labels=["emg"], eval_tasks=["emg"]),
+        "seist_.*?_baz": _bind(
+            HuberLoss, inputs=[_ZNE], labels=["baz"], eval_tasks=["baz"]),
+        "seist_.*?_dis": _bind(
+            HuberLoss, inputs=[_ZNE], labels=["dis"], eval_tasks=["dis"]),
     }
 
     _avl_metrics = ("precision", "recall", "f1", "mean", "rmse", "mae",
@@ -151,35 +116,23 @@ class Config:
     _avl_io_item_types = ("soft", "value", "onehot")
 
     _avl_io_items = {
-        "z": {"type": "soft", "metrics": ["mean", "rmse", "mae"]},
-        "n": {"type": "soft", "metrics": ["mean", "rmse", "mae"]},
-        "e": {"type": "soft", "metrics": ["mean", "rmse", "mae"]},
-        "dz": {"type": "soft", "metrics": ["mean", "rmse", "mae"]},
-        "dn": {"type": "soft", "metrics": ["mean", "rmse", "mae"]},
-        "de": {"type": "soft", "metrics": ["mean", "rmse", "mae"]},
-        "non": {"type": "soft", "metrics": []},
-        "det": {"type": "soft", "metrics": ["precision", "recall", "f1"]},
-        "ppk": {"type": "soft",
-                "metrics": ["precision", "recall", "f1", "mean", "rmse",
-                            "mae", "mape"]},
-        "spk": {"type": "soft",
-                "metrics": ["precision", "recall", "f1", "mean", "rmse",
-                            "mae", "mape"]},
-        "ppk+": {"type": "soft", "metrics": []},
-        "spk+": {"type": "soft", "metrics": []},
-        "det+": {"type": "soft", "metrics": []},
-        "ppks": {"type": "value",
-                 "metrics": ["mean", "rmse", "mae", "mape", "r2"]},
-        "spks": {"type": "value",
-                 "metrics": ["mean", "rmse", "mae", "mape", "r2"]},
-        "emg": {"type": "value", "metrics": ["mean", "rmse", "mae", "r2"]},
-        "smg": {"type": "value", "metrics": ["mean", "rmse", "mae", "r2"]},
-        "baz": {"type": "value", "metrics": ["mean", "rmse", "mae", "r2"]},
-        "dis": {"type": "value", "metrics": ["mean", "rmse", "mae", "r2"]},
-        "pmp": {"type": "onehot", "metrics": ["precision", "recall", "f1"],
-                "num_classes": 2},
-        "clr": {"type": "onehot", "metrics": ["precision", "recall", "f1"],
-                "num_classes": 2},
+        **{ch: _io("soft", _WAVE_M) for ch in ("z", "n", "e",
+                                               "dz", "dn", "de")},
+        "non": _io("soft", ()),
+        "det": _io("soft", _DET_M),
+        "ppk": _io("soft", _PICK_M),
+        "spk": _io("soft", _PICK_M),
+        "ppk+": _io("soft", ()),
+        "spk+": _io("soft", ()),
+        "det+": _io("soft", ()),
+        "ppks": _io("value", _IDX_M),
+        "spks": _io("value", _IDX_M),
+        "emg": _io("value", _REG_M),
+        "smg": _io("value", _REG_M),
+        "baz": _io("value", _REG_M),
+        "dis": _io("value", _REG_M),
+        "pmp": _io("onehot", _DET_M, num_classes=2),
+        "clr": _io("onehot", _DET_M, num_classes=2),
     }
 
     # ------------------------------------------------------------------
