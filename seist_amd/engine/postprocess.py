@@ -177,8 +177,13 @@ def _pick_phase(outputs: torch.Tensor, prob_threshold: float,
         if hi <= lo:
             continue
         ind = cols[lo:hi]
-        if min_peak_dist > 1:
-            v = vals[lo:hi]
+        v = vals[lo:hi]
+        if min_peak_dist <= 1:
+            # keep the topk highest peaks (in index order), matching the
+            # suppression branch's height-ordered selection semantics
+            order = np.argsort(v)[::-1][:topk]
+            ind = np.sort(ind[order])
+        else:
             order = np.argsort(v)[::-1][:topk]
             ind = ind[order]
             idel = np.zeros(ind.size, dtype=bool)
@@ -330,6 +335,20 @@ class ResultSaver:
             self._results_dict[pk].extend(self._convert_type(pv))
             tk, tv = self._process_item(k, targets[k], prefix="tgt_")
             self._results_dict[tk].extend(self._convert_type(tv))
+
+    def gather_to_main(self) -> None:
+        """Merge every rank's rows onto rank 0 (row dicts are small python
+        lists; one gather_object at test end, not per step)."""
+        import torch.distributed as dist
+        world = dist.get_world_size()
+        holder = [None] * world if dist.get_rank() == 0 else None
+        dist.gather_object(dict(self._results_dict), holder, dst=0)
+        if holder is not None:
+            merged = defaultdict(list)
+            for shard in holder:
+                for k, v in shard.items():
+                    merged[k].extend(v)
+            self._results_dict = merged
 
     def save_as_csv(self, path: str) -> None:
         import pandas as pd

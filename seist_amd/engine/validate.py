@@ -51,9 +51,11 @@ def validate(args, tasks, model, loss_fn, val_loader, epoch, device,
         len(val_loader), list(average_meters.values()),
         prefix=f"{'Test' if testing else 'Val'}: [{epoch}/{args.epochs}]")
 
+    # Every rank accumulates its shard's rows; they are gathered to rank 0
+    # before the CSV write, so the saved file covers the whole test set
+    # (the reference saves only rank 0's 1/world_size shard).
     results_saver = (ResultSaver(item_names=tasks)
-                     if testing and args.save_test_results
-                     and pdist.is_main_process() else None)
+                     if testing and args.save_test_results else None)
     sync_per_step = pdist.is_dist() and args.sync_metrics_per_step
 
     with torch.no_grad():
@@ -129,9 +131,12 @@ def validate(args, tasks, model, loss_fn, val_loader, epoch, device,
             metrics_merged[task].synchronize_between_processes()
 
     if results_saver is not None:
-        path = get_safe_path(os.path.join(
-            logger.logdir() or ".",
-            f"test_results_{val_loader.dataset.name()}.csv"))
-        results_saver.save_as_csv(path)
+        if pdist.is_dist():
+            results_saver.gather_to_main()
+        if pdist.is_main_process():
+            path = get_safe_path(os.path.join(
+                logger.logdir() or ".",
+                f"test_results_{val_loader.dataset.name()}.csv"))
+            results_saver.save_as_csv(path)
 
     return average_meters["loss"].avg, metrics_merged

@@ -5,9 +5,9 @@ FocalLoss, BinaryFocalLoss, MSELoss, CombinationLoss, MousaviLoss, plus the
 ``HuberLoss`` re-export). All operate on probabilities (post-sigmoid/softmax
 model outputs) like the reference; epsilon = 1e-6.
 
-On MI355X, every one of these lowers to a fused elementwise+reduction via
-``seist_amd.ops`` when the input is a CUDA tensor (memory-bound: a single
-HBM pass, not a chain of eager temporaries).
+These run as eager PyTorch on purpose: at batch 500 the loss is <1% of the
+step (rocprof, profiles/step_profile_r01.md), so the fusion budget is spent
+on the conv/BN/attention chains instead.
 """
 
 from typing import Tuple

@@ -24,6 +24,12 @@ class FusedAdam(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self._packed = None  # list of (meta, sample, has_master, ptr_sig)
 
+    def load_state_dict(self, state_dict):
+        # The packed chunk metadata bakes m/v/master device pointers in; a
+        # restored state has fresh tensors, so force a repack.
+        super().load_state_dict(state_dict)
+        self._packed = None
+
     def _collect(self, group):
         params, grads, ms, vs, masters, steps = [], [], [], [], [], []
         for p in group["params"]:
@@ -36,6 +42,9 @@ class FusedAdam(torch.optim.Optimizer):
                 state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
                 if p.dtype != torch.float32:
                     state["master"] = p.detach().clone().float()
+            elif torch.is_tensor(state["step"]):
+                # torch.optim.Adam checkpoints store step as a 0-dim tensor
+                state["step"] = int(state["step"])
             state["step"] += 1
             params.append(p)
             grads.append(p.grad)
@@ -81,9 +90,11 @@ class FusedAdam(torch.optim.Optimizer):
         parts = {}
         for p, g, m, v, mst in zip(params, grads, ms, vs, masters):
             parts.setdefault(p.dtype, []).append((p, g, m, v, mst))
-        sig = tuple((g.data_ptr(), p.data_ptr())
-                    for p, g, *_ in
-                    [t for lst in parts.values() for t in lst])
+        sig = tuple(
+            (p.data_ptr(), g.data_ptr(), m.data_ptr(), v.data_ptr(),
+             mst.data_ptr() if mst is not None else 0)
+            for p, g, m, v, mst in
+            [t for lst in parts.values() for t in lst])
         if self._packed is None or self._packed[0] != sig:
             packed = []
             for dtype, lst in parts.items():
