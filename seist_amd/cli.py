@@ -49,6 +49,9 @@ def get_args(argv=None):
     # Distributed
     parser.add_argument("--find-unused-parameters", type=bool_, default=False)
     parser.add_argument("--sync-bn", type=bool_, default=True)
+    parser.add_argument("--use-torch-ddp", type=bool_, default=False,
+                        help="use torch DDP + SyncBatchNorm instead of the "
+                             "native flat-bucket replica + fused SyncBN")
     parser.add_argument("--trace-step-time", type=bool_, default=False,
                         help="log a data/h2d/fwd/loss/bwd/opt/comm/"
                              "postprocess/metrics step-time breakdown")

@@ -20,10 +20,15 @@ def run_conv(conv: nn.Conv1d, x, padl: int = 0, padr: int = 0):
 
 
 def run_bn(bn, x, act: str = "none"):
-    """Fused BatchNorm1d (+act); falls through for Identity/other norms."""
+    """Fused BatchNorm1d (+act); falls through for Identity/other norms.
+
+    A BN module tagged ``_sync_bn`` (see parallel.ddp.enable_native_syncbn)
+    reduces its batch statistics across ranks inside the same fused path.
+    """
     if isinstance(bn, nn.BatchNorm1d):
         y = ops.bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
-                       bn.training, bn.momentum, bn.eps, act=act)
+                       bn.training, bn.momentum, bn.eps, act=act,
+                       sync=getattr(bn, "_sync_bn", False))
         if bn.training and bn.track_running_stats:
             bn.num_batches_tracked += 1
         return y

@@ -78,6 +78,7 @@ def _norm(norm_module, x, act: str = "none"):
             norm_module.momentum,
             norm_module.eps,
             act=act,
+            sync=getattr(norm_module, "_sync_bn", False),
         )
         if norm_module.training and norm_module.track_running_stats:
             norm_module.num_batches_tracked += 1

@@ -168,10 +168,64 @@ def conv1d(x: torch.Tensor, weight: torch.Tensor,
 # ---------------------------------------------------------------------------
 
 
+def _sync_world(group):
+    import torch.distributed as dist
+    if not (dist.is_available() and dist.is_initialized()):
+        return None, 1
+    return group, dist.get_world_size(group)
+
+
 class _BNAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, act):
+                momentum, eps, act, sync, group):
+        import torch.distributed as dist
+        world = 1
+        if sync and training:
+            group, world = _sync_world(group)
+        ctx.sync = sync and training and world > 1
+        ctx.group = group
+        ctx.world = world
+        if ctx.sync:
+            # SyncBN: local (sum, sumsq) -> one small RCCL
+            # all-reduce of (C, 2) floats -> finalize with the global count.
+            # Keeps the native fused kernel in play under distributed
+            # training (the reference's convert_sync_batchnorm drops it).
+            n_local = x.size(0) * x.size(2)
+            count = n_local * world
+            if use_native(x):
+                sums = ext().bn_sums_only(x)
+                dist.all_reduce(sums, group=group)
+                y, mean, invstd = ext().bn_act_fwd_from_sums(
+                    x, sums, float(count), gamma, beta, running_mean,
+                    running_var, momentum, eps, act)
+            else:
+                x32 = x.float()
+                sums = torch.stack([x32.sum(dim=(0, 2)),
+                                    (x32 * x32).sum(dim=(0, 2))], dim=1)
+                dist.all_reduce(sums, group=group)
+                mean = sums[:, 0] / count
+                var = (sums[:, 1] / count - mean * mean).clamp_min_(0)
+                if running_mean is not None:
+                    with torch.no_grad():
+                        running_mean.mul_(1 - momentum).add_(momentum * mean)
+                        unbiased = var * (count / max(count - 1, 1))
+                        running_var.mul_(1 - momentum).add_(
+                            momentum * unbiased)
+                invstd = torch.rsqrt(var + eps)
+                xhat = (x32 - mean[:, None]) * invstd[:, None]
+                pre = xhat * gamma.float()[:, None] + beta.float()[:, None]
+                if act == _ACT_GELU:
+                    y = F.gelu(pre)
+                elif act == _ACT_RELU:
+                    y = F.relu(pre)
+                else:
+                    y = pre
+                y = y.to(x.dtype)
+            ctx.save_for_backward(x, gamma, beta, mean, invstd)
+            ctx.training = training
+            ctx.act = act
+            return y
         if use_native(x):
             y, mean, invstd = ext().bn_act_fwd(
                 x, gamma, beta, running_mean, running_var, training, momentum,
@@ -207,8 +261,48 @@ class _BNAct(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
+        import torch.distributed as dist
         x, gamma, beta, mean, invstd = ctx.saved_tensors
         dy = dy.contiguous()
+        if ctx.sync:
+            count = x.size(0) * x.size(2) * ctx.world
+            if use_native(x):
+                local = ext().bn_bwd_sums_only(dy, x, mean, invstd, gamma,
+                                               beta, ctx.act)
+                # dgamma/dbeta stay LOCAL sums (the later gradient
+                # all-reduce averages them, torch SyncBatchNorm parity);
+                # dx needs the GLOBAL correction sums.
+                gsums = local.clone()
+                dist.all_reduce(gsums, group=ctx.group)
+                dx = ext().bn_bwd_dx_from_sums(dy, x, mean, invstd, gamma,
+                                               beta, gsums, float(count),
+                                               ctx.act)
+                dbeta = local[:, 0].to(gamma.dtype)
+                dgamma = local[:, 1].to(gamma.dtype)
+            else:
+                x32 = x.float()
+                dy32 = dy.float()
+                g = gamma.float()[:, None]
+                b = beta.float()[:, None]
+                xhat = (x32 - mean[:, None]) * invstd[:, None]
+                if ctx.act != _ACT_NONE:
+                    pre = xhat * g + b
+                    if ctx.act == _ACT_GELU:
+                        dy32 = dy32 * _gelu_grad(pre)
+                    else:
+                        dy32 = dy32 * (pre > 0).to(dy32.dtype)
+                dbeta_l = dy32.sum(dim=(0, 2))
+                dgamma_l = (dy32 * xhat).sum(dim=(0, 2))
+                gsums = torch.stack([dbeta_l, dgamma_l], dim=1)
+                dist.all_reduce(gsums, group=ctx.group)
+                dx = (g * invstd[:, None] / count) * (
+                    count * dy32 - gsums[:, 0][:, None]
+                    - xhat * gsums[:, 1][:, None])
+                dx = dx.to(x.dtype)
+                dbeta = dbeta_l.to(gamma.dtype)
+                dgamma = dgamma_l.to(gamma.dtype)
+            return (dx, dgamma, dbeta, None, None, None, None, None, None,
+                    None, None)
         if use_native(x):
             dx, dgamma, dbeta = ext().bn_act_bwd(
                 dy, x, gamma, beta, mean, invstd, ctx.training, ctx.act)
@@ -236,21 +330,26 @@ class _BNAct(torch.autograd.Function):
             dx = dx.to(x.dtype)
             dgamma = dgamma.to(gamma.dtype)
             dbeta = dbeta.to(beta.dtype)
-        return dx, dgamma, dbeta, None, None, None, None, None, None
+        return (dx, dgamma, dbeta, None, None, None, None, None, None, None,
+                None)
 
 
 def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
            running_mean: Optional[torch.Tensor],
            running_var: Optional[torch.Tensor], training: bool,
            momentum: float = 0.1, eps: float = 1e-5,
-           act: str = "none") -> torch.Tensor:
+           act: str = "none", sync: bool = False,
+           process_group=None) -> torch.Tensor:
     """Fused BatchNorm1d (+GELU) — K7/K14 of SURVEY §2.4.
 
     BN statistics and parameters are fp32 regardless of activation dtype.
+    With ``sync=True`` under an initialized process group, batch statistics
+    are reduced across ranks (one (C,2)-float collective before finalize,
+    one more in backward) — K7's SyncBN obligation.
     """
     act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
     return _BNAct.apply(x.contiguous(), gamma, beta, running_mean, running_var,
-                        training, momentum, eps, act_id)
+                        training, momentum, eps, act_id, sync, process_group)
 
 
 # ---------------------------------------------------------------------------

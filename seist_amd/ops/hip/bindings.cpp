@@ -35,6 +35,23 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
                                    const at::Tensor& invstd, bool training,
                                    long act);
 
+at::Tensor bn_sums_only(const at::Tensor& x);
+std::vector<at::Tensor> bn_act_fwd_from_sums(
+    const at::Tensor& x, const at::Tensor& sums, double count,
+    const at::Tensor& gamma, const at::Tensor& beta,
+    const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    long act);
+at::Tensor bn_bwd_sums_only(const at::Tensor& dy, const at::Tensor& x,
+                            const at::Tensor& mean, const at::Tensor& invstd,
+                            const at::Tensor& gamma, const at::Tensor& beta,
+                            long act);
+at::Tensor bn_bwd_dx_from_sums(const at::Tensor& dy, const at::Tensor& x,
+                               const at::Tensor& mean,
+                               const at::Tensor& invstd,
+                               const at::Tensor& gamma, const at::Tensor& beta,
+                               const at::Tensor& sums, double count, long act);
+
 std::vector<at::Tensor> avgmax_pool_fwd(const at::Tensor& x, long k);
 at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
                            long k, long in_len);
@@ -82,6 +99,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "transposed conv1d backward");
   m.def("bn_act_fwd", &bn_act_fwd, "fused batchnorm+act forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused batchnorm+act backward");
+  m.def("bn_sums_only", &bn_sums_only, "local BN (sum, sumsq) to (C,2)");
+  m.def("bn_act_fwd_from_sums", &bn_act_fwd_from_sums,
+        "BN+act forward from externally reduced sums (SyncBN)");
+  m.def("bn_bwd_sums_only", &bn_bwd_sums_only,
+        "local BN backward (dbeta, dgamma) sums to (C,2)");
+  m.def("bn_bwd_dx_from_sums", &bn_bwd_dx_from_sums,
+        "BN backward dx from externally reduced sums (SyncBN)");
   m.def("avgmax_pool_fwd", &avgmax_pool_fwd, "fused avg+max pool forward");
   m.def("avgmax_pool_bwd", &avgmax_pool_bwd, "fused avg+max pool backward");
   m.def("interp_linear_fwd", &interp_linear_fwd, "linear interp forward");

@@ -268,6 +268,135 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
   return {y, mean, invstd};
 }
 
+// --- SyncBN split form -----------------------------------
+// The cross-rank form runs the same kernels with an RCCL all-reduce of the
+// tiny (C, 2) sums tensor between the local reduction and the finalize —
+// the native BN kernel stays in use under distributed training (the
+// reference converts to torch SyncBatchNorm and loses its fused path,
+// train.py:374).
+
+at::Tensor bn_sums_only(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long NL = (long)N * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = x.options().dtype(at::kFloat);
+  const int nsplit = pick_nsplit(N, C);
+  auto part = at::empty({C, nsplit, 2}, opts);
+  auto sums = at::empty({C, 2}, opts);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_sums", [&] {
+        hipLaunchKernelGGL((bn_sums_kernel<scalar_t>), dim3(C, nsplit),
+                           dim3(kBlock), 0, stream.stream(),
+                           x.data_ptr<scalar_t>(), part.data_ptr<float>(),
+                           C, NL, L);
+      });
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
+                     dim3(256), 0, stream.stream(), part.data_ptr<float>(),
+                     sums.data_ptr<float>(), C, nsplit);
+  return sums;
+}
+
+std::vector<at::Tensor> bn_act_fwd_from_sums(
+    const at::Tensor& x, const at::Tensor& sums, double count,
+    const at::Tensor& gamma, const at::Tensor& beta,
+    const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    long act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  TORCH_CHECK(sums.is_cuda() && sums.is_contiguous());
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = x.options().dtype(at::kFloat);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  const bool has_running = running_mean.has_value() && running_mean->defined();
+  // sums is (C, 2) == part layout with nsplit == 1
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
+                     0, stream.stream(), sums.data_ptr<float>(), 1,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     has_running ? running_mean->data_ptr<float>() : nullptr,
+                     has_running ? running_var->data_ptr<float>() : nullptr,
+                     C, (long)count, (float)momentum, (float)eps);
+  auto y = at::empty_like(x);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_apply", [&] {
+        hipLaunchKernelGGL((bn_apply_kernel<scalar_t>),
+                           dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                           dim3(kBlock), 0, stream.stream(),
+                           x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           g32.data_ptr<float>(), b32.data_ptr<float>(),
+                           C, L, total, (int)act);
+      });
+  return {y, mean, invstd};
+}
+
+at::Tensor bn_bwd_sums_only(const at::Tensor& dy, const at::Tensor& x,
+                            const at::Tensor& mean, const at::Tensor& invstd,
+                            const at::Tensor& gamma, const at::Tensor& beta,
+                            long act) {
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long NL = (long)N * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = x.options().dtype(at::kFloat);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  const int nsplit = pick_nsplit(N, C);
+  auto part = at::empty({C, nsplit, 2}, opts);
+  auto sums = at::empty({C, 2}, opts);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_bwd_sums", [&] {
+        hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t>), dim3(C, nsplit),
+                           dim3(kBlock), 0, stream.stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           g32.data_ptr<float>(), b32.data_ptr<float>(),
+                           part.data_ptr<float>(), C, NL, L, (int)act);
+      });
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
+                     dim3(256), 0, stream.stream(), part.data_ptr<float>(),
+                     sums.data_ptr<float>(), C, nsplit);
+  return sums;
+}
+
+at::Tensor bn_bwd_dx_from_sums(const at::Tensor& dy, const at::Tensor& x,
+                               const at::Tensor& mean,
+                               const at::Tensor& invstd,
+                               const at::Tensor& gamma, const at::Tensor& beta,
+                               const at::Tensor& sums, double count,
+                               long act) {
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  auto dx = at::empty_like(x);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_bwd_dx", [&] {
+        hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, true>),
+                           dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                           dim3(kBlock), 0, stream.stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           dx.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), g32.data_ptr<float>(),
+                           b32.data_ptr<float>(), sums.data_ptr<float>(),
+                           C, L, total, (long)count, (int)act);
+      });
+  return dx;
+}
+
 std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
                                    const at::Tensor& gamma,
                                    const at::Tensor& beta,

@@ -36,6 +36,18 @@ def wrap_distributed(model: nn.Module, args) -> nn.Module:
     return model
 
 
+def enable_native_syncbn(model: nn.Module, flag: bool = True) -> nn.Module:
+    """Tag every BatchNorm1d so the fused bn_act path reduces batch
+    statistics across ranks (one (C,2)-float collective around the
+    finalize) instead of converting the module tree to
+    torch SyncBatchNorm — which would bypass the native kernel entirely.
+    The module tree and checkpoint format are unchanged."""
+    for m in model.modules():
+        if isinstance(m, nn.BatchNorm1d):
+            m._sync_bn = flag
+    return model
+
+
 class FlatReplica:
     """Flat-bucket gradient replica for one-process-per-GPU training.
 
