@@ -80,8 +80,10 @@ class BiLSTMBlock(nn.Module):
         self.bn = nn.BatchNorm1d(out_channels)
 
     def forward(self, x):
-        # MIOpen LSTM keeps fp32 weights; cast bf16 encoder activations up
-        x, _ = self.bilstm(x.permute(0, 2, 1).float())
+        # LSTM weights stay fp32; cast bf16 encoder activations up.
+        # On GPU the recurrence runs as the persistent K11 kernel (input
+        # projections = one GEMM), not MIOpen's per-step launches.
+        x = ops.lstm(x.permute(0, 2, 1).float().contiguous(), self.bilstm)
         x = self.dropout(x).permute(0, 2, 1)
         x = run_conv(self.conv, x)
         return run_bn(self.bn, x)
@@ -107,20 +109,11 @@ class AttentionLayer(nn.Module):
 
     def forward(self, x):
         x = x.permute(0, 2, 1)                       # (N,L,C)
-        q = torch.matmul(x, self.Wt).unsqueeze(2)    # (N,L,1,d)
-        k = torch.matmul(x, self.Wx).unsqueeze(1)    # (N,1,L,d)
-        h = torch.tanh(q + k + self.bh)              # (N,L,L,d)
-        # (h @ Wa) is a K=d->1 GEMV over N*L*L rows — rocBLAS runs it ~100x
-        # off the memory roofline; an elementwise multiply-reduce is fast
-        e = (h * self.Wa.view(1, 1, 1, -1)).sum(-1) + self.ba
-        e = torch.exp(e - torch.max(e, dim=-1, keepdim=True).values)
-        if self.attn_width is not None:
-            mask = (torch.ones(e.shape[-2:], dtype=torch.bool, device=e.device)
-                    .tril(self.attn_width // 2 - 1)
-                    .triu(-self.attn_width // 2))
-            e = e.where(mask, torch.zeros((), dtype=e.dtype, device=e.device))
-        s = torch.sum(e, dim=-1, keepdim=True)
-        a = e / (s + _EPS)
+        q = torch.matmul(x, self.Wt)                 # (N,L,d)
+        k = torch.matmul(x, self.Wx)                 # (N,L,d)
+        # fused K10 kernel on GPU: the (N,L,L,d) tanh tensor never exists
+        a = ops.additive_attention_weights(q, k, self.bh, self.Wa, self.ba,
+                                           self.attn_width)
         v = torch.matmul(a, x).permute(0, 2, 1)
         return v, a
 
@@ -154,8 +147,9 @@ class TransformerLayer(nn.Module):
         x = x.float()  # attention/LayerNorm stage stays fp32 (tiny, L=64)
         x1, w = self.attn(x)
         x2 = (x1 + x).permute(0, 2, 1)
-        x2 = self.ln0(x2)
-        x4 = self.ln1(self.ff(x2) + x2).permute(0, 2, 1)
+        x2 = ops.layer_norm(x2, self.ln0.weight, self.ln0.bias, self.ln0.eps)
+        x4 = ops.layer_norm(self.ff(x2) + x2, self.ln1.weight, self.ln1.bias,
+                            self.ln1.eps).permute(0, 2, 1)
         return x4, w
 
 
@@ -262,8 +256,11 @@ class Decoder(nn.Module):
                                   padding=5)
 
     def forward(self, x):
-        x = x.permute(0, 2, 1).float()  # decoder LSTM stays fp32
-        x, _ = self.lstm(x)
+        x = x.permute(0, 2, 1).float().contiguous()  # decoder LSTM stays fp32
+        if isinstance(self.lstm, nn.LSTM):
+            x = ops.lstm(x, self.lstm)
+        else:
+            x, _ = self.lstm(x)
         x = self.lstm_dropout(x).permute(0, 2, 1)
         x, _ = self.transformer(x)
         x = self.upsamplings(x)

@@ -5,6 +5,7 @@ one bidirectional LSTM (hidden 100), linear head emitting (mag, log-var)
 for MousaviLoss.
 """
 
+import torch
 import torch.nn as nn
 
 from .. import ops
@@ -44,8 +45,16 @@ class MagNet(nn.Module):
 
     def forward(self, x):
         x = self.conv_layers(x)
-        hs, (h, c) = self.lstm(x.transpose(-1, -2).float())
-        h = h.transpose(0, 1).flatten(1)
+        x = x.transpose(-1, -2).float().contiguous()
+        if x.is_cuda:
+            # K11 persistent recurrence; h_n per direction is the last
+            # processed step: t = L-1 forward, t = 0 reverse
+            y = ops.lstm(x, self.lstm)
+            H = self.lstm.hidden_size
+            h = torch.cat([y[:, -1, :H], y[:, 0, H:]], dim=-1)
+        else:
+            hs, (h, c) = self.lstm(x)
+            h = h.transpose(0, 1).flatten(1)
         return self.lin(h)
 
 

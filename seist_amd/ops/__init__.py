@@ -63,4 +63,9 @@ from .functional import (  # noqa: E402,F401
     pooled_attention,
     upsample2x,
 )
+from .eqt import (  # noqa: E402,F401
+    additive_attention_weights,
+    layer_norm,
+    lstm,
+)
 from .adam import FusedAdam  # noqa: E402,F401

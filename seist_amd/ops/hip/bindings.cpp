@@ -72,6 +72,25 @@ at::Tensor pooled_attn_fwd(const at::Tensor& q, const at::Tensor& k,
                            const at::Tensor& v);
 at::Tensor interp_linear_bwd(const at::Tensor& dy, long in_len);
 
+std::vector<at::Tensor> ln_fwd(const at::Tensor& x, const at::Tensor& gamma,
+                               const at::Tensor& beta, double eps);
+std::vector<at::Tensor> ln_bwd(const at::Tensor& dy, const at::Tensor& x,
+                               const at::Tensor& gamma,
+                               const at::Tensor& mean, const at::Tensor& rstd);
+std::vector<at::Tensor> addattn_fwd(const at::Tensor& q, const at::Tensor& k,
+                                    const at::Tensor& bh, const at::Tensor& wa,
+                                    double ba, long tril_k, long triu_k);
+std::vector<at::Tensor> addattn_bwd(const at::Tensor& q, const at::Tensor& k,
+                                    const at::Tensor& bh, const at::Tensor& wa,
+                                    const at::Tensor& attn,
+                                    const at::Tensor& dattn,
+                                    const at::Tensor& amax);
+std::vector<at::Tensor> lstm_fwd(const at::Tensor& pre, const at::Tensor& whh,
+                                 at::Tensor y, long dir, bool training);
+at::Tensor lstm_bwd(const at::Tensor& dy, const at::Tensor& y,
+                    const at::Tensor& cstash, const at::Tensor& gstash,
+                    const at::Tensor& whh, long dirs, long dir);
+
 at::Tensor adam_pack(std::vector<at::Tensor> params,
                      std::vector<at::Tensor> grads,
                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
@@ -118,6 +137,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused pooled-KV attention training forward (stats + packed mask)");
   m.def("pooled_attn_bwd", &pooled_attn_bwd,
         "fused pooled-KV attention backward (dq, dk, dv)");
+  m.def("ln_fwd", &ln_fwd, "LayerNorm forward (channel-last rows)");
+  m.def("ln_bwd", &ln_bwd, "LayerNorm backward");
+  m.def("addattn_fwd", &addattn_fwd,
+        "fused additive attention scores+softmax (EQT K10)");
+  m.def("addattn_bwd", &addattn_bwd, "additive attention backward");
+  m.def("lstm_fwd", &lstm_fwd, "LSTM recurrence forward (persistent)");
+  m.def("lstm_bwd", &lstm_bwd, "LSTM recurrence backward (BPTT)");
   m.def("adam_pack", &adam_pack, "pack fused-adam chunk metadata");
   m.def("sum_batch", &sum_batch, "batch-axis sum to fp32");
   m.def("channel_sum", &channel_sum, "per-channel sum to fp32");

@@ -44,6 +44,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "conv_tap.hip"),
         os.path.join(HIP_DIR, "rowscale.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
+        os.path.join(HIP_DIR, "eqt.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
