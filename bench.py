@@ -136,9 +136,7 @@ def main():
     # fused Adam), so the 8-rank step replays ~2 graphs + 1 collective
     # instead of ~4k eager dispatches.
     graphed = False
-    has_rnn = any(isinstance(m, (torch.nn.LSTM, torch.nn.GRU))
-                  for m in model.modules())
-    if args.graph and use_cuda and not has_rnn:
+    if args.graph and use_cuda:
         try:
             stream = torch.cuda.Stream()
             with torch.cuda.stream(stream):

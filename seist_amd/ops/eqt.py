@@ -137,9 +137,9 @@ class _LSTMDir(torch.autograd.Function):
     (N, L, 4H); writes its H-slice of the shared output y in place."""
 
     @staticmethod
-    def forward(ctx, pre, whh, y, dir_):
-        training = torch.is_grad_enabled() and (
-            pre.requires_grad or whh.requires_grad)
+    def forward(ctx, pre, whh, y, dir_, training):
+        # grad mode is disabled inside Function.forward, so the caller
+        # decides whether the backward stashes are needed
         out = ext().lstm_fwd(pre, whh, y, dir_, training)
         if training:
             cstash, gstash = out
@@ -180,7 +180,7 @@ class _LSTMDir(torch.autograd.Function):
         # incoming values contribute nothing
         grad_y = dy.clone()
         grad_y[:, :, dir_ * H:(dir_ + 1) * H] = 0
-        return dgates, dwhh, grad_y, None
+        return dgates, dwhh, grad_y, None, None
 
 
 def lstm(x: torch.Tensor, module: torch.nn.LSTM) -> torch.Tensor:
@@ -197,6 +197,9 @@ def lstm(x: torch.Tensor, module: torch.nn.LSTM) -> torch.Tensor:
     x = x.contiguous()
     N, L, _ = x.shape
     y = x.new_empty(N, L, dirs * H)
+    training = torch.is_grad_enabled() and (
+        x.requires_grad
+        or any(p.requires_grad for p in module.parameters()))
     for dir_ in range(dirs):
         sfx = "_reverse" if dir_ == 1 else ""
         w_ih = getattr(module, f"weight_ih_l0{sfx}")
@@ -206,5 +209,6 @@ def lstm(x: torch.Tensor, module: torch.nn.LSTM) -> torch.Tensor:
             b_ih = getattr(module, f"bias_ih_l0{sfx}")
             b_hh = getattr(module, f"bias_hh_l0{sfx}")
             pre = pre + (b_ih + b_hh)
-        y = _LSTMDir.apply(pre.contiguous(), w_hh.contiguous(), y, dir_)
+        y = _LSTMDir.apply(pre.contiguous(), w_hh.contiguous(), y, dir_,
+                           training)
     return y
