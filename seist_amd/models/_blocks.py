@@ -19,16 +19,18 @@ def run_conv(conv: nn.Conv1d, x, padl: int = 0, padr: int = 0):
     )
 
 
-def run_bn(bn, x, act: str = "none"):
+def run_bn(bn, x, act: str = "none", part=None):
     """Fused BatchNorm1d (+act); falls through for Identity/other norms.
 
     A BN module tagged ``_sync_bn`` (see parallel.ddp.enable_native_syncbn)
     reduces its batch statistics across ranks inside the same fused path.
+    ``part`` is an optional producer-collected (C, nsplit, 2) partial-sums
+    slab (conv->BN fusion step 1).
     """
     if isinstance(bn, nn.BatchNorm1d):
         y = ops.bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                        bn.training, bn.momentum, bn.eps, act=act,
-                       sync=getattr(bn, "_sync_bn", False))
+                       sync=getattr(bn, "_sync_bn", False), part=part)
         if bn.training and bn.track_running_stats:
             bn.num_batches_tracked += 1
         return y
@@ -38,3 +40,29 @@ def run_bn(bn, x, act: str = "none"):
     elif act == "gelu":
         y = ops.gelu(y)
     return y
+
+
+def run_conv_bn(conv: nn.Conv1d, bn, x, act: str = "none", padl: int = 0,
+                padr: int = 0, auto_pad: bool = False):
+    """conv -> BatchNorm1d(+act) chain with the BN statistics accumulated
+    in the conv kernel's epilogue (docs/FUSION_PLAN.md step 1): the
+    standalone bn_sums pass over the conv output disappears."""
+    k = conv.kernel_size[0]
+    s = conv.stride[0]
+    if auto_pad:
+        pl, pr = ops.auto_pad_lr(x.size(-1), k, s)
+    else:
+        p = conv.padding[0] if isinstance(conv.padding, tuple) \
+            else int(conv.padding)
+        pl, pr = padl + p, padr + p
+    collect = (isinstance(bn, nn.BatchNorm1d) and bn.training and x.is_cuda)
+    if collect:
+        y, part = ops.conv1d_stats(x, conv.weight, conv.bias, stride=s,
+                                   padding=(pl, pr), groups=conv.groups,
+                                   dilation=conv.dilation[0])
+    else:
+        y = ops.conv1d(x, conv.weight, conv.bias, stride=s,
+                       padding=(pl, pr), groups=conv.groups,
+                       dilation=conv.dilation[0])
+        part = None
+    return run_bn(bn, y, act=act, part=part)

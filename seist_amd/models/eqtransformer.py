@@ -17,7 +17,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from .. import ops
-from ._blocks import run_bn, run_conv
+from ._blocks import run_bn, run_conv, run_conv_bn
 from ._registry import register_model
 
 _EPS = 1e-6
@@ -64,8 +64,10 @@ class ResConvBlock(nn.Module):
 
     def forward(self, x):
         x1 = self.dropout0(run_bn(self.bn0, x, act="relu"))
-        x1 = run_conv(self.conv0, x1, *self.conv_padding_same)
-        x1 = self.dropout1(run_bn(self.bn1, x1, act="relu"))
+        x1 = run_conv_bn(self.conv0, self.bn1, x1, act="relu",
+                         padl=self.conv_padding_same[0],
+                         padr=self.conv_padding_same[1])
+        x1 = self.dropout1(x1)
         x1 = run_conv(self.conv1, x1, *self.conv_padding_same)
         return x + x1
 

@@ -11,7 +11,7 @@ import torch.nn as nn
 import torch.nn.functional as F  # noqa: F401
 
 from .. import ops
-from ._blocks import run_bn, run_conv
+from ._blocks import run_bn, run_conv, run_conv_bn
 from ._registry import register_model
 
 
@@ -43,11 +43,12 @@ class ConvBlock(nn.Module):
         p = (self.stride - (x.size(-1) % self.stride)) % self.stride \
             + self.kernel_padding
         if isinstance(self.conv0, nn.Conv1d):
-            x = run_conv(self.conv0, x, p // 2, p - p // 2)
-            x = run_bn(self.bn0, x, act="relu")
+            x = run_conv_bn(self.conv0, self.bn0, x, act="relu",
+                            padl=p // 2, padr=p - p // 2)
             x = self.drop0(x)
-        x = run_conv(self.conv1, x, *self.conv_padding_same)
-        x = run_bn(self.bn1, x, act="relu")
+        x = run_conv_bn(self.conv1, self.bn1, x, act="relu",
+                        padl=self.conv_padding_same[0],
+                        padr=self.conv_padding_same[1])
         return self.drop1(x)
 
 
@@ -75,8 +76,9 @@ class ConvTransBlock(nn.Module):
 
     def forward(self, x):
         if isinstance(self.conv0, nn.Conv1d):
-            x = run_conv(self.conv0, x, *self.conv_padding_same)
-            x = run_bn(self.bn0, x, act="relu")
+            x = run_conv_bn(self.conv0, self.bn0, x, act="relu",
+                            padl=self.conv_padding_same[0],
+                            padr=self.conv_padding_same[1])
         x = self.drop0(x)
         if isinstance(self.convt, nn.ConvTranspose1d):
             x = ops.conv_transpose1d(x, self.convt.weight, self.convt.bias,

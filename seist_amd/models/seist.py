@@ -20,6 +20,7 @@ import torch.nn as nn
 
 from .. import ops
 from ..ops.functional import auto_pad_lr
+from ._blocks import run_conv_bn
 from ._registry import register_model
 
 __all__ = ["SeismogramTransformer"]
@@ -65,7 +66,7 @@ def _is_gelu(act_module) -> bool:
     return isinstance(act_module, nn.GELU)
 
 
-def _norm(norm_module, x, act: str = "none"):
+def _norm(norm_module, x, act: str = "none", part=None):
     """Apply a norm module (+fused activation when it is a BatchNorm1d)."""
     if isinstance(norm_module, nn.BatchNorm1d):
         y = ops.bn_act(
@@ -79,6 +80,7 @@ def _norm(norm_module, x, act: str = "none"):
             norm_module.eps,
             act=act,
             sync=getattr(norm_module, "_sync_bn", False),
+            part=part,
         )
         if norm_module.training and norm_module.track_running_stats:
             norm_module.num_batches_tracked += 1
@@ -123,8 +125,7 @@ class LocalAwareAggregationBlock(nn.Module):
     def forward(self, x):
         if self.avg_pool is not None:
             x = ops.avgmax_pool1d(x, self.kernel_size)
-        x = ops.pointwise_conv(x, self.proj.weight, self.proj.bias)
-        return _norm(self.norm, x)
+        return run_conv_bn(self.proj, self.norm, x)
 
 
 class MLP(nn.Module):
@@ -160,8 +161,8 @@ class DSConvNormAct(nn.Module):
     def forward(self, x):
         x = ops.pointwise_conv(x, self.in_proj.weight, self.in_proj.bias)
         x = _conv(self.dconv, x, auto_pad=True)
-        x = ops.pointwise_conv(x, self.pconv.weight, self.pconv.bias)
-        return _norm(self.norm, x, act="gelu" if _is_gelu(self.act) else "none")
+        return run_conv_bn(self.pconv, self.norm, x,
+                           act="gelu" if _is_gelu(self.act) else "none")
 
 
 class StemBlock(nn.Module):
@@ -181,8 +182,7 @@ class StemBlock(nn.Module):
 
     def forward(self, x):
         x = torch.cat([conv(x) for conv in self.convs], dim=1)
-        x = ops.pointwise_conv(x, self.out_proj.weight, self.out_proj.bias)
-        return _norm(self.norm, x)
+        return run_conv_bn(self.out_proj, self.norm, x)
 
 
 class GroupConvBlock(nn.Module):
@@ -203,8 +203,8 @@ class GroupConvBlock(nn.Module):
         self.droppath1 = DropPath(path_drop_rate)
 
     def forward(self, x):
-        y = _conv(self.conv, x, auto_pad=True)
-        y = _norm(self.norm0, y, act="gelu" if _is_gelu(self.act) else "none")
+        y = run_conv_bn(self.conv, self.norm0, x, auto_pad=True,
+                        act="gelu" if _is_gelu(self.act) else "none")
         y = ops.pointwise_conv(y, self.proj.weight, self.proj.bias)
         x = ops.droppath_add(x, y, self.droppath0.drop_prob, self.training)
         y = _norm(self.norm1, x)
@@ -244,7 +244,7 @@ class MultiScaleMixedConv(nn.Module):
     def forward(self, x):
         outs = []
         for proj, norm, conv in zip(self.projs, self.norms, self.convs):
-            xi = _norm(norm, ops.pointwise_conv(x, proj.weight, proj.bias))
+            xi = run_conv_bn(proj, norm, x)
             outs.append(xi + conv(xi))
         x = torch.cat(outs, dim=1)
         return _norm(self.out_norm, x)
@@ -354,15 +354,13 @@ class MultiPathTransformerLayer(nn.Module):
     def forward(self, x):
         outs = []
         if self.has_attn:
-            x1 = _norm(self.norm0, ops.pointwise_conv(
-                x, self.attn_proj.weight, self.attn_proj.bias))
+            x1 = run_conv_bn(self.attn_proj, self.norm0, x)
             x1 = ops.droppath_add(x1, self.attention(x1),
                                   self.attn_droppath.drop_prob,
                                   self.training)
             outs.append(x1)
         if self.has_conv:
-            x2 = _norm(self.norm1, ops.pointwise_conv(
-                x, self.conv_proj.weight, self.conv_proj.bias))
+            x2 = run_conv_bn(self.conv_proj, self.norm1, x)
             x2 = ops.droppath_add(x2, self.gconv(x2),
                                   self.gconv_droppath.drop_prob,
                                   self.training)
@@ -408,9 +406,8 @@ class HeadDetectionPicking(nn.Module):
         up_sizes = self._upsampling_sizes(x.size(-1), x0.size(-1))
         for i, layer in enumerate(self.up_layers):
             x = ops.interp_linear(x, up_sizes[i])
-            x = _conv(layer.conv, x, auto_pad=True)
-            x = _norm(layer.norm, x,
-                      act="gelu" if _is_gelu(layer.act) else "none")
+            x = run_conv_bn(layer.conv, layer.norm, x, auto_pad=True,
+                            act="gelu" if _is_gelu(layer.act) else "none")
         x = _conv(self.out_conv, x)
         return self.out_act(x)
 

@@ -52,6 +52,9 @@ def use_native(x: torch.Tensor) -> bool:
 
 from .functional import (  # noqa: E402,F401
     auto_pad,
+    auto_pad_lr,
+    conv1d_stats,
+    pointwise_conv_stats,
     avgmax_pool1d,
     droppath_add,
     bn_act,

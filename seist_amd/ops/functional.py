@@ -104,6 +104,74 @@ def pointwise_conv(x: torch.Tensor, weight: torch.Tensor,
     return _PointwiseConv.apply(x.contiguous(), weight.contiguous(), bias)
 
 
+class _PointwiseConvStats(torch.autograd.Function):
+    """Forward also returns the (Co, nsplit, 2) BN partial-sums slab,
+    reduced in the conv epilogue (fusion step 1). The slab is
+    non-differentiable auxiliary output."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        y, part = ext().pw_conv_fwd_stats(x, weight, bias)
+        ctx.mark_non_differentiable(part)
+        return y, part
+
+    @staticmethod
+    def backward(ctx, dy, dpart):
+        x, weight = ctx.saved_tensors
+        dx, dw, db = ext().pw_conv_bwd(dy.contiguous(), x, weight,
+                                       ctx.has_bias)
+        return dx, dw, db
+
+
+def pointwise_conv_stats(x, weight, bias=None):
+    """pointwise_conv + BN stats partials; (y, None) off-GPU."""
+    if weight.dim() == 3:
+        weight = weight.squeeze(-1)
+    if x.dtype != weight.dtype:
+        x = x.to(weight.dtype)
+    if not use_native(x):
+        return pointwise_conv(x, weight, bias), None
+    return _PointwiseConvStats.apply(x.contiguous(), weight.contiguous(),
+                                     bias)
+
+
+class _Conv1dStats(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padl, padr, groups, dilation):
+        ctx.save_for_backward(x, weight)
+        ctx.conf = (stride, padl, padr, groups, dilation, bias is not None)
+        y, part = ext().conv1d_fwd_stats(x, weight, bias, stride, padl,
+                                         padr, groups, dilation)
+        ctx.mark_non_differentiable(part)
+        return y, part
+
+    @staticmethod
+    def backward(ctx, dy, dpart):
+        x, weight = ctx.saved_tensors
+        stride, padl, padr, groups, dilation, has_bias = ctx.conf
+        dx, dw, db = ext().conv1d_bwd(dy.contiguous(), x, weight, stride,
+                                      padl, padr, groups, dilation, has_bias)
+        return dx, dw, db, None, None, None, None, None
+
+
+def conv1d_stats(x, weight, bias=None, stride=1, padding=(0, 0), groups=1,
+                 dilation=1):
+    """conv1d + BN stats partials; (y, None) off-GPU."""
+    padl, padr = padding
+    if x.dtype != weight.dtype:
+        x = x.to(weight.dtype)
+    if not use_native(x):
+        return conv1d(x, weight, bias, stride, padding, groups,
+                      dilation), None
+    if (weight.size(-1) == 1 and stride == 1 and padl == 0 and padr == 0
+            and groups == 1):
+        return pointwise_conv_stats(x, weight, bias)
+    return _Conv1dStats.apply(x.contiguous(), weight.contiguous(), bias,
+                              stride, padl, padr, groups, dilation)
+
+
 # ---------------------------------------------------------------------------
 # general direct conv1d (depthwise / grouped / dense, strided, pre-padded)
 # ---------------------------------------------------------------------------
@@ -178,7 +246,7 @@ def _sync_world(group):
 class _BNAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var, training,
-                momentum, eps, act, sync, group):
+                momentum, eps, act, sync, group, part):
         import torch.distributed as dist
         world = 1
         if sync and training:
@@ -194,7 +262,8 @@ class _BNAct(torch.autograd.Function):
             n_local = x.size(0) * x.size(2)
             count = n_local * world
             if use_native(x):
-                sums = ext().bn_sums_only(x)
+                sums = (ext().bn_part_to_sums(part) if part is not None
+                        else ext().bn_sums_only(x))
                 dist.all_reduce(sums, group=group)
                 y, mean, invstd = ext().bn_act_fwd_from_sums(
                     x, sums, float(count), gamma, beta, running_mean,
@@ -227,9 +296,16 @@ class _BNAct(torch.autograd.Function):
             ctx.act = act
             return y
         if use_native(x):
-            y, mean, invstd = ext().bn_act_fwd(
-                x, gamma, beta, running_mean, running_var, training, momentum,
-                eps, act)
+            if part is not None and training:
+                # producer-collected partials (fusion step 1): finalize
+                # directly, no bn_sums pass over x
+                y, mean, invstd = ext().bn_act_fwd_with_part(
+                    x, part, gamma, beta, running_mean, running_var,
+                    momentum, eps, act)
+            else:
+                y, mean, invstd = ext().bn_act_fwd(
+                    x, gamma, beta, running_mean, running_var, training,
+                    momentum, eps, act)
         else:
             x32 = x.float()
             if training:
@@ -302,7 +378,7 @@ class _BNAct(torch.autograd.Function):
                 dbeta = dbeta_l.to(gamma.dtype)
                 dgamma = dgamma_l.to(gamma.dtype)
             return (dx, dgamma, dbeta, None, None, None, None, None, None,
-                    None, None)
+                    None, None, None)
         if use_native(x):
             dx, dgamma, dbeta = ext().bn_act_bwd(
                 dy, x, gamma, beta, mean, invstd, ctx.training, ctx.act)
@@ -331,7 +407,7 @@ class _BNAct(torch.autograd.Function):
             dgamma = dgamma.to(gamma.dtype)
             dbeta = dbeta.to(beta.dtype)
         return (dx, dgamma, dbeta, None, None, None, None, None, None, None,
-                None)
+                None, None)
 
 
 def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
@@ -339,7 +415,8 @@ def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
            running_var: Optional[torch.Tensor], training: bool,
            momentum: float = 0.1, eps: float = 1e-5,
            act: str = "none", sync: bool = False,
-           process_group=None) -> torch.Tensor:
+           process_group=None, part: Optional[torch.Tensor] = None
+           ) -> torch.Tensor:
     """Fused BatchNorm1d (+GELU) — K7/K14 of SURVEY §2.4.
 
     BN statistics and parameters are fp32 regardless of activation dtype.
@@ -349,7 +426,8 @@ def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
     """
     act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
     return _BNAct.apply(x.contiguous(), gamma, beta, running_mean, running_var,
-                        training, momentum, eps, act_id, sync, process_group)
+                        training, momentum, eps, act_id, sync, process_group,
+                        part)
 
 
 # ---------------------------------------------------------------------------

@@ -36,6 +36,19 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
                                    long act);
 
 at::Tensor bn_sums_only(const at::Tensor& x);
+std::vector<at::Tensor> pw_conv_fwd_stats(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& bias);
+std::vector<at::Tensor> conv1d_fwd_stats(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& bias, long stride, long padl,
+    long padr, long groups, long dilation);
+std::vector<at::Tensor> bn_act_fwd_with_part(
+    const at::Tensor& x, const at::Tensor& part, const at::Tensor& gamma,
+    const at::Tensor& beta, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    long act);
+at::Tensor bn_part_to_sums(const at::Tensor& part);
 std::vector<at::Tensor> bn_act_fwd_from_sums(
     const at::Tensor& x, const at::Tensor& sums, double count,
     const at::Tensor& gamma, const at::Tensor& beta,
@@ -119,6 +132,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd, "fused batchnorm+act forward");
   m.def("bn_act_bwd", &bn_act_bwd, "fused batchnorm+act backward");
   m.def("bn_sums_only", &bn_sums_only, "local BN (sum, sumsq) to (C,2)");
+  m.def("pw_conv_fwd_stats", &pw_conv_fwd_stats,
+        "pointwise conv forward + BN stats partials (fusion step 1)");
+  m.def("conv1d_fwd_stats", &conv1d_fwd_stats,
+        "conv1d forward + BN stats partials (fusion step 1)");
+  m.def("bn_act_fwd_with_part", &bn_act_fwd_with_part,
+        "BN+act forward from producer-collected partials");
+  m.def("bn_part_to_sums", &bn_part_to_sums,
+        "(C,nsplit,2) partial slab -> (C,2) sums");
   m.def("bn_act_fwd_from_sums", &bn_act_fwd_from_sums,
         "BN+act forward from externally reduced sums (SyncBN)");
   m.def("bn_bwd_sums_only", &bn_bwd_sums_only,

@@ -339,6 +339,62 @@ std::vector<at::Tensor> bn_act_fwd_from_sums(
   return {y, mean, invstd};
 }
 
+// fwd consuming a producer-collected (C, nsplit, 2) partial slab
+// (fusion step 1): no bn_sums pass over x.
+std::vector<at::Tensor> bn_act_fwd_with_part(
+    const at::Tensor& x, const at::Tensor& part, const at::Tensor& gamma,
+    const at::Tensor& beta, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum, double eps,
+    long act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  TORCH_CHECK(part.is_contiguous() && part.dim() == 3
+              && part.size(0) == x.size(1) && part.size(2) == 2);
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long NL = (long)N * L;
+  const long total = (long)N * C * L;
+  const int nsplit = part.size(1);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = x.options().dtype(at::kFloat);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  const bool has_running = running_mean.has_value() && running_mean->defined();
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
+                     0, stream.stream(), part.data_ptr<float>(), nsplit,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     has_running ? running_mean->data_ptr<float>() : nullptr,
+                     has_running ? running_var->data_ptr<float>() : nullptr,
+                     C, NL, (float)momentum, (float)eps);
+  auto y = at::empty_like(x);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_apply", [&] {
+        hipLaunchKernelGGL((bn_apply_kernel<scalar_t>),
+                           dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                           dim3(kBlock), 0, stream.stream(),
+                           x.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           g32.data_ptr<float>(), b32.data_ptr<float>(),
+                           C, L, total, (int)act);
+      });
+  return {y, mean, invstd};
+}
+
+// reduce a (C, nsplit, 2) slab to (C, 2) — used by the SyncBN path to
+// all-reduce producer-collected partials
+at::Tensor bn_part_to_sums(const at::Tensor& part) {
+  const int C = part.size(0);
+  const int nsplit = part.size(1);
+  auto sums = at::empty({C, 2}, part.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
+                     dim3(256), 0, stream.stream(), part.data_ptr<float>(),
+                     sums.data_ptr<float>(), C, nsplit);
+  return sums;
+}
+
 at::Tensor bn_bwd_sums_only(const at::Tensor& dy, const at::Tensor& x,
                             const at::Tensor& mean, const at::Tensor& invstd,
                             const at::Tensor& gamma, const at::Tensor& beta,

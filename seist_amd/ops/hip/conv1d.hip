@@ -26,19 +26,22 @@ at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor sum_mid(const at::Tensor& in);
 bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
-               long padl, long dilation, bool is_dx);
+               long padl, long dilation, bool is_dx,
+               at::Tensor* stats_out = nullptr);
 at::Tensor channel_sum(const at::Tensor& in);
+at::Tensor bn_sums_only(const at::Tensor& x);
 c10::optional<at::Tensor> dw_mfma_try(const at::Tensor& dy,
                                       const at::Tensor& x, long stride,
                                       long padl, long groups, long dilation,
                                       int K);
 bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
                    const c10::optional<at::Tensor>& bias, at::Tensor& y,
-                   long padl, long dilation, long groups, bool is_dx);
+                   long padl, long dilation, long groups, bool is_dx,
+                   at::Tensor* stats_out = nullptr);
 bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias, at::Tensor& y,
                      long stride, long padl, long dilation, long groups,
-                     bool is_dx);
+                     bool is_dx, at::Tensor* stats_out = nullptr);
 
 namespace {
 
@@ -454,6 +457,39 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
         }
       });
   return y;
+}
+
+// forward + per-out-channel (sum, sumsq) BN partials (fusion step 1);
+// falls back to conv1d_fwd + a bn_sums pass outside the MFMA envelope.
+std::vector<at::Tensor> conv1d_fwd_stats(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& bias, long stride, long padl,
+    long padr, long groups, long dilation) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
+  const int N = x.size(0);
+  const long L = x.size(2);
+  const int Co = w.size(0), K = w.size(2);
+  const long Lp = L + padl + padr;
+  const long Lo = (Lp - ((long)K - 1) * dilation - 1) / stride + 1;
+  TORCH_CHECK(Lo > 0, "empty conv output");
+  auto y = at::empty({N, Co, Lo}, x.options());
+  at::Tensor part;
+  if (stride == 1
+      && conv_tap_mfma(x, w, bias, y, padl, dilation, groups,
+                       /*is_dx=*/false, &part)) {
+    return {y, part};
+  }
+  if (conv_tap_s_mfma(x, w, bias, y, stride, padl, dilation, groups,
+                      /*is_dx=*/false, &part)) {
+    return {y, part};
+  }
+  if (groups == 1 && stride == 1
+      && conv_mfma(x, w, bias, y, padl, dilation, /*is_dx=*/false, &part)) {
+    return {y, part};
+  }
+  y = conv1d_fwd(x, w, bias, stride, padl, padr, groups, dilation);
+  part = bn_sums_only(y).view({Co, 1, 2});
+  return {y, part};
 }
 
 // Gather pass shared by the conv input-gradient and the transposed-conv

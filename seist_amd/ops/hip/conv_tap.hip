@@ -46,12 +46,14 @@ void conv_tap_kernel(const sa_bf16* __restrict__ x,
                      const sa_bf16* __restrict__ w,
                      const sa_bf16* __restrict__ bias,
                      sa_bf16* __restrict__ y,
+                     float* __restrict__ stats,
                      int N, int Cin, int Cout, long Lin, long Lout,
                      int K, int padl, int dil, int G, int xext,
                      int xpitch) {
   extern __shared__ sa_bf16 smem[];
   sa_bf16* w_s = smem;                    // [16][K][kCT] (c contiguous)
   sa_bf16* x_s = smem + 16 * K * kCT;     // [kCT][xpitch]
+  __shared__ float stats_s[16 * 2];
 
   const int n = blockIdx.y;
   const int Cg_out = Cout / G;  // out channels per group (m axis)
@@ -165,6 +167,8 @@ void conv_tap_kernel(const sa_bf16* __restrict__ x,
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
   const int Cm = Cout;
+  float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+  float ssum2[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int nrep = 0; nrep < 4; ++nrep) {
     const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
@@ -177,7 +181,39 @@ void conv_tap_kernel(const sa_bf16* __restrict__ x,
       if (mg < Cm) {
         float v = acc[r];
         if (HAS_BIAS) v += (float)bias[mg];
-        y[((long)n * Cm + mg) * Lout + lg] = (sa_bf16)v;
+        const sa_bf16 vb = (sa_bf16)v;
+        y[((long)n * Cm + mg) * Lout + lg] = vb;
+        if (stats != nullptr) {
+          const float vf = (float)vb;
+          ssum[r] += vf;
+          ssum2[r] += vf * vf;
+        }
+      }
+    }
+  }
+  if (stats != nullptr) {
+    __syncthreads();
+    for (int t = tid; t < 16 * 2; t += kBlock) stats_s[t] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int b = 1; b < 16; b <<= 1) {
+        ssum[r] += __shfl_xor(ssum[r], b, sa::kWave);
+        ssum2[r] += __shfl_xor(ssum2[r], b, sa::kWave);
+      }
+      if (d_col == 0) {
+        atomicAdd(&stats_s[(d_row0 + r) * 2 + 0], ssum[r]);
+        atomicAdd(&stats_s[(d_row0 + r) * 2 + 1], ssum2[r]);
+      }
+    }
+    __syncthreads();
+    const int nsplit = N * gridDim.x;
+    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    for (int t = tid; t < 16 * 2; t += kBlock) {
+      const int mg = m0 + (t >> 1);
+      if (mg < Cm) {
+        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -188,7 +224,8 @@ void conv_tap_kernel(const sa_bf16* __restrict__ x,
 // returns false if the shape/dtype is outside this kernel's envelope
 bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
                    const c10::optional<at::Tensor>& bias, at::Tensor& y,
-                   long padl, long dilation, long groups, bool is_dx) {
+                   long padl, long dilation, long groups, bool is_dx,
+                   at::Tensor* stats_out) {
   if (x.scalar_type() != at::kBFloat16 || w.scalar_type() != at::kBFloat16)
     return false;
   const int N = x.size(0), Cin = x.size(1);
@@ -222,12 +259,19 @@ bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
   const sa_bf16* bp =
       has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
 
+  float* sp = nullptr;
+  if (stats_out != nullptr) {
+    const long nsplit = (long)N * grid.x;
+    *stats_out = at::empty({Cout, nsplit, 2}, x.options().dtype(at::kFloat));
+    sp = stats_out->data_ptr<float>();
+  }
+
   auto launch = [&](auto dx_t, auto hb_t) {
     hipLaunchKernelGGL(
         (conv_tap_kernel<decltype(dx_t)::value, decltype(hb_t)::value>),
         grid, dim3(kBlock), lds, stream.stream(),
         (const sa_bf16*)x.data_ptr(), (const sa_bf16*)w.data_ptr(), bp,
-        (sa_bf16*)y.data_ptr(), N, Cin, Cout, Lin, Lout, K, (int)padl,
+        (sa_bf16*)y.data_ptr(), sp, N, Cin, Cout, Lin, Lout, K, (int)padl,
         (int)dilation, G, xext, xpitch);
   };
   if (is_dx) {
@@ -260,11 +304,13 @@ void conv_tap_s_kernel(const sa_bf16* __restrict__ x,
                        const sa_bf16* __restrict__ w,
                        const sa_bf16* __restrict__ bias,
                        sa_bf16* __restrict__ y,
+                       float* __restrict__ stats,
                        int N, int Cin, int Cout, long Lin, long Lout,
                        int K, int padl, int xext, int xpitch) {
   extern __shared__ sa_bf16 smem[];
   sa_bf16* w_s = smem;                         // [16][K][kCT] (c contiguous)
   sa_bf16* x_s = smem + 16 * K * kCT;          // [kCT][xpitch]
+  __shared__ float stats_s[16 * 2];
 
   const int n = blockIdx.y;
   const int m0 = blockIdx.z * 16;
@@ -364,14 +410,49 @@ void conv_tap_s_kernel(const sa_bf16* __restrict__ x,
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
   const long lg = l0 + wid * 16 + d_col;
-  if (lg >= Lout) return;
+  float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+  float ssum2[4] = {0.f, 0.f, 0.f, 0.f};
+  if (lg < Lout) {
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int mg = m0 + d_row0 + r;
-    if (mg < Cout) {
-      float v = acc[r];
-      if (HAS_BIAS) v += (float)bias[mg];
-      y[((long)n * Cout + mg) * Lout + lg] = (sa_bf16)v;
+    for (int r = 0; r < 4; ++r) {
+      const int mg = m0 + d_row0 + r;
+      if (mg < Cout) {
+        float v = acc[r];
+        if (HAS_BIAS) v += (float)bias[mg];
+        const sa_bf16 vb = (sa_bf16)v;
+        y[((long)n * Cout + mg) * Lout + lg] = vb;
+        if (stats != nullptr) {
+          const float vf = (float)vb;
+          ssum[r] += vf;
+          ssum2[r] += vf * vf;
+        }
+      }
+    }
+  }
+  if (stats != nullptr) {
+    __syncthreads();
+    for (int t = tid; t < 16 * 2; t += kBlock) stats_s[t] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int b = 1; b < 16; b <<= 1) {
+        ssum[r] += __shfl_xor(ssum[r], b, sa::kWave);
+        ssum2[r] += __shfl_xor(ssum2[r], b, sa::kWave);
+      }
+      if (d_col == 0) {
+        atomicAdd(&stats_s[(d_row0 + r) * 2 + 0], ssum[r]);
+        atomicAdd(&stats_s[(d_row0 + r) * 2 + 1], ssum2[r]);
+      }
+    }
+    __syncthreads();
+    const int nsplit = N * gridDim.x;
+    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    for (int t = tid; t < 16 * 2; t += kBlock) {
+      const int mg = m0 + (t >> 1);
+      if (mg < Cout) {
+        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
+      }
     }
   }
 }
@@ -382,7 +463,7 @@ void conv_tap_s_kernel(const sa_bf16* __restrict__ x,
 bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias, at::Tensor& y,
                      long stride, long padl, long dilation, long groups,
-                     bool is_dx) {
+                     bool is_dx, at::Tensor* stats_out) {
   if (x.scalar_type() != at::kBFloat16 || w.scalar_type() != at::kBFloat16)
     return false;
   if (groups != 1 || dilation != 1 || (stride != 2 && stride != 4))
@@ -408,13 +489,20 @@ bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
   if (has_bias) bct = bias->to(x.scalar_type()).contiguous();
   const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
 
+  float* sp = nullptr;
+  if (stats_out != nullptr) {
+    const long nsplit = (long)N * grid.x;
+    *stats_out = at::empty({Cout, nsplit, 2}, x.options().dtype(at::kFloat));
+    sp = stats_out->data_ptr<float>();
+  }
+
   auto launch = [&](auto dx_t, auto hb_t, auto s_t) {
     hipLaunchKernelGGL(
         (conv_tap_s_kernel<decltype(dx_t)::value, decltype(hb_t)::value,
                            decltype(s_t)::value>),
         grid, dim3(kBlock), lds, stream.stream(),
         (const sa_bf16*)x.data_ptr(), (const sa_bf16*)w.data_ptr(), bp,
-        (sa_bf16*)y.data_ptr(), N, Cin, Cout, Lin, Lout, K, (int)padl,
+        (sa_bf16*)y.data_ptr(), sp, N, Cin, Cout, Lin, Lout, K, (int)padl,
         xext, xpitch);
   };
   auto launch_hb = [&](auto dx_t, auto s_t) {

@@ -18,9 +18,10 @@
 
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor channel_sum(const at::Tensor& in);
+at::Tensor bn_sums_only(const at::Tensor& x);
 bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
                   const c10::optional<at::Tensor>& bias, at::Tensor& y,
-                  bool trans);
+                  bool trans, at::Tensor* stats_out = nullptr);
 
 namespace {
 
@@ -118,6 +119,27 @@ at::Tensor pw_conv_fwd(const at::Tensor& x, const at::Tensor& w,
         }
       });
   return y;
+}
+
+// forward + per-out-channel (sum, sumsq) partials for the following
+// BatchNorm (conv->BN fusion step 1, docs/FUSION_PLAN.md): the stats are
+// reduced in the conv epilogue while the tile is still in registers, so
+// bn_sums never re-reads y.
+std::vector<at::Tensor> pw_conv_fwd_stats(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& bias) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int N = x.size(0);
+  const long L = x.size(2);
+  const int Co = w.size(0);
+  auto y = at::empty({N, Co, L}, x.options());
+  at::Tensor part;
+  if (pw_mfma_gemm(x, w, bias, y, /*trans=*/false, &part)) {
+    return {y, part};
+  }
+  y = pw_conv_fwd(x, w, bias);
+  part = bn_sums_only(y).view({Co, 1, 2});
+  return {y, part};
 }
 
 std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,

@@ -36,15 +36,22 @@ constexpr int kKT = 32;     // ci per K-step
 constexpr int kWPitch = 40;   // LDS row pitch of the W chunk (bf16)
 constexpr int kXPitch = 136;  // LDS row pitch of the X chunk (bf16)
 
+// When `stats` is non-null the epilogue also accumulates per-out-channel
+// (sum, sumsq) of the bf16-rounded outputs into a partial slab
+// (Co, nsplit, 2), nsplit = N * gridDim.x — the producer half of the
+// conv->BN chain fusion (docs/FUSION_PLAN.md step 1): the following
+// BatchNorm consumes the slab and never re-reads y for its statistics.
 template <bool TRANS, bool HAS_BIAS>
 __global__ __launch_bounds__(kBlock)
 void pw_mfma_kernel(const sa_bf16* __restrict__ x,
                     const sa_bf16* __restrict__ w,
                     const sa_bf16* __restrict__ bias,
                     sa_bf16* __restrict__ y,
+                    float* __restrict__ stats,
                     int N, int Ci, int Co, long L) {
   __shared__ sa_bf16 w_s[kCoT * kWPitch];
   __shared__ sa_bf16 x_s[kKT * kXPitch];
+  __shared__ float stats_s[kCoT * 2];
 
   const int n = blockIdx.y;
   const int co0 = blockIdx.z * kCoT;
@@ -124,6 +131,8 @@ void pw_mfma_kernel(const sa_bf16* __restrict__ x,
   // ---- epilogue: D col = lane&15 (l), row = (lane>>4)*4 + r (co) ----
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
+  float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+  float ssum2[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int nrep = 0; nrep < 4; ++nrep) {
     const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
@@ -136,7 +145,42 @@ void pw_mfma_kernel(const sa_bf16* __restrict__ x,
       if (mg < Co) {
         float v = acc[r];
         if (HAS_BIAS) v += (float)bias[mg];
-        y[((long)n * Co + mg) * L + lg] = (sa_bf16)v;
+        const sa_bf16 vb = (sa_bf16)v;
+        y[((long)n * Co + mg) * L + lg] = vb;
+        if (stats != nullptr) {
+          const float vf = (float)vb;  // stats over the stored values
+          ssum[r] += vf;
+          ssum2[r] += vf * vf;
+        }
+      }
+    }
+  }
+  if (stats != nullptr) {
+    __syncthreads();
+    for (int t = tid; t < kCoT * 2; t += kBlock) stats_s[t] = 0.0f;
+    __syncthreads();
+    // reduce over the 16 l-columns of each channel group, then LDS-merge
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int b = 1; b < 16; b <<= 1) {
+        ssum[r] += __shfl_xor(ssum[r], b, sa::kWave);
+        ssum2[r] += __shfl_xor(ssum2[r], b, sa::kWave);
+      }
+      if (d_col == 0) {
+        const int mloc = wr * 16 + d_row0 + r;
+        atomicAdd(&stats_s[mloc * 2 + 0], ssum[r]);
+        atomicAdd(&stats_s[mloc * 2 + 1], ssum2[r]);
+      }
+    }
+    __syncthreads();
+    const int nsplit = N * gridDim.x;
+    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    for (int t = tid; t < kCoT * 2; t += kBlock) {
+      const int m = t >> 1;
+      const int mg = co0 + m;
+      if (mg < Co) {
+        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -144,10 +188,11 @@ void pw_mfma_kernel(const sa_bf16* __restrict__ x,
 
 }  // namespace
 
-// host entry; returns false if the shape/dtype is not handled
+// host entry; returns false if the shape/dtype is not handled.
+// `stats_out` (optional): receives the (Co, nsplit, 2) partial-sums slab.
 bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
                   const c10::optional<at::Tensor>& bias, at::Tensor& y,
-                  bool trans) {
+                  bool trans, at::Tensor* stats_out) {
   if (x.scalar_type() != at::ScalarType::BFloat16) return false;
   const int N = x.size(0), Ci = x.size(1);
   const long L = x.size(2);
@@ -167,22 +212,25 @@ bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
   const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
   sa_bf16* yp = (sa_bf16*)y.data_ptr();
 
+  float* sp = nullptr;
+  if (stats_out != nullptr) {
+    const long nsplit = (long)N * grid.x;
+    *stats_out = at::empty({Co, nsplit, 2}, x.options().dtype(at::kFloat));
+    sp = stats_out->data_ptr<float>();
+  }
+
+  auto launch = [&](auto tr_, auto hb_) {
+    hipLaunchKernelGGL((pw_mfma_kernel<decltype(tr_)::value,
+                                       decltype(hb_)::value>),
+                       grid, dim3(kBlock), 0, stream.stream(), xp, wp, bp,
+                       yp, sp, N, Ci, Co, L);
+  };
   if (trans) {
-    if (has_bias) {
-      hipLaunchKernelGGL((pw_mfma_kernel<true, true>), grid, dim3(kBlock), 0,
-                         stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
-    } else {
-      hipLaunchKernelGGL((pw_mfma_kernel<true, false>), grid, dim3(kBlock),
-                         0, stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
-    }
+    if (has_bias) launch(std::true_type{}, std::true_type{});
+    else launch(std::true_type{}, std::false_type{});
   } else {
-    if (has_bias) {
-      hipLaunchKernelGGL((pw_mfma_kernel<false, true>), grid, dim3(kBlock),
-                         0, stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
-    } else {
-      hipLaunchKernelGGL((pw_mfma_kernel<false, false>), grid, dim3(kBlock),
-                         0, stream.stream(), xp, wp, bp, yp, N, Ci, Co, L);
-    }
+    if (has_bias) launch(std::false_type{}, std::true_type{});
+    else launch(std::false_type{}, std::false_type{});
   }
   return true;
 }
@@ -204,10 +252,12 @@ void conv_mfma_kernel(const sa_bf16* __restrict__ x,
                       const sa_bf16* __restrict__ w,
                       const sa_bf16* __restrict__ bias,
                       sa_bf16* __restrict__ y,
+                      float* __restrict__ stats,
                       int N, int Cin, int Cout, long Lin, long Lout,
                       int K, int padl, int dil) {
   __shared__ sa_bf16 w_s[kCoT * kWPitch];
   __shared__ sa_bf16 x_s[kKT * kXPitch];
+  __shared__ float stats_s[kCoT * 2];
 
   const int n = blockIdx.y;
   const int m0 = blockIdx.z * kCoT;
@@ -296,6 +346,8 @@ void conv_mfma_kernel(const sa_bf16* __restrict__ x,
 
   const int d_col = lane & 15;
   const int d_row0 = (lane >> 4) * 4;
+  float ssum[4] = {0.f, 0.f, 0.f, 0.f};
+  float ssum2[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int nrep = 0; nrep < 4; ++nrep) {
     const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
@@ -308,7 +360,41 @@ void conv_mfma_kernel(const sa_bf16* __restrict__ x,
       if (mg < Cout) {
         float v = acc[r];
         if (HAS_BIAS) v += (float)bias[mg];
-        y[((long)n * Cout + mg) * Lout + lg] = (sa_bf16)v;
+        const sa_bf16 vb = (sa_bf16)v;
+        y[((long)n * Cout + mg) * Lout + lg] = vb;
+        if (stats != nullptr) {
+          const float vf = (float)vb;
+          ssum[r] += vf;
+          ssum2[r] += vf * vf;
+        }
+      }
+    }
+  }
+  if (stats != nullptr) {
+    __syncthreads();
+    for (int t = tid; t < kCoT * 2; t += kBlock) stats_s[t] = 0.0f;
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+#pragma unroll
+      for (int b = 1; b < 16; b <<= 1) {
+        ssum[r] += __shfl_xor(ssum[r], b, sa::kWave);
+        ssum2[r] += __shfl_xor(ssum2[r], b, sa::kWave);
+      }
+      if (d_col == 0) {
+        const int mloc = wr * 16 + d_row0 + r;
+        atomicAdd(&stats_s[mloc * 2 + 0], ssum[r]);
+        atomicAdd(&stats_s[mloc * 2 + 1], ssum2[r]);
+      }
+    }
+    __syncthreads();
+    const int nsplit = N * gridDim.x;
+    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    for (int t = tid; t < kCoT * 2; t += kBlock) {
+      const int m = t >> 1;
+      const int mg = m0 + m;
+      if (mg < Cout) {
+        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -444,7 +530,8 @@ void conv_mfma_f32_kernel(const float* __restrict__ x,
 // Dense stride-1 conv fwd/dx on MFMA; returns false if not applicable.
 bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
-               long padl, long dilation, bool is_dx) {
+               long padl, long dilation, bool is_dx,
+               at::Tensor* stats_out) {
   const bool is_bf16 = x.scalar_type() == at::ScalarType::BFloat16;
   const bool is_f32 = x.scalar_type() == at::ScalarType::Float;
   if (!is_bf16 && !is_f32) return false;
@@ -489,11 +576,18 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
   const sa_bf16* bp = has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr;
   sa_bf16* yp = (sa_bf16*)y.data_ptr();
 
+  float* sp = nullptr;
+  if (stats_out != nullptr) {
+    const long nsplit = (long)N * grid.x;
+    *stats_out = at::empty({Cout, nsplit, 2}, x.options().dtype(at::kFloat));
+    sp = stats_out->data_ptr<float>();
+  }
+
   auto launch = [&](auto dxp_, auto hb_) {
     hipLaunchKernelGGL((conv_mfma_kernel<decltype(dxp_)::value,
                                          decltype(hb_)::value>),
                        grid, dim3(kBlock), 0, stream.stream(), xp, wp, bp,
-                       yp, N, Cin, Cout, Lin, Lout, K, (int)padl,
+                       yp, sp, N, Cin, Cout, Lin, Lout, K, (int)padl,
                        (int)dilation);
   };
   if (is_dx) {

@@ -473,3 +473,58 @@ def test_pooled_attention_mask_varies(dev):
     _, _, m1 = ext().pooled_attn_train_fwd(q, k, v, 0.3)
     _, _, m2 = ext().pooled_attn_train_fwd(q, k, v, 0.3)
     assert not torch.equal(m1, m2), "dropout mask frozen across calls"
+
+
+@pytest.mark.parametrize("shape", [
+    # (Ci, Co, L, k, groups)
+    (24, 32, 1024, 1, 1),    # pointwise MFMA
+    (32, 32, 512, 3, 2),     # grouped conv_tap
+    (16, 48, 2048, 7, 1),    # dense conv_tap
+    (8, 16, 2048, 7, 1),     # strided falls to other kernels
+])
+def test_conv_stats_partials_match_bn_sums(shape):
+    """Fusion step 1: the producer-epilogue (C, nsplit, 2) slab reduces to
+    the same per-channel sums bn_sums computes from the output tensor."""
+    from seist_amd.ops import ext
+    torch.manual_seed(0)
+    Ci, Co, L, k, g = shape
+    x = torch.randn(5, Ci, L, device="cuda:0", dtype=torch.bfloat16)
+    w = torch.randn(Co, Ci // g, k, device="cuda:0",
+                    dtype=torch.bfloat16) * 0.2
+    y, part = ops.conv1d_stats(x, w, None, stride=1,
+                               padding=(k // 2, k - 1 - k // 2), groups=g)
+    assert part is not None
+    sums = part.sum(dim=1)
+    ref = ext().bn_sums_only(y.contiguous())
+    scale = ref.abs().max().item() or 1.0
+    d = (sums - ref).abs().max().item() / scale
+    assert d < 1e-4, f"stats mismatch {d}"
+
+
+def test_run_conv_bn_matches_separate_ops():
+    """run_conv_bn (stats in conv epilogue) == conv followed by bn_act."""
+    import torch.nn as nn
+    from seist_amd.models._blocks import run_conv_bn
+    torch.manual_seed(1)
+    conv = nn.Conv1d(24, 32, 5, padding=2, bias=False).to(
+        "cuda:0", torch.bfloat16)
+    bn = nn.BatchNorm1d(32).to("cuda:0")
+    bn.train()
+    x = torch.randn(6, 24, 1024, device="cuda:0", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y1 = run_conv_bn(conv, bn, x, act="gelu")
+    rm1, rv1 = bn.running_mean.clone(), bn.running_var.clone()
+    g1 = torch.autograd.grad(y1.float().pow(2).sum(), x)[0]
+
+    bn2 = nn.BatchNorm1d(32).to("cuda:0")
+    bn2.train()
+    x2 = x.detach().clone().requires_grad_(True)
+    z = ops.conv1d(x2, conv.weight, None, stride=1, padding=(2, 2))
+    y2 = ops.bn_act(z, bn2.weight, bn2.bias, bn2.running_mean,
+                    bn2.running_var, True, bn2.momentum, bn2.eps, act="gelu")
+    g2 = torch.autograd.grad(y2.float().pow(2).sum(), x2)[0]
+
+    assert torch.allclose(y1.float(), y2.float(), atol=1e-2)
+    assert torch.allclose(rm1, bn2.running_mean, atol=1e-4)
+    assert torch.allclose(rv1, bn2.running_var, atol=1e-4)
+    assert torch.allclose(g1.float(), g2.float(), atol=1e-2)
