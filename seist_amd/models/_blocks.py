@@ -1,9 +1,14 @@
 """Shared building helpers: run nn.Conv1d / nn.BatchNorm1d parameter
 containers through the MI355X op layer."""
 
+import os
+
 import torch.nn as nn
 
 from .. import ops
+
+# escape hatch for A/B runs (docs/FUSION_PLAN.md validation plan)
+_NO_FUSION = os.environ.get("SEIST_AMD_NO_FUSION") == "1"
 
 
 def run_conv(conv: nn.Conv1d, x, padl: int = 0, padr: int = 0):
@@ -55,7 +60,8 @@ def run_conv_bn(conv: nn.Conv1d, bn, x, act: str = "none", padl: int = 0,
         p = conv.padding[0] if isinstance(conv.padding, tuple) \
             else int(conv.padding)
         pl, pr = padl + p, padr + p
-    collect = (isinstance(bn, nn.BatchNorm1d) and bn.training and x.is_cuda)
+    collect = (isinstance(bn, nn.BatchNorm1d) and bn.training and x.is_cuda
+               and not _NO_FUSION)
     if collect:
         y, part = ops.conv1d_stats(x, conv.weight, conv.bias, stride=s,
                                    padding=(pl, pr), groups=conv.groups,

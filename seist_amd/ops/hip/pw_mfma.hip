@@ -27,6 +27,8 @@ typedef __bf16 sa_bf16;
 typedef sa_bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
+at::Tensor bn_sums_only(const at::Tensor& x);
+
 namespace {
 
 constexpr int kBlock = 256;
@@ -567,6 +569,10 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
     } else {
       if (has_bias) launch(std::false_type{}, std::true_type{});
       else launch(std::false_type{}, std::false_type{});
+    }
+    if (stats_out != nullptr) {
+      // fp32 kernel has no stats epilogue: one bn_sums pass instead
+      *stats_out = bn_sums_only(y).view({Cout, 1, 2});
     }
     return true;
   }
