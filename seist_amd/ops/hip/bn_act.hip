@@ -51,8 +51,11 @@ __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
 // one wave per channel: the per-channel partial reduction runs across the
 // 64 lanes instead of as one serial dependent-load loop (the thread-per-
 // channel version was latency-bound at ~12 us for KB-scale reads)
+// CMAJOR: part layout (C, nsplit, 2); else split-major (nsplit, C, 2)
+// (the conv-epilogue slabs are split-major so each block flushes one
+// coalesced run). The two coincide at nsplit == 1.
 __global__ void bn_finalize_kernel(const float* __restrict__ part,
-                                   int nsplit,
+                                   int nsplit, int cmajor,
                                    float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
@@ -65,8 +68,9 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
   if (c >= C) return;
   float s = 0.0f, s2 = 0.0f;
   for (int j = lane; j < nsplit; j += sa::kWave) {
-    s += part[((long)c * nsplit + j) * 2 + 0];
-    s2 += part[((long)c * nsplit + j) * 2 + 1];
+    const long o = cmajor ? ((long)c * nsplit + j) : ((long)j * C + c);
+    s += part[o * 2 + 0];
+    s2 += part[o * 2 + 1];
   }
   s = sa::warp_reduce_sum(s);
   s2 = sa::warp_reduce_sum(s2);
@@ -149,15 +153,16 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
 // reduce (C, nsplit, 2) partials -> (C, 2); one wave per channel
 __global__ void bn_part_reduce_kernel(const float* __restrict__ part,
                                       float* __restrict__ out,
-                                      int C, int nsplit) {
+                                      int C, int nsplit, int cmajor) {
   const int lane = threadIdx.x & (sa::kWave - 1);
   const int c = blockIdx.x * (blockDim.x / sa::kWave)
                 + threadIdx.x / sa::kWave;
   if (c >= C) return;
   float s1 = 0.0f, s2 = 0.0f;
   for (int j = lane; j < nsplit; j += sa::kWave) {
-    s1 += part[((long)c * nsplit + j) * 2 + 0];
-    s2 += part[((long)c * nsplit + j) * 2 + 1];
+    const long o = cmajor ? ((long)c * nsplit + j) : ((long)j * C + c);
+    s1 += part[o * 2 + 0];
+    s2 += part[o * 2 + 1];
   }
   s1 = sa::warp_reduce_sum(s1);
   s2 = sa::warp_reduce_sum(s2);
@@ -242,7 +247,8 @@ std::vector<at::Tensor> bn_act_fwd(const at::Tensor& x, const at::Tensor& gamma,
         });
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)),
                        dim3(256), 0, stream.stream(),
-                       part.data_ptr<float>(), nsplit, mean.data_ptr<float>(),
+                       part.data_ptr<float>(), nsplit, 1,
+                       mean.data_ptr<float>(),
                        invstd.data_ptr<float>(),
                        has_running ? running_mean->data_ptr<float>() : nullptr,
                        has_running ? running_var->data_ptr<float>() : nullptr,
@@ -295,7 +301,7 @@ at::Tensor bn_sums_only(const at::Tensor& x) {
       });
   hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
                      dim3(256), 0, stream.stream(), part.data_ptr<float>(),
-                     sums.data_ptr<float>(), C, nsplit);
+                     sums.data_ptr<float>(), C, nsplit, 1);
   return sums;
 }
 
@@ -319,7 +325,7 @@ std::vector<at::Tensor> bn_act_fwd_from_sums(
   const bool has_running = running_mean.has_value() && running_mean->defined();
   // sums is (C, 2) == part layout with nsplit == 1
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
-                     0, stream.stream(), sums.data_ptr<float>(), 1,
+                     0, stream.stream(), sums.data_ptr<float>(), 1, 1,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      has_running ? running_mean->data_ptr<float>() : nullptr,
                      has_running ? running_var->data_ptr<float>() : nullptr,
@@ -348,12 +354,12 @@ std::vector<at::Tensor> bn_act_fwd_with_part(
     long act) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
   TORCH_CHECK(part.is_contiguous() && part.dim() == 3
-              && part.size(0) == x.size(1) && part.size(2) == 2);
+              && part.size(1) == x.size(1) && part.size(2) == 2);
   const int N = x.size(0), C = x.size(1);
   const long L = x.size(2);
   const long NL = (long)N * L;
   const long total = (long)N * C * L;
-  const int nsplit = part.size(1);
+  const int nsplit = part.size(0);
   auto stream = at::hip::getCurrentHIPStream();
   auto opts = x.options().dtype(at::kFloat);
   auto g32 = gamma.to(at::kFloat).contiguous();
@@ -362,7 +368,7 @@ std::vector<at::Tensor> bn_act_fwd_with_part(
   auto invstd = at::empty({C}, opts);
   const bool has_running = running_mean.has_value() && running_mean->defined();
   hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
-                     0, stream.stream(), part.data_ptr<float>(), nsplit,
+                     0, stream.stream(), part.data_ptr<float>(), nsplit, 0,
                      mean.data_ptr<float>(), invstd.data_ptr<float>(),
                      has_running ? running_mean->data_ptr<float>() : nullptr,
                      has_running ? running_var->data_ptr<float>() : nullptr,
@@ -385,13 +391,14 @@ std::vector<at::Tensor> bn_act_fwd_with_part(
 // reduce a (C, nsplit, 2) slab to (C, 2) — used by the SyncBN path to
 // all-reduce producer-collected partials
 at::Tensor bn_part_to_sums(const at::Tensor& part) {
-  const int C = part.size(0);
-  const int nsplit = part.size(1);
+  // split-major producer slab (nsplit, C, 2)
+  const int C = part.size(1);
+  const int nsplit = part.size(0);
   auto sums = at::empty({C, 2}, part.options());
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
                      dim3(256), 0, stream.stream(), part.data_ptr<float>(),
-                     sums.data_ptr<float>(), C, nsplit);
+                     sums.data_ptr<float>(), C, nsplit, 0);
   return sums;
 }
 
@@ -421,7 +428,7 @@ at::Tensor bn_bwd_sums_only(const at::Tensor& dy, const at::Tensor& x,
       });
   hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
                      dim3(256), 0, stream.stream(), part.data_ptr<float>(),
-                     sums.data_ptr<float>(), C, nsplit);
+                     sums.data_ptr<float>(), C, nsplit, 1);
   return sums;
 }
 
@@ -485,7 +492,7 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
       });
   hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
                      dim3(256), 0, stream.stream(), part.data_ptr<float>(),
-                     sums.data_ptr<float>(), C, nsplit);
+                     sums.data_ptr<float>(), C, nsplit, 1);
 
   auto dx = at::empty_like(x);
   AT_DISPATCH_FLOATING_TYPES_AND2(

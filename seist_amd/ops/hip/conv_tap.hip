@@ -208,12 +208,11 @@ void conv_tap_kernel(const sa_bf16* __restrict__ x,
       }
     }
     __syncthreads();
-    const int nsplit = N * gridDim.x;
-    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    const long split = (long)blockIdx.y * gridDim.x + blockIdx.x;
     for (int t = tid; t < 16 * 2; t += kBlock) {
       const int mg = m0 + (t >> 1);
       if (mg < Cm) {
-        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
+        stats[(split * Cm + mg) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -262,7 +261,7 @@ bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
   float* sp = nullptr;
   if (stats_out != nullptr) {
     const long nsplit = (long)N * grid.x;
-    *stats_out = at::empty({Cout, nsplit, 2}, x.options().dtype(at::kFloat));
+    *stats_out = at::empty({nsplit, Cout, 2}, x.options().dtype(at::kFloat));
     sp = stats_out->data_ptr<float>();
   }
 
@@ -446,12 +445,11 @@ void conv_tap_s_kernel(const sa_bf16* __restrict__ x,
       }
     }
     __syncthreads();
-    const int nsplit = N * gridDim.x;
-    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    const long split = (long)blockIdx.y * gridDim.x + blockIdx.x;
     for (int t = tid; t < 16 * 2; t += kBlock) {
       const int mg = m0 + (t >> 1);
       if (mg < Cout) {
-        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
+        stats[(split * Cout + mg) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -492,7 +490,7 @@ bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
   float* sp = nullptr;
   if (stats_out != nullptr) {
     const long nsplit = (long)N * grid.x;
-    *stats_out = at::empty({Cout, nsplit, 2}, x.options().dtype(at::kFloat));
+    *stats_out = at::empty({nsplit, Cout, 2}, x.options().dtype(at::kFloat));
     sp = stats_out->data_ptr<float>();
   }
 

@@ -176,13 +176,14 @@ void pw_mfma_kernel(const sa_bf16* __restrict__ x,
       }
     }
     __syncthreads();
-    const int nsplit = N * gridDim.x;
-    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    // split-major slab (nsplit, Co, 2): one coalesced 256 B run per
+    // block instead of 64 scattered dwords at stride nsplit
+    const long split = (long)blockIdx.y * gridDim.x + blockIdx.x;
     for (int t = tid; t < kCoT * 2; t += kBlock) {
       const int m = t >> 1;
       const int mg = co0 + m;
       if (mg < Co) {
-        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
+        stats[(split * Co + mg) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -217,7 +218,7 @@ bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
   float* sp = nullptr;
   if (stats_out != nullptr) {
     const long nsplit = (long)N * grid.x;
-    *stats_out = at::empty({Co, nsplit, 2}, x.options().dtype(at::kFloat));
+    *stats_out = at::empty({nsplit, Co, 2}, x.options().dtype(at::kFloat));
     sp = stats_out->data_ptr<float>();
   }
 
@@ -390,13 +391,12 @@ void conv_mfma_kernel(const sa_bf16* __restrict__ x,
       }
     }
     __syncthreads();
-    const int nsplit = N * gridDim.x;
-    const int split = blockIdx.y * gridDim.x + blockIdx.x;
+    const long split = (long)blockIdx.y * gridDim.x + blockIdx.x;
     for (int t = tid; t < kCoT * 2; t += kBlock) {
       const int m = t >> 1;
       const int mg = m0 + m;
       if (mg < Cout) {
-        stats[((long)mg * nsplit + split) * 2 + (t & 1)] = stats_s[t];
+        stats[(split * Cout + mg) * 2 + (t & 1)] = stats_s[t];
       }
     }
   }
@@ -572,7 +572,7 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
     }
     if (stats_out != nullptr) {
       // fp32 kernel has no stats epilogue: one bn_sums pass instead
-      *stats_out = bn_sums_only(y).view({Cout, 1, 2});
+      *stats_out = bn_sums_only(y).view({1, Cout, 2});
     }
     return true;
   }
@@ -585,7 +585,7 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
   float* sp = nullptr;
   if (stats_out != nullptr) {
     const long nsplit = (long)N * grid.x;
-    *stats_out = at::empty({Cout, nsplit, 2}, x.options().dtype(at::kFloat));
+    *stats_out = at::empty({nsplit, Cout, 2}, x.options().dtype(at::kFloat));
     sp = stats_out->data_ptr<float>();
   }
 

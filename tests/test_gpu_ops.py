@@ -494,7 +494,7 @@ def test_conv_stats_partials_match_bn_sums(shape):
     y, part = ops.conv1d_stats(x, w, None, stride=1,
                                padding=(k // 2, k - 1 - k // 2), groups=g)
     assert part is not None
-    sums = part.sum(dim=1)
+    sums = part.sum(dim=0)  # split-major (nsplit, C, 2)
     ref = ext().bn_sums_only(y.contiguous())
     scale = ref.abs().max().item() or 1.0
     d = (sums - ref).abs().max().item() / scale
