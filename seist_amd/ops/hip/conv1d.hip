@@ -488,7 +488,7 @@ std::vector<at::Tensor> conv1d_fwd_stats(
     return {y, part};
   }
   y = conv1d_fwd(x, w, bias, stride, padl, padr, groups, dilation);
-  part = bn_sums_only(y).view({Co, 1, 2});
+  part = bn_sums_only(y).view({1, Co, 2});
   return {y, part};
 }
 

@@ -138,7 +138,7 @@ std::vector<at::Tensor> pw_conv_fwd_stats(
     return {y, part};
   }
   y = pw_conv_fwd(x, w, bias);
-  part = bn_sums_only(y).view({Co, 1, 2});
+  part = bn_sums_only(y).view({1, Co, 2});
   return {y, part};
 }
 
