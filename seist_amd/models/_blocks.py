@@ -7,8 +7,12 @@ import torch.nn as nn
 
 from .. import ops
 
-# escape hatch for A/B runs (docs/FUSION_PLAN.md validation plan)
-_NO_FUSION = os.environ.get("SEIST_AMD_NO_FUSION") == "1"
+# Producer-side BN-stats fusion (FUSION_PLAN step 1) measured NET NEGATIVE
+# on seist_m_dpk (35.7 -> 36.6 ms/step, same box A/B): the bn_sums pass it
+# removes is L2-resident and cheap (~9 us/site) while the conv-epilogue
+# reduction (+3.8 us/site) and the big-slab finalize (+7 us/site) cost
+# more. Opt-in for experiments; see profiles/step_profile_r02.md.
+_STATS_FUSION = os.environ.get("SEIST_AMD_STATS_FUSION") == "1"
 
 
 def run_conv(conv: nn.Conv1d, x, padl: int = 0, padr: int = 0):
@@ -61,7 +65,7 @@ def run_conv_bn(conv: nn.Conv1d, bn, x, act: str = "none", padl: int = 0,
             else int(conv.padding)
         pl, pr = padl + p, padr + p
     collect = (isinstance(bn, nn.BatchNorm1d) and bn.training and x.is_cuda
-               and not _NO_FUSION)
+               and _STATS_FUSION)
     if collect:
         y, part = ops.conv1d_stats(x, conv.weight, conv.bias, stride=s,
                                    padding=(pl, pr), groups=conv.groups,
