@@ -75,22 +75,25 @@ class Metrics:
     def _order_phases(self, targets: torch.Tensor, preds: torch.Tensor
                       ) -> Tuple[torch.Tensor, torch.Tensor]:
         """Greedy nearest-match ordering of multi-event pick lists
-        (reference metrics.py:101-125)."""
-        num_phases = targets.size(-1)
-        _targets = targets.detach().cpu().numpy()
-        _preds = preds.detach().cpu().numpy()
-        for i, (t_i, p_i) in enumerate(zip(_targets, _preds)):
-            ordered = np.zeros_like(p_i)
-            dmat = np.abs(t_i[:, None].astype(np.int64)
-                          - p_i[None, :].astype(np.int64))
-            for _ in range(num_phases):
-                ind = dmat.argmin()
-                ito, ifr = divmod(int(ind), num_phases)
-                ordered[ito] = p_i[ifr]
-                dmat[ito, :] = int(1 / self._epsilon)
-                dmat[:, ifr] = int(1 / self._epsilon)
-            _preds[i] = ordered
-        preds.copy_(torch.from_numpy(_preds))
+        (reference metrics.py:101-125), batched on device (K19): the K
+        greedy rounds run as tensor ops over the whole batch — no
+        per-sample Python, no D2H (the reference loops rows in numpy)."""
+        K = targets.size(-1)
+        N = targets.size(0)
+        big = int(1 / self._epsilon)
+        dmat = (targets[:, :, None].long()
+                - preds[:, None, :].long()).abs()        # (N, K, K)
+        ordered = torch.zeros_like(preds)
+        rows = torch.arange(N, device=preds.device)
+        for _ in range(K):
+            flat = dmat.reshape(N, -1)
+            ind = flat.argmin(dim=1)                      # first-min, as numpy
+            ito = torch.div(ind, K, rounding_mode="floor")
+            ifr = ind - ito * K
+            ordered[rows, ito] = preds[rows, ifr]
+            dmat[rows, ito, :] = big
+            dmat[rows, :, ifr] = big
+        preds.copy_(ordered)
         return targets, preds
 
     @torch.no_grad()

@@ -144,113 +144,100 @@ def _pick_phase(outputs: torch.Tensor, prob_threshold: float,
                 padding_value: int) -> torch.Tensor:
     """Batch phase picking -> (N, topk) sample indices, padded.
 
-    On GPU the rising-edge candidate mask is computed on-device and only
-    the sparse (row, col, value) candidate lists cross D2H — the reference
-    copies every full probability trace to the host per step
-    (postprocess.py:181)."""
+    Fully batched tensor ops, K18 of SURVEY §2.4: rising-edge candidate
+    mask, per-row top-k by height, then the reference's greedy +-mpd
+    suppression run as topk tiny batched steps — zero per-sample Python
+    and zero D2H in the step path (the reference copies every probability
+    trace to the host and loops rows, postprocess.py:181)."""
     x = outputs.detach().float()
     N, L = x.shape
-    if x.is_cuda:
-        dx = x[:, 1:] - x[:, :-1]
-        cand = torch.zeros_like(x, dtype=torch.bool)
-        cand[:, 1:-1] = (dx[:, 1:] <= 0) & (dx[:, :-1] > 0)
-        cand &= x >= prob_threshold
-        cand[:, 0] = False
-        cand[:, -1] = False
-        nz = cand.nonzero()
-        rows = nz[:, 0].cpu().numpy()
-        cols = nz[:, 1].cpu().numpy()
-        vals = x[cand].cpu().numpy()
+    dev = x.device
+    dx = x[:, 1:] - x[:, :-1]
+    cand = torch.zeros_like(x, dtype=torch.bool)
+    cand[:, 1:-1] = (dx[:, 1:] <= 0) & (dx[:, :-1] > 0)
+    cand &= x >= prob_threshold
+    cand[:, 0] = False
+    cand[:, -1] = False
+
+    k = min(topk, L)
+    heights = torch.where(cand, x, torch.full_like(x, float("-inf")))
+    vals, idx = heights.topk(k, dim=1)           # height-ordered, desc
+    valid = vals > float("-inf")
+
+    if min_peak_dist > 1:
+        idel = torch.zeros_like(valid)
+        for j in range(k):
+            active = valid[:, j] & ~idel[:, j]
+            ref = idx[:, j:j + 1]
+            rng = (idx >= ref - min_peak_dist) & (idx <= ref + min_peak_dist)
+            idel = idel | (rng & active[:, None])
+            idel[:, j] = torch.where(active,
+                                     torch.zeros_like(idel[:, j]),
+                                     idel[:, j])
+        kept = valid & ~idel
     else:
-        batch = x.numpy()
-        cand = _batch_candidate_peaks(batch, prob_threshold)
-        cand[:, 0] = False
-        cand[:, -1] = False
-        rows, cols = np.where(cand)
-        vals = batch[rows, cols]
-    out = np.full((N, topk), padding_value, dtype=np.int64)
-    # candidate counts per row (nonzero is row-major sorted)
-    counts = np.bincount(rows, minlength=N)
-    offsets = np.concatenate([[0], np.cumsum(counts)])
-    for i in range(N):
-        lo, hi = offsets[i], offsets[i + 1]
-        if hi <= lo:
-            continue
-        ind = cols[lo:hi]
-        v = vals[lo:hi]
-        if min_peak_dist <= 1:
-            # keep the topk highest peaks (in index order), matching the
-            # suppression branch's height-ordered selection semantics
-            order = np.argsort(v)[::-1][:topk]
-            ind = np.sort(ind[order])
-        else:
-            order = np.argsort(v)[::-1][:topk]
-            ind = ind[order]
-            idel = np.zeros(ind.size, dtype=bool)
-            for j in range(ind.size):
-                if not idel[j]:
-                    idel = idel | (ind >= ind[j] - min_peak_dist) \
-                        & (ind <= ind[j] + min_peak_dist)
-                    idel[j] = 0
-            ind = np.sort(ind[~idel])
-        k = min(topk, ind.size)
-        out[i, :k] = ind[:k]
-    return torch.tensor(out, dtype=torch.long, device=outputs.device)
+        kept = valid
+
+    big = L + 1
+    masked = torch.where(kept, idx, torch.full_like(idx, big))
+    ordered = masked.sort(dim=1).values
+    out = torch.where(ordered < big, ordered,
+                      torch.full_like(ordered, padding_value))
+    if k < topk:
+        pad = out.new_full((N, topk - k), padding_value)
+        out = torch.cat([out, pad], dim=1)
+    return out.to(dtype=torch.long)
 
 
 def _detect_event(outputs: torch.Tensor, prob_threshold: float,
                   topk: int) -> torch.Tensor:
-    """Batch event detection -> (N, 2*topk) [on,off] pairs, longest first,
-    padded with [1, 0].
+    """Batch event detection -> (N, 2*topk) [on,off] pairs, longest first
+    (stable on ties), padded with [1, 0].
 
-    Run detection is one vectorised pass over the whole (N, L) batch
-    (the reference calls obspy's trigger_onset per trace in a Python loop,
-    postprocess.py:129); only the tiny per-row top-k selection stays
-    sequential."""
+    Fully batched (K18): run starts/ends land in dense per-row tables via
+    scatter on the run ordinal, lengths are sorted stably per row — no
+    per-sample Python, no D2H (the reference calls obspy trigger_onset per
+    trace, postprocess.py:129)."""
     x = outputs.detach().float()
     N, L = x.shape
-    if x.is_cuda:
-        above_t = x > prob_threshold
-        d_t = above_t[:, 1:].char() - above_t[:, :-1].char()
-        s_nz = (d_t == 1).nonzero()
-        e_nz = (d_t == -1).nonzero()
-        srow = s_nz[:, 0].cpu().numpy()
-        scol = s_nz[:, 1].cpu().numpy()
-        erow = e_nz[:, 0].cpu().numpy()
-        ecol = e_nz[:, 1].cpu().numpy()
-        first = above_t[:, 0].nonzero().flatten().cpu().numpy()
-        last = above_t[:, -1].nonzero().flatten().cpu().numpy()
-    else:
-        batch = x.numpy()
-        above = batch > prob_threshold
-        d = np.diff(above.astype(np.int8), axis=1)
-        srow, scol = np.where(d == 1)
-        erow, ecol = np.where(d == -1)
-        first = np.where(above[:, 0])[0]
-        last = np.where(above[:, -1])[0]
-    srow = np.concatenate([srow, first])
-    scol = np.concatenate([scol + 1, np.zeros(len(first), dtype=scol.dtype)])
-    erow = np.concatenate([erow, last])
-    ecol = np.concatenate([ecol, np.full(len(last), L - 1,
-                                         dtype=ecol.dtype)])
-    s_order = np.lexsort((scol, srow))
-    e_order = np.lexsort((ecol, erow))
-    srow, scol = srow[s_order], scol[s_order]
-    ecol = ecol[e_order]
-    # starts and ends are paired in order within each row
-    counts = np.bincount(srow, minlength=N)
-    offsets = np.concatenate([[0], np.cumsum(counts)])
-    lengths = ecol - scol
-    out = np.tile(np.array([1, 0], dtype=np.int64), (N, topk))
-    for i in range(N):
-        lo, hi = offsets[i], offsets[i + 1]
-        if hi <= lo:
-            continue
-        order = np.argsort(-lengths[lo:hi], kind="stable")[:topk]
-        for j, oi in enumerate(order):
-            out[i, 2 * j] = scol[lo + oi]
-            out[i, 2 * j + 1] = ecol[lo + oi]
-    return torch.tensor(out, dtype=torch.long, device=outputs.device)
+    dev = x.device
+    above = x > prob_threshold
+    d = above[:, 1:].to(torch.int8) - above[:, :-1].to(torch.int8)
+    S = torch.zeros_like(above)
+    S[:, 1:] = d == 1
+    S[:, 0] = above[:, 0]
+    E = torch.zeros_like(above)
+    E[:, :-1] = d == -1
+    E[:, -1] = above[:, -1]
+
+    M = L // 2 + 2                      # max runs per row + trash slot 0
+    rid_s = torch.cumsum(S.long(), dim=1) * S.long()   # ordinal at starts
+    rid_e = torch.cumsum(E.long(), dim=1) * E.long()
+    cols = torch.arange(L, device=dev).expand(N, L)
+    start_tab = torch.full((N, M), -1, dtype=torch.long, device=dev)
+    end_tab = torch.full((N, M), -1, dtype=torch.long, device=dev)
+    start_tab.scatter_(1, rid_s, cols)  # non-starts collide on slot 0
+    end_tab.scatter_(1, rid_e, cols)
+    starts = start_tab[:, 1:]
+    ends = end_tab[:, 1:]
+    lengths = torch.where(starts >= 0, ends - starts,
+                          torch.full_like(starts, -1))
+
+    lv, li = lengths.sort(dim=1, descending=True, stable=True)
+    k = min(topk, M - 1)
+    li = li[:, :k]
+    lv = lv[:, :k]
+    on = starts.gather(1, li)
+    off = ends.gather(1, li)
+    ok = lv >= 0
+    on = torch.where(ok, on, torch.ones_like(on))
+    off = torch.where(ok, off, torch.zeros_like(off))
+    out = torch.stack([on, off], dim=2).reshape(N, 2 * k)
+    if k < topk:
+        pad = torch.tensor([1, 0], dtype=torch.long,
+                           device=dev).repeat(N, topk - k)
+        out = torch.cat([out, pad], dim=1)
+    return out
 
 
 def process_outputs(args: argparse.Namespace,
