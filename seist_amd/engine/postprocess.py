@@ -128,17 +128,6 @@ def trigger_onset(charfct: np.ndarray, thres1: float, thres2: float,
     return picks
 
 
-def _batch_candidate_peaks(batch: np.ndarray, mph: float):
-    """Vectorised rising-edge peak candidates for a (N, L) batch."""
-    dx = np.diff(batch, axis=1)
-    rising = np.zeros_like(batch, dtype=bool)
-    # (dx_next <= 0) & (dx_prev > 0) at interior points
-    rising[:, 1:-1] = (dx[:, 1:] <= 0) & (dx[:, :-1] > 0)
-    if mph is not None:
-        rising &= batch >= mph
-    return rising
-
-
 def _pick_phase(outputs: torch.Tensor, prob_threshold: float,
                 min_peak_dist: int, topk: int,
                 padding_value: int) -> torch.Tensor:
