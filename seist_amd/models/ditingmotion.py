@@ -13,6 +13,7 @@ import torch.nn.functional as F
 
 from ..ops.functional import auto_pad_lr
 from ._blocks import run_conv
+from .. import ops
 from ._registry import register_model
 
 
@@ -58,7 +59,8 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         x1 = self.conv_layers(x)
-        return self.pool(torch.cat([x, x1], dim=1))
+        return ops.max_pool1d(torch.cat([x, x1], dim=1),
+                              self.pool.kernel_size[0] if isinstance(self.pool.kernel_size, tuple) else self.pool.kernel_size)
 
 
 class SideLayer(nn.Module):

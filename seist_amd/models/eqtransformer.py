@@ -46,7 +46,7 @@ class ConvBlock(nn.Module):
     def forward(self, x):
         x = run_conv(self.conv, x, *self.conv_padding_same).relu()
         x = F.pad(x, (0, x.size(-1) % 2), "constant", -1 / _EPS)
-        return self.pool(x)
+        return ops.max_pool1d(x, 2)
 
 
 class ResConvBlock(nn.Module):

@@ -26,7 +26,7 @@ class ConvBlock(nn.Module):
         pl, pr = auto_pad_lr(x.size(-1), self.conv.kernel_size[0])
         x = run_conv(self.conv, x, pl, pr)
         x = self.dropout(x)
-        return self.pool(x)
+        return ops.max_pool1d(x, self.pool.kernel_size, ceil_mode=True)
 
 
 class MagNet(nn.Module):

@@ -421,7 +421,7 @@ class HeadClassification(nn.Module):
         self.out_act = out_act_layer()
 
     def forward(self, x, _: torch.Tensor = None):
-        x = self.flatten(x.mean(dim=-1, keepdim=True))
+        x = self.flatten(ops.global_avg_pool1d(x))
         return self.out_act(self.lin(x))
 
 
@@ -434,7 +434,7 @@ class HeadRegression(nn.Module):
         self.out_act = out_act_layer()
 
     def forward(self, x, _: torch.Tensor = None):
-        x = self.flatten(x.mean(dim=-1, keepdim=True))
+        x = self.flatten(ops.global_avg_pool1d(x))
         return self.out_act(self.lin(x))
 
 

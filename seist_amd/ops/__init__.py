@@ -62,6 +62,8 @@ from .functional import (  # noqa: E402,F401
     conv_transpose1d,
     gelu,
     interp_linear,
+    max_pool1d,
+    global_avg_pool1d,
     pointwise_conv,
     pooled_attention,
     upsample2x,

@@ -478,6 +478,62 @@ def avgmax_pool1d(x: torch.Tensor, k: int) -> torch.Tensor:
     return _AvgMaxPool.apply(x.contiguous(), k)
 
 
+class _MaxPool1d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k, ceil_mode):
+        ctx.k = k
+        ctx.in_len = x.size(-1)
+        if use_native(x):
+            y, idx = ext().max_pool1d_fwd(x, k, ceil_mode)
+            ctx.save_for_backward(idx)
+            return y
+        y, idx = F.max_pool1d(x.float(), k, ceil_mode=ceil_mode,
+                              return_indices=True)
+        ctx.save_for_backward(idx)
+        return y.to(x.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        dy = dy.contiguous()
+        if dy.is_cuda and has_ext():
+            return ext().max_pool1d_bwd(dy, idx, ctx.k, ctx.in_len), \
+                None, None
+        N, C, Lo = dy.shape
+        dx = dy.new_zeros(N, C, ctx.in_len)
+        dx.scatter_(2, idx, dy)
+        return dx, None, None
+
+
+def max_pool1d(x: torch.Tensor, k: int,
+               ceil_mode: bool = False) -> torch.Tensor:
+    """MaxPool1d with stride == kernel (K12): EQT/MagNet/DiTingMotion
+    encoder pools."""
+    return _MaxPool1d.apply(x.contiguous(), k, ceil_mode)
+
+
+class _GlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.in_len = x.size(-1)
+        if use_native(x):
+            return ext().gap_fwd(x)
+        return x.mean(dim=-1, keepdim=True)
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous()
+        if dy.is_cuda and has_ext():
+            return ext().gap_bwd(dy, ctx.in_len)
+        return (dy / ctx.in_len).expand(dy.size(0), dy.size(1), ctx.in_len) \
+            .contiguous()
+
+
+def global_avg_pool1d(x: torch.Tensor) -> torch.Tensor:
+    """AdaptiveAvgPool1d(1) (K12): classification/regression heads."""
+    return _GlobalAvgPool.apply(x.contiguous())
+
+
 # ---------------------------------------------------------------------------
 # linear interpolation resize (F.interpolate mode='linear', align_corners=False)
 # ---------------------------------------------------------------------------

@@ -66,6 +66,12 @@ at::Tensor bn_bwd_dx_from_sums(const at::Tensor& dy, const at::Tensor& x,
                                const at::Tensor& sums, double count, long act);
 
 std::vector<at::Tensor> avgmax_pool_fwd(const at::Tensor& x, long k);
+std::vector<at::Tensor> max_pool1d_fwd(const at::Tensor& x, long k,
+                                       bool ceil_mode);
+at::Tensor max_pool1d_bwd(const at::Tensor& dy, const at::Tensor& argmax,
+                          long k, long in_len);
+at::Tensor gap_fwd(const at::Tensor& x);
+at::Tensor gap_bwd(const at::Tensor& dy, long in_len);
 at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
                            long k, long in_len);
 at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len);
@@ -147,6 +153,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_bwd_dx_from_sums", &bn_bwd_dx_from_sums,
         "BN backward dx from externally reduced sums (SyncBN)");
   m.def("avgmax_pool_fwd", &avgmax_pool_fwd, "fused avg+max pool forward");
+  m.def("max_pool1d_fwd", &max_pool1d_fwd, "max pool (stride=k) forward");
+  m.def("max_pool1d_bwd", &max_pool1d_bwd, "max pool backward");
+  m.def("gap_fwd", &gap_fwd, "global average pool forward");
+  m.def("gap_bwd", &gap_bwd, "global average pool backward");
   m.def("avgmax_pool_bwd", &avgmax_pool_bwd, "fused avg+max pool backward");
   m.def("interp_linear_fwd", &interp_linear_fwd, "linear interp forward");
   m.def("interp_linear_bwd", &interp_linear_bwd, "linear interp backward");

@@ -188,6 +188,78 @@ __global__ void interp_bwd_lds_kernel(const scalar_t* __restrict__ dy,
   }
 }
 
+template <typename scalar_t>
+__global__ void maxpool_fwd_kernel(const scalar_t* __restrict__ x,
+                                   scalar_t* __restrict__ y,
+                                   int* __restrict__ argmax,
+                                   long L, long Lo, int k, long rows) {
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= rows * Lo) return;
+    const long row = i / Lo;
+    const long lo = i - row * Lo;
+    const long lo0 = lo * k;
+    const long lo1 = min(lo0 + (long)k, L);
+    const scalar_t* xr = x + row * L;
+    float mx = -INFINITY;
+    int mi = (int)lo0;
+    for (long l = lo0; l < lo1; ++l) {
+      const float v = (float)xr[l];
+      if (v > mx) {
+        mx = v;
+        mi = (int)l;
+      }
+    }
+    y[i] = (scalar_t)mx;
+    argmax[i] = mi;
+  }
+}
+
+template <typename scalar_t>
+__global__ void maxpool_bwd_kernel(const scalar_t* __restrict__ dy,
+                                   const int* __restrict__ argmax,
+                                   scalar_t* __restrict__ dx,
+                                   long L, long Lo, int k, long rows) {
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= rows * L) return;
+    const long row = i / L;
+    const long li = i - row * L;
+    const long lo = li / k;
+    const long oi = row * Lo + lo;
+    dx[i] = (argmax[oi] == (int)li) ? dy[oi] : (scalar_t)0.0f;
+  }
+}
+
+// global average pool to length 1 (AdaptiveAvgPool1d(1)): one wave per row
+template <typename scalar_t>
+__global__ void gap_fwd_kernel(const scalar_t* __restrict__ x,
+                               scalar_t* __restrict__ y, long L, long rows) {
+  const int lane = threadIdx.x & (sa::kWave - 1);
+  const long row = (long)blockIdx.x * (blockDim.x / sa::kWave)
+                   + threadIdx.x / sa::kWave;
+  if (row >= rows) return;
+  const scalar_t* xr = x + row * L;
+  float s = 0.0f;
+  for (long l = lane; l < L; l += sa::kWave) s += (float)xr[l];
+  s = sa::warp_reduce_sum(s);
+  if (lane == 0) y[row] = (scalar_t)(s / (float)L);
+}
+
+template <typename scalar_t>
+__global__ void gap_bwd_kernel(const scalar_t* __restrict__ dy,
+                               scalar_t* __restrict__ dx, long L, long rows) {
+#pragma unroll
+  for (int t = 0; t < kEwTile; ++t) {
+    const long i = ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
+    if (i >= rows * L) return;
+    const long row = i / L;
+    dx[i] = (scalar_t)((float)dy[row] / (float)L);
+  }
+}
+
 }  // namespace
 
 std::vector<at::Tensor> avgmax_pool_fwd(const at::Tensor& x, long k) {
@@ -358,6 +430,83 @@ at::Tensor upsample2x_bwd(const at::Tensor& dy) {
                            dim3(kBlock), 0, stream.stream(),
                            dy.data_ptr<scalar_t>(), dx.data_ptr<scalar_t>(),
                            Li, rows);
+      });
+  return dx;
+}
+
+
+std::vector<at::Tensor> max_pool1d_fwd(const at::Tensor& x, long k,
+                                       bool ceil_mode) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const long L = x.size(2);
+  const long rows = x.size(0) * x.size(1);
+  const long Lo = ceil_mode ? (L + k - 1) / k : L / k;
+  auto y = at::empty({x.size(0), x.size(1), Lo}, x.options());
+  auto argmax = at::empty({x.size(0), x.size(1), Lo},
+                          x.options().dtype(at::kInt));
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "max_pool1d_fwd", [&] {
+        hipLaunchKernelGGL(
+            (maxpool_fwd_kernel<scalar_t>),
+            dim3(sa::ceil_div(rows * Lo, (long)kBlock * kEwTile)),
+            dim3(kBlock), 0, stream.stream(), x.data_ptr<scalar_t>(),
+            y.data_ptr<scalar_t>(), argmax.data_ptr<int>(), L, Lo, (int)k,
+            rows);
+      });
+  return {y, argmax};
+}
+
+at::Tensor max_pool1d_bwd(const at::Tensor& dy, const at::Tensor& argmax,
+                          long k, long in_len) {
+  const long rows = dy.size(0) * dy.size(1);
+  const long Lo = dy.size(2);
+  auto dx = at::empty({dy.size(0), dy.size(1), in_len}, dy.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
+      "max_pool1d_bwd", [&] {
+        hipLaunchKernelGGL(
+            (maxpool_bwd_kernel<scalar_t>),
+            dim3(sa::ceil_div(rows * in_len, (long)kBlock * kEwTile)),
+            dim3(kBlock), 0, stream.stream(), dy.data_ptr<scalar_t>(),
+            argmax.data_ptr<int>(), dx.data_ptr<scalar_t>(), in_len, Lo,
+            (int)k, rows);
+      });
+  return dx;
+}
+
+at::Tensor gap_fwd(const at::Tensor& x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const long L = x.size(2);
+  const long rows = x.size(0) * x.size(1);
+  auto y = at::empty({x.size(0), x.size(1), 1}, x.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  const int rpb = kBlock / sa::kWave;
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "gap_fwd", [&] {
+        hipLaunchKernelGGL((gap_fwd_kernel<scalar_t>),
+                           dim3(sa::ceil_div(rows, (long)rpb)), dim3(kBlock),
+                           0, stream.stream(), x.data_ptr<scalar_t>(),
+                           y.data_ptr<scalar_t>(), L, rows);
+      });
+  return y;
+}
+
+at::Tensor gap_bwd(const at::Tensor& dy, long in_len) {
+  const long rows = dy.size(0) * dy.size(1);
+  auto dx = at::empty({dy.size(0), dy.size(1), in_len}, dy.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, dy.scalar_type(),
+      "gap_bwd", [&] {
+        hipLaunchKernelGGL(
+            (gap_bwd_kernel<scalar_t>),
+            dim3(sa::ceil_div(rows * in_len, (long)kBlock * kEwTile)),
+            dim3(kBlock), 0, stream.stream(), dy.data_ptr<scalar_t>(),
+            dx.data_ptr<scalar_t>(), in_len, rows);
       });
   return dx;
 }

@@ -528,3 +528,32 @@ def test_run_conv_bn_matches_separate_ops():
     assert torch.allclose(rm1, bn2.running_mean, atol=1e-4)
     assert torch.allclose(rv1, bn2.running_var, atol=1e-4)
     assert torch.allclose(g1.float(), g2.float(), atol=1e-2)
+
+
+@pytest.mark.parametrize("geo", [(6, 8, 8192, 2, False), (5, 64, 2048, 4, True),
+                                 (3, 18, 1000, 3, True)])
+def test_max_pool1d_native(geo):
+    N, C, L, k, ceil = geo
+    torch.manual_seed(3)
+    x = torch.randn(N, C, L, device="cuda:0", requires_grad=True)
+    y = ops.max_pool1d(x, k, ceil_mode=ceil)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x2 = x.detach().cpu().requires_grad_(True)
+    y2 = F.max_pool1d(x2, k, ceil_mode=ceil)
+    y2.backward(dy.cpu())
+    assert torch.allclose(y.cpu(), y2, atol=1e-6)
+    assert torch.allclose(x.grad.cpu(), x2.grad, atol=1e-6)
+
+
+def test_global_avg_pool_native():
+    torch.manual_seed(4)
+    x = torch.randn(7, 96, 128, device="cuda:0", requires_grad=True)
+    y = ops.global_avg_pool1d(x)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x2 = x.detach().cpu().requires_grad_(True)
+    y2 = x2.mean(-1, keepdim=True)
+    y2.backward(dy.cpu())
+    assert torch.allclose(y.cpu(), y2, atol=1e-5)
+    assert torch.allclose(x.grad.cpu(), x2.grad, atol=1e-6)
