@@ -40,6 +40,8 @@ def make_batch(model_name, batch, in_samples, device, dtype, seed):
         t = torch.rand(batch, 3, in_samples, generator=g)
     elif labels == ["emg"] or labels == ["baz"] or labels == ["dis"]:
         t = torch.rand(batch, 1, generator=g) * 5.0
+    elif labels == ["pmp"]:  # one-hot polarity
+        t = torch.eye(2)[torch.randint(0, 2, (batch,), generator=g)]
     else:
         t = torch.rand(batch, 3, in_samples, generator=g)
     return (x.to(device=device, dtype=dtype),

@@ -171,3 +171,38 @@ def test_seist_activation_checkpointing_matches():
     m1(x1).sum().backward()
     m2(x2).sum().backward()
     assert torch.allclose(x1.grad, x2.grad, atol=1e-6)
+
+
+@pytest.mark.parametrize("name", ["seist_l_dpk", "seist_l_pmp", "seist_l_emg",
+                                  "seist_l_baz", "seist_l_dis"])
+def test_seist_l_five_heads_train_step(name):
+    """configs[4] milestone: all five seist_l task heads run a full
+    CPU train step (forward, loss, backward, optimizer)."""
+    from seist_amd.config import Config
+    from seist_amd.ops import FusedAdam
+
+    torch.manual_seed(0)
+    m = create_model(name, in_channels=3, in_samples=2048).train()
+    opt = FusedAdam(m.parameters(), lr=1e-4)
+    x = torch.randn(3, 3, 2048)
+    loss_fn = Config.get_loss(name)
+    labels, tgt_trans = Config.get_model_config_(
+        name, "labels", "targets_transform_for_loss")
+    if labels == [["det", "ppk", "spk"]]:
+        t = torch.rand(3, 3, 2048)
+    elif labels == ["pmp"]:
+        t = torch.eye(2)[torch.randint(0, 2, (3,))]
+    else:
+        t = torch.rand(3, 1) * 5.0
+    if tgt_trans is not None:
+        t = tgt_trans(t)
+    out = m(x)
+    out = [o.float() for o in out] if isinstance(out, (list, tuple)) \
+        else out.float()
+    loss = loss_fn(out, t)
+    opt.zero_grad()
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(loss).item()
+    grads = [p.grad for p in m.parameters() if p.requires_grad]
+    assert all(g is not None and torch.isfinite(g).all() for g in grads)
