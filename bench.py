@@ -26,6 +26,7 @@ import torch
 from seist_amd.config import Config
 from seist_amd.engine.precision import convert_to_bf16
 from seist_amd.models import create_model
+from seist_amd.models._blocks import manage_bn_counters
 from seist_amd.ops import FusedAdam
 from seist_amd.parallel import dist as pdist
 from seist_amd.parallel.ddp import FlatReplica
@@ -88,6 +89,7 @@ def main():
     model = model.to(device).train()
 
     replica = FlatReplica(model)
+    bn_tick = manage_bn_counters(model)
     optimizer = FusedAdam(model.parameters(), lr=8e-5)
     loss_fn = Config.get_loss(args.model).to(device)
     tgt_trans = Config.get_model_config_(args.model,
@@ -122,6 +124,7 @@ def main():
         def exchange():
             replica.allreduce()
             optimizer.step()
+            bn_tick()
 
     def step():
         compute()

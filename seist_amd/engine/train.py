@@ -27,6 +27,7 @@ import torch
 from ..config import Config
 from ..data import SeismicDataset
 from ..models import create_model, load_checkpoint, save_checkpoint
+from ..models._blocks import manage_bn_counters
 from ..ops import FusedAdam
 from ..parallel import dist as pdist
 from ..parallel.ddp import FlatReplica, enable_native_syncbn, wrap_distributed
@@ -50,6 +51,9 @@ def _to_device(x, device, dtype=None):
 def train(args, tasks, model, optimizer, scheduler, loss_fn, train_loader,
           epoch, device, scalar_writer, replica=None) -> Union[list, dict]:
     model.train()
+    # one _foreach_add_ per step for all BN counters instead of one tiny
+    # kernel per layer per forward
+    bn_tick = manage_bn_counters(model)
 
     train_loss_per_step = []
     average_meters = {}
@@ -123,6 +127,7 @@ def train(args, tasks, model, optimizer, scheduler, loss_fn, train_loader,
                 replica.allreduce()
         with timer.phase("optimizer"):
             optimizer.step()
+            bn_tick()
             if scheduler is not None:
                 scheduler.step()
                 lr = scheduler.get_last_lr()[0]

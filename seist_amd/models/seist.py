@@ -82,7 +82,8 @@ def _norm(norm_module, x, act: str = "none", part=None):
             sync=getattr(norm_module, "_sync_bn", False),
             part=part,
         )
-        if norm_module.training and norm_module.track_running_stats:
+        if norm_module.training and norm_module.track_running_stats \
+                and not getattr(norm_module, "_managed_nbt", False):
             norm_module.num_batches_tracked += 1
         return y
     y = norm_module(x)
