@@ -225,3 +225,6 @@ def main():
 
 if __name__ == "__main__":
     main()
+    import torch.distributed as _dist
+    if _dist.is_available() and _dist.is_initialized():
+        _dist.destroy_process_group()

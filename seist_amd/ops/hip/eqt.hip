@@ -395,6 +395,120 @@ __global__ void lstm_fwd_kernel(const float* __restrict__ pre,  // (N,L,4H)
   }
 }
 
+// Wave-per-sample variant for 4H <= 64 (EQT hidden 16): the whole gate
+// vector fits one wavefront, so the h/c exchange runs on cross-lane
+// shuffles with ZERO barriers and W_hh lives in registers — samples never
+// wait on each other (the workgroup variant's per-step __syncthreads made
+// 8 independent samples stride together: 170 us/call vs ~8 us here).
+__global__ void lstm_fwd_wave_kernel(const float* __restrict__ pre,
+                                     const float* __restrict__ whh,
+                                     float* __restrict__ y,
+                                     float* __restrict__ cstash,
+                                     float* __restrict__ gstash,
+                                     int N, int L, int H, int dirs,
+                                     int dir) {
+  const int spb = blockDim.x / sa::kWave;
+  const int slot = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (sa::kWave - 1);
+  const int n = blockIdx.x * spb + slot;
+  const int G4 = 4 * H;
+  const bool active = (n < N) && (lane < G4);
+
+  float w[16];
+#pragma unroll
+  for (int hh = 0; hh < 16; ++hh) {
+    w[hh] = (active && hh < H) ? whh[(long)lane * H + hh] : 0.0f;
+  }
+  float hcur = 0.0f, ccur = 0.0f;  // valid on lanes < H
+
+  for (int step = 0; step < L; ++step) {
+    const int t = dir ? (L - 1 - step) : step;
+    const long base = ((long)n * L + t);
+    float acc = active ? pre[base * G4 + lane] : 0.0f;
+    for (int hh = 0; hh < H; ++hh) {
+      acc += w[hh] * __shfl(hcur, hh, sa::kWave);
+    }
+    const float gate = (lane / H == 2) ? tanhf(acc)
+                                       : 1.0f / (1.0f + expf(-acc));
+    const float iv = __shfl(gate, lane, sa::kWave);
+    const float fv = __shfl(gate, H + lane, sa::kWave);
+    const float gv = __shfl(gate, 2 * H + lane, sa::kWave);
+    const float ov = __shfl(gate, 3 * H + lane, sa::kWave);
+    if (active && lane < H) {
+      ccur = fv * ccur + iv * gv;
+      hcur = ov * tanhf(ccur);
+      y[base * (dirs * H) + dir * H + lane] = hcur;
+      if (cstash != nullptr) cstash[base * H + lane] = ccur;
+    }
+    if (active && gstash != nullptr) gstash[base * G4 + lane] = gate;
+  }
+}
+
+__global__ void lstm_bwd_wave_kernel(const float* __restrict__ dy,
+                                     const float* __restrict__ cstash,
+                                     const float* __restrict__ gstash,
+                                     const float* __restrict__ whh,
+                                     float* __restrict__ dgates,
+                                     int N, int L, int H, int dirs,
+                                     int dir) {
+  const int spb = blockDim.x / sa::kWave;
+  const int slot = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (sa::kWave - 1);
+  const int n = blockIdx.x * spb + slot;
+  const int G4 = 4 * H;
+  const bool active = (n < N) && (lane < G4);
+  const int h = lane % H;
+  const int chunk = lane / H;
+
+  // W column h in registers (lanes < H need it for the dh recursion)
+  float wcol[64];
+#pragma unroll 16
+  for (int g = 0; g < 64; ++g) {
+    wcol[g] = (active && lane < H && g < G4)
+        ? whh[(long)g * H + lane] : 0.0f;
+  }
+
+  float dh = 0.0f, dc = 0.0f;  // lanes < H
+  for (int step = L - 1; step >= 0; --step) {
+    const int t = dir ? (L - 1 - step) : step;
+    const long base = ((long)n * L + t);
+    const float gown = active ? gstash[base * G4 + lane] : 0.0f;
+    float dhv = 0.0f, dcv = 0.0f, tc = 0.0f, cprev = 0.0f;
+    if (active && lane < H) {
+      const float cv = cstash[base * H + lane];
+      tc = tanhf(cv);
+      const float ov = __shfl(gown, 3 * H + lane, sa::kWave);
+      dhv = dh + dy[base * (dirs * H) + dir * H + lane];
+      dcv = dc + dhv * ov * (1.0f - tc * tc);
+      cprev = (step == 0)
+          ? 0.0f
+          : cstash[((long)n * L + (dir ? (L - step) : (t - 1))) * H + lane];
+    }
+    // broadcast per-h quantities to all gate lanes
+    const float dcv_h = __shfl(dcv, h, sa::kWave);
+    const float dhv_h = __shfl(dhv, h, sa::kWave);
+    const float tc_h = __shfl(tc, h, sa::kWave);
+    const float cprev_h = __shfl(cprev, h, sa::kWave);
+    const float i_h = __shfl(gown, h, sa::kWave);            // i-chunk value
+    const float g_h = __shfl(gown, 2 * H + h, sa::kWave);    // g-chunk value
+    float dg = 0.0f;
+    if (chunk == 0) dg = dcv_h * g_h * gown * (1.0f - gown);
+    else if (chunk == 1) dg = dcv_h * cprev_h * gown * (1.0f - gown);
+    else if (chunk == 2) dg = dcv_h * i_h * (1.0f - gown * gown);
+    else dg = dhv_h * tc_h * gown * (1.0f - gown);
+    if (active) dgates[base * G4 + lane] = dg;
+    // dh_{t-1}[h] = sum_g w[g][h] * dg[g];  dc_{t-1} = dcv * f
+    float acc = 0.0f;
+    for (int g = 0; g < G4; ++g) {
+      acc += wcol[g] * __shfl(dg, g, sa::kWave);
+    }
+    if (active && lane < H) {
+      dh = acc;
+      dc = dcv * __shfl(gown, H + lane, sa::kWave);
+    }
+  }
+}
+
 // reverse-time BPTT; consumes the stashes; emits per-element dgates
 // (N,L,4H) — dW_ih/dW_hh/db and dx are GEMMs outside.
 __global__ void lstm_bwd_kernel(const float* __restrict__ dy,  // (N,L,D*H)
@@ -601,13 +715,26 @@ std::vector<at::Tensor> lstm_fwd(const at::Tensor& pre, const at::Tensor& whh,
   TORCH_CHECK(lds <= 160 * 1024, "LSTM hidden too large for LDS staging");
   auto cstash = training ? at::empty({N, L, H}, opts) : at::Tensor();
   auto gstash = training ? at::empty({N, L, G4}, opts) : at::Tensor();
-  hipLaunchKernelGGL(lstm_fwd_kernel, dim3(sa::ceil_div(N, spb)),
-                     dim3(spb * gwaves * sa::kWave), lds, stream.stream(),
-                     pre.data_ptr<float>(), whh.data_ptr<float>(),
-                     y.data_ptr<float>(),
-                     training ? cstash.data_ptr<float>() : nullptr,
-                     training ? gstash.data_ptr<float>() : nullptr,
-                     N, L, H, dirs, (int)dir, spb);
+  if (G4 <= sa::kWave) {
+    // wave-per-sample, barrier-free (H <= 16)
+    const int wspb = 4;
+    hipLaunchKernelGGL(lstm_fwd_wave_kernel,
+                       dim3(sa::ceil_div(N, wspb)),
+                       dim3(wspb * sa::kWave), 0, stream.stream(),
+                       pre.data_ptr<float>(), whh.data_ptr<float>(),
+                       y.data_ptr<float>(),
+                       training ? cstash.data_ptr<float>() : nullptr,
+                       training ? gstash.data_ptr<float>() : nullptr,
+                       N, L, H, dirs, (int)dir);
+  } else {
+    hipLaunchKernelGGL(lstm_fwd_kernel, dim3(sa::ceil_div(N, spb)),
+                       dim3(spb * gwaves * sa::kWave), lds, stream.stream(),
+                       pre.data_ptr<float>(), whh.data_ptr<float>(),
+                       y.data_ptr<float>(),
+                       training ? cstash.data_ptr<float>() : nullptr,
+                       training ? gstash.data_ptr<float>() : nullptr,
+                       N, L, H, dirs, (int)dir, spb);
+  }
   if (training) return {cstash, gstash};
   return {};
 }
@@ -624,11 +751,22 @@ at::Tensor lstm_bwd(const at::Tensor& dy, const at::Tensor& y,
   const long lds = (long)G4 * H * sizeof(float)
                    + (long)spb * (2 * H + G4) * sizeof(float);
   auto dgates = at::empty({N, L, G4}, dy.options());
-  hipLaunchKernelGGL(lstm_bwd_kernel, dim3(sa::ceil_div(N, spb)),
-                     dim3(spb * gwaves * sa::kWave), lds, stream.stream(),
-                     dy.contiguous().data_ptr<float>(), y.data_ptr<float>(),
-                     cstash.data_ptr<float>(), gstash.data_ptr<float>(),
-                     whh.data_ptr<float>(), dgates.data_ptr<float>(),
-                     N, L, H, (int)dirs, (int)dir, spb);
+  auto dy_c = dy.contiguous();
+  if (G4 <= sa::kWave) {
+    const int wspb = 4;
+    hipLaunchKernelGGL(lstm_bwd_wave_kernel, dim3(sa::ceil_div(N, wspb)),
+                       dim3(wspb * sa::kWave), 0, stream.stream(),
+                       dy_c.data_ptr<float>(), cstash.data_ptr<float>(),
+                       gstash.data_ptr<float>(), whh.data_ptr<float>(),
+                       dgates.data_ptr<float>(), N, L, H, (int)dirs,
+                       (int)dir);
+  } else {
+    hipLaunchKernelGGL(lstm_bwd_kernel, dim3(sa::ceil_div(N, spb)),
+                       dim3(spb * gwaves * sa::kWave), lds, stream.stream(),
+                       dy_c.data_ptr<float>(), y.data_ptr<float>(),
+                       cstash.data_ptr<float>(), gstash.data_ptr<float>(),
+                       whh.data_ptr<float>(), dgates.data_ptr<float>(),
+                       N, L, H, (int)dirs, (int)dir, spb);
+  }
   return dgates;
 }
