@@ -154,6 +154,7 @@ def main():
                 g_update = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g_update):
                     optimizer.step()
+                    bn_tick()
 
                 def step():  # noqa: F811
                     g_compute.replay()
