@@ -473,13 +473,16 @@ __global__ void lstm_bwd_wave_kernel(const float* __restrict__ dy,
     const int t = dir ? (L - 1 - step) : step;
     const long base = ((long)n * L + t);
     const float gown = active ? gstash[base * G4 + lane] : 0.0f;
+    // shuffles must run on ALL lanes (a shfl under a divergent branch
+    // reads zeros from exec-masked source lanes)
+    const float ov_l = __shfl(gown, 3 * H + h, sa::kWave);
+    const float fv_l = __shfl(gown, H + h, sa::kWave);
     float dhv = 0.0f, dcv = 0.0f, tc = 0.0f, cprev = 0.0f;
     if (active && lane < H) {
       const float cv = cstash[base * H + lane];
       tc = tanhf(cv);
-      const float ov = __shfl(gown, 3 * H + lane, sa::kWave);
       dhv = dh + dy[base * (dirs * H) + dir * H + lane];
-      dcv = dc + dhv * ov * (1.0f - tc * tc);
+      dcv = dc + dhv * ov_l * (1.0f - tc * tc);
       cprev = (step == 0)
           ? 0.0f
           : cstash[((long)n * L + (dir ? (L - step) : (t - 1))) * H + lane];
@@ -504,7 +507,7 @@ __global__ void lstm_bwd_wave_kernel(const float* __restrict__ dy,
     }
     if (active && lane < H) {
       dh = acc;
-      dc = dcv * __shfl(gown, H + lane, sa::kWave);
+      dc = dcv * fv_l;
     }
   }
 }
