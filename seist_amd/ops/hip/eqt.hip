@@ -425,7 +425,10 @@ __global__ void lstm_fwd_wave_kernel(const float* __restrict__ pre,
     const int t = dir ? (L - 1 - step) : step;
     const long base = ((long)n * L + t);
     float acc = active ? pre[base * G4 + lane] : 0.0f;
-    for (int hh = 0; hh < H; ++hh) {
+    // compile-time bound so w[] stays in registers (w is zero-padded
+    // past H, so the extra shuffles contribute nothing)
+#pragma unroll
+    for (int hh = 0; hh < 16; ++hh) {
       acc += w[hh] * __shfl(hcur, hh, sa::kWave);
     }
     const float gate = (lane / H == 2) ? tanhf(acc)
@@ -501,8 +504,10 @@ __global__ void lstm_bwd_wave_kernel(const float* __restrict__ dy,
     else dg = dhv_h * tc_h * gown * (1.0f - gown);
     if (active) dgates[base * G4 + lane] = dg;
     // dh_{t-1}[h] = sum_g w[g][h] * dg[g];  dc_{t-1} = dcv * f
+    // compile-time bound keeps wcol in registers (zero-padded past G4)
     float acc = 0.0f;
-    for (int g = 0; g < G4; ++g) {
+#pragma unroll
+    for (int g = 0; g < 64; ++g) {
       acc += wcol[g] * __shfl(dg, g, sa::kWave);
     }
     if (active && lane < H) {
