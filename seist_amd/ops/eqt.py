@@ -86,7 +86,9 @@ class _AdditiveAttn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, bh, wa, ba, tril_k, triu_k):
-        attn, ssum, amax = ext().addattn_fwd(q, k, bh, wa, float(ba),
+        # ba stays a device scalar (a host float() here would D2H-sync and
+        # break hipGraph capture)
+        attn, ssum, amax = ext().addattn_fwd(q, k, bh, wa, ba.reshape(1),
                                              tril_k, triu_k)
         ctx.save_for_backward(q, k, bh, wa, attn, amax)
         return attn

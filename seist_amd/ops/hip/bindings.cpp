@@ -98,7 +98,8 @@ std::vector<at::Tensor> ln_bwd(const at::Tensor& dy, const at::Tensor& x,
                                const at::Tensor& mean, const at::Tensor& rstd);
 std::vector<at::Tensor> addattn_fwd(const at::Tensor& q, const at::Tensor& k,
                                     const at::Tensor& bh, const at::Tensor& wa,
-                                    double ba, long tril_k, long triu_k);
+                                    const at::Tensor& ba, long tril_k,
+                                    long triu_k);
 std::vector<at::Tensor> addattn_bwd(const at::Tensor& q, const at::Tensor& k,
                                     const at::Tensor& bh, const at::Tensor& wa,
                                     const at::Tensor& attn,

@@ -123,7 +123,7 @@ __global__ void addattn_fwd_kernel(const float* __restrict__ q,   // (N,L,d)
                                    const float* __restrict__ k,   // (N,L,d)
                                    const float* __restrict__ bh,  // (d)
                                    const float* __restrict__ wa,  // (d)
-                                   float ba,
+                                   const float* __restrict__ ba_p,
                                    float* __restrict__ attn,      // (N,L,L)
                                    float* __restrict__ ssum,      // (N,L)
                                    int* __restrict__ amax,        // (N,L)
@@ -156,7 +156,7 @@ __global__ void addattn_fwd_kernel(const float* __restrict__ q,   // (N,L,d)
     const int j = tIdx * sa::kWave + lane;
     float e = -1e30f;
     if (j < L) {
-      e = ba;
+      e = ba_p[0];
       const float* kj = kt + (long)j * d;
       const float* qi = qt + (long)wid * d;
       for (int c = 0; c < d; ++c) {
@@ -645,7 +645,8 @@ std::vector<at::Tensor> ln_bwd(const at::Tensor& dy, const at::Tensor& x,
 
 std::vector<at::Tensor> addattn_fwd(const at::Tensor& q, const at::Tensor& k,
                                     const at::Tensor& bh, const at::Tensor& wa,
-                                    double ba, long tril_k, long triu_k) {
+                                    const at::Tensor& ba, long tril_k,
+                                    long triu_k) {
   TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kFloat, "additive attention is fp32");
   const int N = q.size(0), L = q.size(1), d = q.size(2);
@@ -660,7 +661,8 @@ std::vector<at::Tensor> addattn_fwd(const at::Tensor& q, const at::Tensor& k,
                      dim3((long)N * (L / kAttnRows)),
                      dim3(kAttnRows * sa::kWave), lds, stream.stream(),
                      q.data_ptr<float>(), k.data_ptr<float>(),
-                     bh.data_ptr<float>(), wa.data_ptr<float>(), (float)ba,
+                     bh.data_ptr<float>(), wa.data_ptr<float>(),
+                     ba.data_ptr<float>(),
                      attn.data_ptr<float>(), ssum.data_ptr<float>(),
                      amax.data_ptr<int>(), L, d, (int)tril_k, (int)triu_k);
   return {attn, ssum, amax};
