@@ -182,8 +182,10 @@ class StemBlock(nn.Module):
         self.norm = norm_layer(out_dim)
 
     def forward(self, x):
-        x = torch.cat([conv(x) for conv in self.convs], dim=1)
-        return run_conv_bn(self.out_proj, self.norm, x)
+        xs = [conv(x) for conv in self.convs]
+        y = ops.pointwise_conv_cat(xs, self.out_proj.weight,
+                                   self.out_proj.bias)
+        return _norm(self.norm, y)
 
 
 class GroupConvBlock(nn.Module):

@@ -65,6 +65,7 @@ from .functional import (  # noqa: E402,F401
     max_pool1d,
     global_avg_pool1d,
     pointwise_conv,
+    pointwise_conv_cat,
     pooled_attention,
     upsample2x,
 )

@@ -731,3 +731,45 @@ class _Upsample2x(torch.autograd.Function):
 def upsample2x(x: torch.Tensor) -> torch.Tensor:
     """Nearest-neighbour x2 upsample along the last dim."""
     return _Upsample2x.apply(x.contiguous())
+
+
+# ---------------------------------------------------------------------------
+# concat-fused pointwise conv: y = W @ cat(xs, dim=1) without the cat
+# ---------------------------------------------------------------------------
+
+
+class _PointwiseConvMulti(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, weight, bias, *xs):
+        ctx.save_for_backward(weight, *xs)
+        ctx.has_bias = bias is not None
+        return ext().pw_conv_multi_fwd(list(xs), weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        weight, *xs = ctx.saved_tensors
+        dy = dy.contiguous()
+        dxs = ext().pw_conv_multi_dx(dy, weight, [x.size(1) for x in xs])
+        # dw slices per input (three small hipblaslt GEMMs on the virtual
+        # concat's pieces), concatenated on the tiny (Co, Ci) weight grad
+        dw = torch.cat(
+            [ext().sum_batch(torch.bmm(dy, x.transpose(1, 2)))
+             for x in xs], dim=1).to(weight.dtype)
+        db = ext().channel_sum(dy).to(weight.dtype) if ctx.has_bias else None
+        return (dw, db) + tuple(dxs)
+
+
+def pointwise_conv_cat(xs, weight, bias=None):
+    """1x1 conv over the channel-concat of ``xs`` with the concat fused
+    into the kernel's LDS staging (forward) and the input gradient written
+    straight into per-input contiguous tensors (backward) — the stem's
+    3-path concat never exists (reference models/seist.py:187-195)."""
+    if weight.dim() == 3:
+        weight = weight.squeeze(-1)
+    if (use_native(xs[0]) and 2 <= len(xs) <= 3
+            and xs[0].dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and sum(x.size(1) for x in xs) >= 16):
+        return _PointwiseConvMulti.apply(
+            weight.contiguous(), bias, *[x.contiguous() for x in xs])
+    return pointwise_conv(torch.cat(xs, dim=1), weight, bias)

@@ -4,6 +4,11 @@
 
 at::Tensor pw_conv_fwd(const at::Tensor& x, const at::Tensor& w,
                        const c10::optional<at::Tensor>& bias);
+at::Tensor pw_conv_multi_fwd(std::vector<at::Tensor> xs, const at::Tensor& w,
+                             const c10::optional<at::Tensor>& bias);
+std::vector<at::Tensor> pw_conv_multi_dx(const at::Tensor& dy,
+                                         const at::Tensor& w,
+                                         std::vector<long> sizes);
 std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,
                                     const at::Tensor& w, bool has_bias);
 
@@ -129,6 +134,10 @@ void adam_step_packed(const at::Tensor& meta, const at::Tensor& sample,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pw_conv_fwd", &pw_conv_fwd, "pointwise conv forward (MFMA GEMM)");
+  m.def("pw_conv_multi_fwd", &pw_conv_multi_fwd,
+        "pointwise conv over a virtual channel-concat of inputs");
+  m.def("pw_conv_multi_dx", &pw_conv_multi_dx,
+        "input-gradient written into per-input contiguous tensors");
   m.def("pw_conv_bwd", &pw_conv_bwd, "pointwise conv backward");
   m.def("conv1d_fwd", &conv1d_fwd, "direct conv1d forward");
   m.def("conv1d_bwd", &conv1d_bwd, "direct conv1d backward");

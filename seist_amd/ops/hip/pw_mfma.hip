@@ -605,3 +605,209 @@ bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
   }
   return true;
 }
+
+// ---------------------------------------------------------------------------
+// Multi-input / multi-output pointwise GEMM: y = W @ concat(x0,x1,x2) with
+// the concat done during LDS staging (forward), and dx = W^T @ dy written
+// directly into per-input contiguous tensors (backward input-gradient) —
+// the stem's 3-path concat (reference models/seist.py:187-195) never
+// materializes and its backward needs no narrow+copy pass.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <bool TRANS, bool HAS_BIAS>
+__global__ __launch_bounds__(kBlock)
+void pw_mfma_multi_kernel(const sa_bf16* __restrict__ x0,
+                          const sa_bf16* __restrict__ x1,
+                          const sa_bf16* __restrict__ x2,
+                          int cb1, int cb2,     // input channel boundaries
+                          const sa_bf16* __restrict__ w,
+                          const sa_bf16* __restrict__ bias,
+                          sa_bf16* __restrict__ y0,
+                          sa_bf16* __restrict__ y1,
+                          sa_bf16* __restrict__ y2,
+                          int ob1, int ob2,     // output channel boundaries
+                          int N, int Ci, int Co, long L) {
+  __shared__ sa_bf16 w_s[kCoT * kWPitch];
+  __shared__ sa_bf16 x_s[kKT * kXPitch];
+
+  const int n = blockIdx.y;
+  const int co0 = blockIdx.z * kCoT;
+  const long l0 = (long)blockIdx.x * kLT;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wr = wid >> 1;
+  const int wc = wid & 1;
+  const int frag_m = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc3 = {0.f, 0.f, 0.f, 0.f};
+
+  for (int k0 = 0; k0 < Ci; k0 += kKT) {
+    __syncthreads();
+    for (int idx = tid; idx < kCoT * kKT; idx += kBlock) {
+      const int m = idx / kKT;
+      const int k = idx - m * kKT;
+      const int mg = co0 + m;
+      const int kg = k0 + k;
+      float v = 0.0f;
+      if (mg < Co && kg < Ci) {
+        v = TRANS ? (float)w[(long)kg * Co + mg]
+                  : (float)w[(long)mg * Ci + kg];
+      }
+      w_s[m * kWPitch + k] = (sa_bf16)v;
+    }
+    for (int idx = tid; idx < kKT * (kLT / 8); idx += kBlock) {
+      const int k = idx / (kLT / 8);
+      const int c8 = idx - k * (kLT / 8);
+      const long lg = l0 + c8 * 8;
+      const int kg = k0 + k;
+      bf16x8 v = {};
+      if (kg < Ci) {
+        const sa_bf16* row;
+        if (kg < cb1) {
+          row = x0 + ((long)n * cb1 + kg) * L;
+        } else if (kg < cb2) {
+          row = x1 + ((long)n * (cb2 - cb1) + (kg - cb1)) * L;
+        } else {
+          row = x2 + ((long)n * (Ci - cb2) + (kg - cb2)) * L;
+        }
+        if (lg + 8 <= L) {
+          v = *(const bf16x8*)(row + lg);
+        } else {
+          for (int j = 0; j < 8; ++j) {
+            v[j] = (lg + j < L) ? row[lg + j] : (sa_bf16)0.f;
+          }
+        }
+      }
+      *(bf16x8*)(x_s + k * kXPitch + c8 * 8) = v;
+    }
+    __syncthreads();
+
+    const bf16x8 a =
+        *(const bf16x8*)(w_s + (wr * 16 + frag_m) * kWPitch + kbase);
+    const int lb = wc * 64 + frag_m;
+#pragma unroll
+    for (int nrep = 0; nrep < 4; ++nrep) {
+      bf16x8 b;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        b[j] = x_s[(kbase + j) * kXPitch + nrep * 16 + lb];
+      }
+      switch (nrep) {
+        case 0: acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0); break;
+        case 1: acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc1, 0, 0, 0); break;
+        case 2: acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc2, 0, 0, 0); break;
+        case 3: acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc3, 0, 0, 0); break;
+      }
+    }
+  }
+
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int nrep = 0; nrep < 4; ++nrep) {
+    const f32x4 acc = (nrep == 0) ? acc0 : (nrep == 1) ? acc1
+                      : (nrep == 2) ? acc2 : acc3;
+    const long lg = l0 + wc * 64 + nrep * 16 + d_col;
+    if (lg >= L) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int mg = co0 + wr * 16 + d_row0 + r;
+      if (mg < Co) {
+        float v = acc[r];
+        if (HAS_BIAS) v += (float)bias[mg];
+        sa_bf16* dst;
+        int ml, cw;
+        if (mg < ob1) { dst = y0; ml = mg; cw = ob1; }
+        else if (mg < ob2) { dst = y1; ml = mg - ob1; cw = ob2 - ob1; }
+        else { dst = y2; ml = mg - ob2; cw = Co - ob2; }
+        dst[((long)n * cw + ml) * L + lg] = (sa_bf16)v;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// y = W @ concat(xs) along channels; xs size 2 or 3, bf16, same (N, *, L)
+at::Tensor pw_conv_multi_fwd(std::vector<at::Tensor> xs, const at::Tensor& w,
+                             const c10::optional<at::Tensor>& bias) {
+  TORCH_CHECK(xs.size() >= 2 && xs.size() <= 3);
+  const int N = xs[0].size(0);
+  const long L = xs[0].size(2);
+  int Ci = 0;
+  for (auto& x : xs) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous()
+                && x.scalar_type() == at::kBFloat16
+                && x.size(0) == N && x.size(2) == L);
+    Ci += x.size(1);
+  }
+  const int Co = w.size(0);
+  TORCH_CHECK(w.size(1) == Ci && Ci >= 16);
+  auto y = at::empty({N, Co, L}, xs[0].options());
+  const bool has_bias = bias.has_value() && bias->defined();
+  at::Tensor bct;
+  if (has_bias) bct = bias->to(at::kBFloat16).contiguous();
+  const int cb1 = xs[0].size(1);
+  const int cb2 = cb1 + xs[1].size(1);
+  const sa_bf16* x2p = xs.size() > 2 ? (const sa_bf16*)xs[2].data_ptr()
+                                     : (const sa_bf16*)xs[1].data_ptr();
+  dim3 grid(sa::ceil_div(L, kLT), N, sa::ceil_div(Co, kCoT));
+  auto stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto hb_) {
+    hipLaunchKernelGGL((pw_mfma_multi_kernel<false, decltype(hb_)::value>),
+                       grid, dim3(kBlock), 0, stream.stream(),
+                       (const sa_bf16*)xs[0].data_ptr(),
+                       (const sa_bf16*)xs[1].data_ptr(), x2p, cb1,
+                       xs.size() > 2 ? cb2 : Ci,
+                       (const sa_bf16*)w.data_ptr(),
+                       has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr,
+                       (sa_bf16*)y.data_ptr(), (sa_bf16*)y.data_ptr(),
+                       (sa_bf16*)y.data_ptr(), Co, Co, N, Ci, Co, L);
+  };
+  if (has_bias) launch(std::true_type{});
+  else launch(std::false_type{});
+  return y;
+}
+
+// dxs = split(W^T @ dy) written straight into per-input contiguous tensors
+std::vector<at::Tensor> pw_conv_multi_dx(const at::Tensor& dy,
+                                         const at::Tensor& w,
+                                         std::vector<long> sizes) {
+  TORCH_CHECK(sizes.size() >= 2 && sizes.size() <= 3);
+  const int N = dy.size(0);
+  const long L = dy.size(2);
+  const int Cin = dy.size(1);   // GEMM reduce dim (= conv out channels)
+  long Ct = 0;
+  for (long s : sizes) Ct += s;
+  // w is the ORIGINAL conv weight (Co_orig=Cin rows, Ci_orig=Ct cols);
+  // dx[c] = sum_m w[m][c] * dy[m] -> the TRANS access w[kg * Co + mg]
+  // with the kernel's Co == Ct reads w[m * Ct + c], exactly w[m][c].
+  TORCH_CHECK(w.size(0) == Cin && w.size(1) == Ct);
+  std::vector<at::Tensor> dxs;
+  for (long s : sizes) dxs.push_back(at::empty({N, s, L}, dy.options()));
+  const int ob1 = sizes[0];
+  const int ob2 = ob1 + sizes[1];
+  sa_bf16* y2p = sizes.size() > 2 ? (sa_bf16*)dxs[2].data_ptr()
+                                  : (sa_bf16*)dxs[1].data_ptr();
+  dim3 grid(sa::ceil_div(L, kLT), N, sa::ceil_div((long)Ct, (long)kCoT));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL((pw_mfma_multi_kernel<true, false>), grid,
+                     dim3(kBlock), 0, stream.stream(),
+                     (const sa_bf16*)dy.data_ptr(),
+                     (const sa_bf16*)dy.data_ptr(),
+                     (const sa_bf16*)dy.data_ptr(), Cin, Cin,
+                     (const sa_bf16*)w.data_ptr(), nullptr,
+                     (sa_bf16*)dxs[0].data_ptr(),
+                     (sa_bf16*)dxs[1].data_ptr(), y2p, ob1,
+                     sizes.size() > 2 ? ob2 : (int)Ct,
+                     N, Cin, (int)Ct, L);
+  return dxs;
+}

@@ -557,3 +557,32 @@ def test_global_avg_pool_native():
     y2.backward(dy.cpu())
     assert torch.allclose(y.cpu(), y2, atol=1e-5)
     assert torch.allclose(x.grad.cpu(), x2.grad, atol=1e-6)
+
+
+@pytest.mark.parametrize("widths", [(16, 16, 16), (8, 16, 24), (16, 16)])
+def test_pointwise_conv_cat_fused(widths):
+    """Concat-fused pw conv == pw conv over torch.cat (fwd + all grads)."""
+    torch.manual_seed(5)
+    N, L, Co = 4, 512, 32
+    Ci = sum(widths)
+    xs = [torch.randn(N, c, L, device="cuda:0", dtype=torch.bfloat16,
+                      requires_grad=True) for c in widths]
+    w = (torch.randn(Co, Ci, device="cuda:0", dtype=torch.bfloat16) * 0.1
+         ).requires_grad_(True)
+    b = torch.randn(Co, device="cuda:0", dtype=torch.bfloat16
+                    ).requires_grad_(True)
+
+    y1 = ops.pointwise_conv_cat(xs, w, b)
+    dy = torch.randn_like(y1)
+    g1 = torch.autograd.grad(y1, xs + [w, b], dy)
+
+    xs2 = [x.detach().clone().requires_grad_(True) for x in xs]
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = ops.pointwise_conv(torch.cat(xs2, dim=1), w2, b2)
+    g2 = torch.autograd.grad(y2, xs2 + [w2, b2], dy)
+
+    assert torch.allclose(y1.float(), y2.float(), atol=1e-2)
+    for a, bb in zip(g1, g2):
+        assert torch.allclose(a.float(), bb.float(), atol=5e-2), \
+            (a - bb).abs().max().item()
