@@ -142,8 +142,13 @@ class MLP(nn.Module):
 
     def forward(self, x):
         x = ops.pointwise_conv(x, self.lin0.weight, self.lin0.bias)
-        x = ops.gelu(x) if _is_gelu(self.act) else self.act(x)
-        x = ops.pointwise_conv(x, self.lin1.weight, self.lin1.bias)
+        if _is_gelu(self.act):
+            # GELU fused into lin1's LDS staging (never hits HBM)
+            x = ops.act_pw(x, "gelu", self.lin1.weight, self.lin1.bias,
+                           module=self.lin1)
+        else:
+            x = self.act(x)
+            x = ops.pointwise_conv(x, self.lin1.weight, self.lin1.bias)
         return self.dropout(x)
 
 
@@ -206,12 +211,23 @@ class GroupConvBlock(nn.Module):
         self.droppath1 = DropPath(path_drop_rate)
 
     def forward(self, x):
-        y = run_conv_bn(self.conv, self.norm0, x, auto_pad=True,
-                        act="gelu" if _is_gelu(self.act) else "none")
-        y = ops.pointwise_conv(y, self.proj.weight, self.proj.bias)
+        # grouped conv -> [BN+act fused into proj's staging] -> residual
+        y = _conv(self.conv, x, auto_pad=True)
+        y = ops.bn_act_pw(y, self.norm0,
+                          "gelu" if _is_gelu(self.act) else "none",
+                          self.proj.weight, self.proj.bias)
         x = ops.droppath_add(x, y, self.droppath0.drop_prob, self.training)
-        y = _norm(self.norm1, x)
-        y = self.mlp(y)
+        # [BN fused into mlp.lin0's staging] -> GELU fused into lin1
+        y = ops.bn_act_pw(x, self.norm1, "none", self.mlp.lin0.weight,
+                          self.mlp.lin0.bias)
+        if _is_gelu(self.mlp.act):
+            y = ops.act_pw(y, "gelu", self.mlp.lin1.weight,
+                           self.mlp.lin1.bias, module=self.mlp.lin1)
+        else:
+            y = self.mlp.act(y)
+            y = ops.pointwise_conv(y, self.mlp.lin1.weight,
+                                   self.mlp.lin1.bias)
+        y = self.mlp.dropout(y)
         return ops.droppath_add(x, y, self.droppath1.drop_prob,
                                 self.training)
 

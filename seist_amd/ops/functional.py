@@ -773,3 +773,107 @@ def pointwise_conv_cat(xs, weight, bias=None):
         return _PointwiseConvMulti.apply(
             weight.contiguous(), bias, *[x.contiguous() for x in xs])
     return pointwise_conv(torch.cat(xs, dim=1), weight, bias)
+
+
+# ---------------------------------------------------------------------------
+# fused BN(+act) -> pointwise conv (FUSION_PLAN step 2): the normalized/
+# activated tensor never exists in HBM — it is produced inside the
+# consumer conv's LDS staging (forward) and inside the split-K weight-
+# gradient kernel's staging (backward).
+# ---------------------------------------------------------------------------
+
+
+class _BNActPw(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, mean, invstd, weight, bias, act,
+                bn_training, act_only=False):
+        scale = (gamma.float() * invstd).contiguous()
+        shift = (beta.float() - mean * scale).contiguous()
+        y = ext().pw_conv_pre_fwd(x, weight, bias, scale, shift, act)
+        ctx.save_for_backward(x, gamma, beta, mean, invstd, weight, scale,
+                              shift)
+        ctx.act = act
+        ctx.has_bias = bias is not None
+        ctx.bn_training = bn_training
+        ctx.act_only = act_only
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x, gamma, beta, mean, invstd, weight, scale,
+         shift) = ctx.saved_tensors
+        dy = dy.contiguous()
+        dz = ext().pw_conv_dx(dy, weight)
+        dw = ext().pw_dw_pre(dy, x, scale, shift, ctx.act).to(weight.dtype)
+        db = ext().channel_sum(dy).to(weight.dtype) if ctx.has_bias else None
+        if ctx.act_only:
+            # identity-BN: dx = act_grad(x) * dz, no dgamma/dbeta passes
+            dx = ext().bn_bwd_dx_eval(dz, x, mean, invstd, gamma, beta,
+                                      ctx.act)
+            dgamma = dbeta = None
+        else:
+            dx, dgamma, dbeta = ext().bn_act_bwd(dz, x, gamma, beta, mean,
+                                                 invstd, ctx.bn_training,
+                                                 ctx.act)
+        return (dx, dgamma, dbeta, None, None, dw, db, None, None, None)
+
+
+def bn_act_pw(x, bn, act, weight, bias):
+    """BatchNorm(+act) fused into the following 1x1 conv. ``bn`` is the
+    nn.BatchNorm1d module (stats/running updates handled here); the module
+    tree and checkpoint format are untouched."""
+    if weight.dim() == 3:
+        weight = weight.squeeze(-1)
+    act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
+    fusable = (use_native(x) and x.dtype == torch.bfloat16
+               and weight.dtype == torch.bfloat16 and x.size(1) >= 16
+               and not getattr(bn, "_sync_bn", False))
+    if not fusable:
+        y = bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+                   bn.training, bn.momentum, bn.eps, act=act,
+                   sync=getattr(bn, "_sync_bn", False))
+        if bn.training and bn.track_running_stats \
+                and not getattr(bn, "_managed_nbt", False):
+            bn.num_batches_tracked += 1
+        return pointwise_conv(y, weight, bias)
+    x = x.contiguous()
+    if bn.training:
+        sums = ext().bn_sums_only(x)
+        count = x.size(0) * x.size(2)
+        mean, invstd = ext().bn_finalize_only(
+            sums, float(count), bn.running_mean, bn.running_var,
+            bn.momentum, bn.eps)
+        if bn.track_running_stats \
+                and not getattr(bn, "_managed_nbt", False):
+            bn.num_batches_tracked += 1
+    else:
+        mean = bn.running_mean.float()
+        invstd = torch.rsqrt(bn.running_var.float() + bn.eps)
+    return _BNActPw.apply(x, bn.weight, bn.bias, mean, invstd,
+                          weight.contiguous(), bias, act_id, bn.training)
+
+
+def act_pw(x, act, weight, bias, module=None):
+    """Elementwise activation fused into the following 1x1 conv (the MLP's
+    GELU between lin0 and lin1 — reference models/seist.py:99-121)."""
+    if weight.dim() == 3:
+        weight = weight.squeeze(-1)
+    if not (use_native(x) and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16 and x.size(1) >= 16):
+        y = gelu(x) if act == "gelu" else (x.relu() if act == "relu" else x)
+        return pointwise_conv(y, weight, bias)
+    act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
+    x = x.contiguous()
+    C = x.size(1)
+    cache = getattr(module, "_actpw_cache", None) if module is not None \
+        else None
+    if cache is None or cache[0].numel() != C \
+            or cache[0].device != x.device:
+        ones = torch.ones(C, dtype=torch.float32, device=x.device)
+        zeros = torch.zeros(C, dtype=torch.float32, device=x.device)
+        cache = (ones, zeros)
+        if module is not None:
+            module._actpw_cache = cache
+    ones, zeros = cache
+    return _BNActPw.apply(x, ones, zeros, zeros, ones,
+                          weight.contiguous(), bias, act_id, False, True)

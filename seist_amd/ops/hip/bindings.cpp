@@ -54,6 +54,23 @@ std::vector<at::Tensor> bn_act_fwd_with_part(
     const c10::optional<at::Tensor>& running_var, double momentum, double eps,
     long act);
 at::Tensor bn_part_to_sums(const at::Tensor& part);
+std::vector<at::Tensor> bn_finalize_only(
+    const at::Tensor& sums, double count,
+    const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum,
+    double eps);
+at::Tensor pw_conv_pre_fwd(const at::Tensor& x, const at::Tensor& w,
+                           const c10::optional<at::Tensor>& bias,
+                           const at::Tensor& scale, const at::Tensor& shift,
+                           long act);
+at::Tensor pw_conv_dx(const at::Tensor& dy, const at::Tensor& w);
+at::Tensor bn_bwd_dx_eval(const at::Tensor& dy, const at::Tensor& x,
+                          const at::Tensor& mean, const at::Tensor& invstd,
+                          const at::Tensor& gamma, const at::Tensor& beta,
+                          long act);
+at::Tensor pw_dw_pre(const at::Tensor& dy, const at::Tensor& x,
+                     const c10::optional<at::Tensor>& scale,
+                     const c10::optional<at::Tensor>& shift, long act);
 std::vector<at::Tensor> bn_act_fwd_from_sums(
     const at::Tensor& x, const at::Tensor& sums, double count,
     const at::Tensor& gamma, const at::Tensor& beta,
@@ -156,6 +173,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "BN+act forward from producer-collected partials");
   m.def("bn_part_to_sums", &bn_part_to_sums,
         "(C,nsplit,2) partial slab -> (C,2) sums");
+  m.def("bn_finalize_only", &bn_finalize_only,
+        "mean/invstd + running update from (C,2) sums");
+  m.def("pw_conv_pre_fwd", &pw_conv_pre_fwd,
+        "pointwise conv with BN/act transform applied in staging");
+  m.def("pw_conv_dx", &pw_conv_dx, "pointwise input-gradient only");
+  m.def("bn_bwd_dx_eval", &bn_bwd_dx_eval,
+        "eval-mode BN/act dx only (no dgamma/dbeta)");
+  m.def("pw_dw_pre", &pw_dw_pre,
+        "split-K MFMA weight grad with transform-staged x");
   m.def("bn_act_fwd_from_sums", &bn_act_fwd_from_sums,
         "BN+act forward from externally reduced sums (SyncBN)");
   m.def("bn_bwd_sums_only", &bn_bwd_sums_only,

@@ -402,6 +402,56 @@ at::Tensor bn_part_to_sums(const at::Tensor& part) {
   return sums;
 }
 
+// finalize-only: mean/invstd (+ running update) from (C, 2) sums —
+// used by the BN+act+conv fused op, which never materializes the apply
+std::vector<at::Tensor> bn_finalize_only(
+    const at::Tensor& sums, double count,
+    const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, double momentum,
+    double eps) {
+  const int C = sums.size(0);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = sums.options();
+  auto mean = at::empty({C}, opts);
+  auto invstd = at::empty({C}, opts);
+  const bool has_running = running_mean.has_value() && running_mean->defined();
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
+                     0, stream.stream(), sums.data_ptr<float>(), 1, 1,
+                     mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                     has_running ? running_mean->data_ptr<float>() : nullptr,
+                     has_running ? running_var->data_ptr<float>() : nullptr,
+                     C, (long)count, (float)momentum, (float)eps);
+  return {mean, invstd};
+}
+
+// eval-mode dx only (dx = act_grad * gamma * invstd * dy): the act-only
+// fused conv's backward — no dgamma/dbeta reductions
+at::Tensor bn_bwd_dx_eval(const at::Tensor& dy, const at::Tensor& x,
+                          const at::Tensor& mean, const at::Tensor& invstd,
+                          const at::Tensor& gamma, const at::Tensor& beta,
+                          long act) {
+  const int N = x.size(0), C = x.size(1);
+  const long L = x.size(2);
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  auto dx = at::empty_like(x);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, x.scalar_type(),
+      "bn_bwd_dx_eval", [&] {
+        hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, false>),
+                           dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                           dim3(kBlock), 0, stream.stream(),
+                           dy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),
+                           dx.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), g32.data_ptr<float>(),
+                           b32.data_ptr<float>(), nullptr, C, L, total,
+                           (long)N * L, (int)act);
+      });
+  return dx;
+}
+
 at::Tensor bn_bwd_sums_only(const at::Tensor& dy, const at::Tensor& x,
                             const at::Tensor& mean, const at::Tensor& invstd,
                             const at::Tensor& gamma, const at::Tensor& beta,

@@ -142,6 +142,18 @@ std::vector<at::Tensor> pw_conv_fwd_stats(
   return {y, part};
 }
 
+// input-gradient only (dx = W^T @ dy) — used by the fused BN+act+conv op
+at::Tensor pw_conv_dx(const at::Tensor& dy, const at::Tensor& w) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous());
+  const int N = dy.size(0);
+  const long L = dy.size(2);
+  const int Ci = w.size(1);
+  auto dx = at::empty({N, Ci, L}, dy.options());
+  TORCH_CHECK(pw_mfma_gemm(dy, w, c10::nullopt, dx, /*trans=*/true),
+              "pw_conv_dx: bf16 MFMA path required");
+  return dx;
+}
+
 std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,
                                     const at::Tensor& w, bool has_bias) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous());

@@ -43,13 +43,18 @@ constexpr int kXPitch = 136;  // LDS row pitch of the X chunk (bf16)
 // (Co, nsplit, 2), nsplit = N * gridDim.x — the producer half of the
 // conv->BN chain fusion (docs/FUSION_PLAN.md step 1): the following
 // BatchNorm consumes the slab and never re-reads y for its statistics.
-template <bool TRANS, bool HAS_BIAS>
+// PRE: x rows are normalized/activated during staging
+// (z = act(x * pre_scale[c] + pre_shift[c])) so the BN-apply output never
+// exists in HBM — conv->BN fusion step 2 (docs/FUSION_PLAN.md).
+template <bool TRANS, bool HAS_BIAS, bool PRE = false>
 __global__ __launch_bounds__(kBlock)
 void pw_mfma_kernel(const sa_bf16* __restrict__ x,
                     const sa_bf16* __restrict__ w,
                     const sa_bf16* __restrict__ bias,
                     sa_bf16* __restrict__ y,
                     float* __restrict__ stats,
+                    const float* __restrict__ pre_scale,
+                    const float* __restrict__ pre_shift, int pre_act,
                     int N, int Ci, int Co, long L) {
   __shared__ sa_bf16 w_s[kCoT * kWPitch];
   __shared__ sa_bf16 x_s[kKT * kXPitch];
@@ -103,6 +108,14 @@ void pw_mfma_kernel(const sa_bf16* __restrict__ x,
         } else {
           for (int j = 0; j < 8; ++j) {
             v[j] = (lg + j < L) ? xb[(long)kg * L + lg + j] : (sa_bf16)0.f;
+          }
+        }
+        if (PRE) {
+          const float sc = pre_scale[kg];
+          const float sh = pre_shift[kg];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            v[j] = (sa_bf16)sa::act_fwd((float)v[j] * sc + sh, pre_act);
           }
         }
       }
@@ -226,7 +239,7 @@ bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
     hipLaunchKernelGGL((pw_mfma_kernel<decltype(tr_)::value,
                                        decltype(hb_)::value>),
                        grid, dim3(kBlock), 0, stream.stream(), xp, wp, bp,
-                       yp, sp, N, Ci, Co, L);
+                       yp, sp, nullptr, nullptr, 0, N, Ci, Co, L);
   };
   if (trans) {
     if (has_bias) launch(std::true_type{}, std::true_type{});
@@ -810,4 +823,169 @@ std::vector<at::Tensor> pw_conv_multi_dx(const at::Tensor& dy,
                      sizes.size() > 2 ? ob2 : (int)Ct,
                      N, Cin, (int)Ct, L);
   return dxs;
+}
+
+// ---------------------------------------------------------------------------
+// BN(+act)+pointwise fusion entry points (FUSION_PLAN step 2)
+// ---------------------------------------------------------------------------
+
+// y = W @ act(x * scale[c] + shift[c]): the BN-apply output never exists
+at::Tensor pw_conv_pre_fwd(const at::Tensor& x, const at::Tensor& w,
+                           const c10::optional<at::Tensor>& bias,
+                           const at::Tensor& scale, const at::Tensor& shift,
+                           long act) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous()
+              && x.scalar_type() == at::kBFloat16);
+  const int N = x.size(0), Ci = x.size(1);
+  const long L = x.size(2);
+  const int Co = w.size(0);
+  TORCH_CHECK(w.size(1) == Ci && Ci >= 16);
+  TORCH_CHECK(scale.is_contiguous() && shift.is_contiguous()
+              && scale.scalar_type() == at::kFloat);
+  auto y = at::empty({N, Co, L}, x.options());
+  const bool has_bias = bias.has_value() && bias->defined();
+  at::Tensor bct;
+  if (has_bias) bct = bias->to(at::kBFloat16).contiguous();
+  dim3 grid(sa::ceil_div(L, kLT), N, sa::ceil_div(Co, kCoT));
+  auto stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto hb_) {
+    hipLaunchKernelGGL((pw_mfma_kernel<false, decltype(hb_)::value, true>),
+                       grid, dim3(kBlock), 0, stream.stream(),
+                       (const sa_bf16*)x.data_ptr(),
+                       (const sa_bf16*)w.data_ptr(),
+                       has_bias ? (const sa_bf16*)bct.data_ptr() : nullptr,
+                       (sa_bf16*)y.data_ptr(), nullptr,
+                       scale.data_ptr<float>(), shift.data_ptr<float>(),
+                       (int)act, N, Ci, Co, L);
+  };
+  if (has_bias) launch(std::true_type{});
+  else launch(std::false_type{});
+  return y;
+}
+
+// ---------------------------------------------------------------------------
+// split-K weight gradient with the BN transform applied while staging x:
+//   dw[m][c] = sum_{n,l} dy[n,m,l] * act(x[n,c,l]*scale[c]+shift[c])
+// One block per (32m x 32c tile, n); 2x2 waves, each a 16x16 MFMA tile,
+// K-stepping l in 32-chunks. Partials land in a per-sample fp32 slab
+// summed by sum_batch (same reduction the hipblaslt bmm path uses).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int kDwL = 32;  // l per K-step
+
+template <bool PRE>
+__global__ __launch_bounds__(kBlock)
+void pw_dw_kernel(const sa_bf16* __restrict__ dy,
+                  const sa_bf16* __restrict__ x,
+                  const float* __restrict__ pre_scale,
+                  const float* __restrict__ pre_shift, int pre_act,
+                  float* __restrict__ slab,  // (N, Co, Ci)
+                  int N, int Co, int Ci, long L) {
+  __shared__ sa_bf16 dy_s[32 * 40];   // [32 m][32 l] padded
+  __shared__ sa_bf16 x_s[32 * 40];    // [32 c][32 l] padded
+  const int n = blockIdx.y;
+  const int mtiles = (Co + 31) / 32;
+  const int m0 = (blockIdx.x % mtiles) * 32;
+  const int c0 = (blockIdx.x / mtiles) * 32;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = wid >> 1;   // 16-m sub-tile
+  const int wc = wid & 1;    // 16-c sub-tile
+  const int frag_m = lane & 15;
+  const int kbase = (lane >> 4) * 8;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const sa_bf16* dyb = dy + ((long)n * Co) * L;
+  const sa_bf16* xb = x + ((long)n * Ci) * L;
+
+  for (long l0 = 0; l0 < L; l0 += kDwL) {
+    __syncthreads();
+    // stage dy rows [32 m][32 l] and x rows [32 c][32 l]
+    for (int idx = tid; idx < 32 * (kDwL / 8); idx += kBlock) {
+      const int r = idx / (kDwL / 8);
+      const int c8 = idx - r * (kDwL / 8);
+      const long lg = l0 + c8 * 8;
+      bf16x8 vd = {}, vx = {};
+      if (m0 + r < Co) {
+        const sa_bf16* row = dyb + (long)(m0 + r) * L;
+        if (lg + 8 <= L) vd = *(const bf16x8*)(row + lg);
+        else for (int j = 0; j < 8; ++j)
+          vd[j] = (lg + j < L) ? row[lg + j] : (sa_bf16)0.f;
+      }
+      if (c0 + r < Ci) {
+        const sa_bf16* row = xb + (long)(c0 + r) * L;
+        if (lg + 8 <= L) vx = *(const bf16x8*)(row + lg);
+        else for (int j = 0; j < 8; ++j)
+          vx[j] = (lg + j < L) ? row[lg + j] : (sa_bf16)0.f;
+        if (PRE) {
+          const float sc = pre_scale[c0 + r];
+          const float sh = pre_shift[c0 + r];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            vx[j] = (sa_bf16)sa::act_fwd((float)vx[j] * sc + sh, pre_act);
+          }
+        }
+      }
+      *(bf16x8*)(dy_s + r * 40 + c8 * 8) = vd;
+      *(bf16x8*)(x_s + r * 40 + c8 * 8) = vx;
+    }
+    __syncthreads();
+
+    // A[i=m][k=l] from dy_s; B[k=l][j=c] = z[c=j][l=k] — both contiguous
+    // 8-element runs because the LDS images are [row][l]
+    const bf16x8 a = *(const bf16x8*)(dy_s + (wm * 16 + frag_m) * 40 + kbase);
+    const bf16x8 b =
+        *(const bf16x8*)(x_s + (wc * 16 + frag_m) * 40 + kbase);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+
+  // D col = lane&15 (c), row = (lane>>4)*4 + r (m)
+  const int d_col = lane & 15;
+  const int d_row0 = (lane >> 4) * 4;
+  const int cg = c0 + wc * 16 + d_col;
+  if (cg >= Ci) return;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int mg = m0 + wm * 16 + d_row0 + r;
+    if (mg < Co) {
+      slab[((long)n * Co + mg) * Ci + cg] = acc[r];
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor sum_batch(const at::Tensor& in);
+
+// dw = sum_n dy_n @ act(x_n * scale + shift)^T  (fp32 out)
+at::Tensor pw_dw_pre(const at::Tensor& dy, const at::Tensor& x,
+                     const c10::optional<at::Tensor>& scale,
+                     const c10::optional<at::Tensor>& shift, long act) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  const int N = dy.size(0), Co = dy.size(1);
+  const long L = dy.size(2);
+  const int Ci = x.size(1);
+  auto slab = at::empty({N, Co, Ci}, dy.options().dtype(at::kFloat));
+  const int mtiles = (Co + 31) / 32;
+  const int ctiles = (Ci + 31) / 32;
+  dim3 grid(mtiles * ctiles, N);
+  auto stream = at::hip::getCurrentHIPStream();
+  const bool pre = scale.has_value() && scale->defined();
+  if (pre) {
+    hipLaunchKernelGGL((pw_dw_kernel<true>), grid, dim3(kBlock), 0,
+                       stream.stream(), (const sa_bf16*)dy.data_ptr(),
+                       (const sa_bf16*)x.data_ptr(),
+                       scale->data_ptr<float>(), shift->data_ptr<float>(),
+                       (int)act, slab.data_ptr<float>(), N, Co, Ci, L);
+  } else {
+    hipLaunchKernelGGL((pw_dw_kernel<false>), grid, dim3(kBlock), 0,
+                       stream.stream(), (const sa_bf16*)dy.data_ptr(),
+                       (const sa_bf16*)x.data_ptr(), nullptr, nullptr, 0,
+                       slab.data_ptr<float>(), N, Co, Ci, L);
+  }
+  return sum_batch(slab);
 }

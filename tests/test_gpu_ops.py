@@ -586,3 +586,63 @@ def test_pointwise_conv_cat_fused(widths):
     for a, bb in zip(g1, g2):
         assert torch.allclose(a.float(), bb.float(), atol=5e-2), \
             (a - bb).abs().max().item()
+
+
+def test_bn_act_pw_fused_matches_composite():
+    """FUSION_PLAN step 2: BN(+GELU) fused into the consumer pointwise
+    conv == bn_act followed by pointwise_conv (fwd + all grads)."""
+    import torch.nn as nn
+    torch.manual_seed(7)
+    N, C, Co, L = 5, 32, 48, 512
+    x = torch.randn(N, C, L, device="cuda:0", dtype=torch.bfloat16,
+                    requires_grad=True)
+    bn = nn.BatchNorm1d(C).to("cuda:0").train()
+    w = (torch.randn(Co, C, device="cuda:0", dtype=torch.bfloat16) * 0.1
+         ).requires_grad_(True)
+    b = torch.randn(Co, device="cuda:0", dtype=torch.bfloat16
+                    ).requires_grad_(True)
+
+    y1 = ops.bn_act_pw(x, bn, "gelu", w, b)
+    rm1, rv1 = bn.running_mean.clone(), bn.running_var.clone()
+    dy = torch.randn_like(y1)
+    g1 = torch.autograd.grad(y1, [x, bn.weight, bn.bias, w, b], dy)
+
+    bn2 = nn.BatchNorm1d(C).to("cuda:0").train()
+    x2 = x.detach().clone().requires_grad_(True)
+    z = ops.bn_act(x2, bn2.weight, bn2.bias, bn2.running_mean,
+                   bn2.running_var, True, bn2.momentum, bn2.eps, act="gelu")
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    y2 = ops.pointwise_conv(z, w2, b2)
+    g2 = torch.autograd.grad(y2, [x2, bn2.weight, bn2.bias, w2, b2], dy)
+
+    assert torch.allclose(y1.float(), y2.float(), atol=5e-2), \
+        (y1 - y2).abs().max().item()
+    assert torch.allclose(rm1, bn2.running_mean, atol=1e-4)
+    assert torch.allclose(rv1, bn2.running_var, atol=1e-4)
+    names = ["dx", "dgamma", "dbeta", "dw", "db"]
+    for a, bb, nm in zip(g1, g2, names):
+        scale = bb.float().abs().max().item() or 1.0
+        d = (a.float() - bb.float()).abs().max().item() / scale
+        assert d < 5e-2, f"{nm}: rel diff {d}"
+
+
+def test_act_pw_fused_matches_composite():
+    torch.manual_seed(8)
+    N, C, Co, L = 4, 64, 32, 256
+    x = torch.randn(N, C, L, device="cuda:0", dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = (torch.randn(Co, C, device="cuda:0", dtype=torch.bfloat16) * 0.1
+         ).requires_grad_(True)
+    y1 = ops.act_pw(x, "gelu", w, None)
+    dy = torch.randn_like(y1)
+    g1 = torch.autograd.grad(y1, [x, w], dy)
+
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    y2 = ops.pointwise_conv(ops.gelu(x2), w2, None)
+    g2 = torch.autograd.grad(y2, [x2, w2], dy)
+    assert torch.allclose(y1.float(), y2.float(), atol=5e-2)
+    for a, bb in zip(g1, g2):
+        scale = bb.float().abs().max().item() or 1.0
+        assert (a.float() - bb.float()).abs().max().item() / scale < 5e-2
