@@ -825,10 +825,12 @@ def bn_act_pw(x, bn, act, weight, bias):
     if weight.dim() == 3:
         weight = weight.squeeze(-1)
     act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
+    import os
     fusable = (use_native(x) and x.dtype == torch.bfloat16
                and weight.dtype == torch.bfloat16 and x.size(1) >= 16
                and weight.size(0) >= 16  # dx GEMM reduces over Co
-               and not getattr(bn, "_sync_bn", False))
+               and not getattr(bn, "_sync_bn", False)
+               and os.environ.get("SEIST_AMD_NO_PW_FUSION") != "1")
     if not fusable:
         y = bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                    bn.training, bn.momentum, bn.eps, act=act,
@@ -859,9 +861,11 @@ def act_pw(x, act, weight, bias, module=None):
     GELU between lin0 and lin1 — reference models/seist.py:99-121)."""
     if weight.dim() == 3:
         weight = weight.squeeze(-1)
+    import os
     if not (use_native(x) and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and x.size(1) >= 16
-            and weight.size(0) >= 16):
+            and weight.size(0) >= 16
+            and os.environ.get("SEIST_AMD_NO_PW_FUSION") != "1"):
         y = gelu(x) if act == "gelu" else (x.relu() if act == "relu" else x)
         return pointwise_conv(y, weight, bias)
     act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
