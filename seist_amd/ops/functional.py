@@ -830,7 +830,10 @@ def bn_act_pw(x, bn, act, weight, bias):
                and weight.dtype == torch.bfloat16 and x.size(1) >= 16
                and weight.size(0) >= 16  # dx GEMM reduces over Co
                and not getattr(bn, "_sync_bn", False)
-               and os.environ.get("SEIST_AMD_PW_FUSION") == "1")
+               # net win in inference (no backward); A/B-negative in
+               # training (profiles/step_profile_r02.md) -> opt-in there
+               and (not torch.is_grad_enabled()
+                    or os.environ.get("SEIST_AMD_PW_FUSION") == "1"))
     if not fusable:
         y = bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
                    bn.training, bn.momentum, bn.eps, act=act,
@@ -865,7 +868,8 @@ def act_pw(x, act, weight, bias, module=None):
     if not (use_native(x) and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and x.size(1) >= 16
             and weight.size(0) >= 16
-            and os.environ.get("SEIST_AMD_PW_FUSION") == "1"):
+            and (not torch.is_grad_enabled()
+                 or os.environ.get("SEIST_AMD_PW_FUSION") == "1")):
         y = gelu(x) if act == "gelu" else (x.relu() if act == "relu" else x)
         return pointwise_conv(y, weight, bias)
     act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
