@@ -43,8 +43,16 @@ def make_batch(model_name, batch, in_samples, device, dtype, seed):
         t = torch.rand(batch, 1, generator=g) * 5.0
     elif labels == ["pmp"]:  # one-hot polarity
         t = torch.eye(2)[torch.randint(0, 2, (batch,), generator=g)]
+    elif labels == ["clr", "pmp"]:
+        # ditingmotion: (clarity, polarity) one-hot pair
+        t = tuple(torch.eye(2)[torch.randint(0, 2, (batch,), generator=g)]
+                  for _ in labels)
     else:
         t = torch.rand(batch, 3, in_samples, generator=g)
+    if isinstance(t, tuple):
+        return (x.to(device=device, dtype=dtype),
+                tuple(ti.to(device=device, dtype=torch.float32)
+                      for ti in t))
     return (x.to(device=device, dtype=dtype),
             t.to(device=device, dtype=torch.float32))
 
