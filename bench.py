@@ -69,6 +69,9 @@ def main():
                    choices=["bf16", "fp32"])
     p.add_argument("--graph", type=int, default=1,
                    help="capture the step in a hipGraph (1=try, 0=off)")
+    p.add_argument("--lazy-grads", type=int, default=1,
+                   help="FlatReplica lazy mode: steal grads + one foreach "
+                        "pack instead of per-param accumulate adds")
     p.add_argument("--mode", type=str, default="train",
                    choices=["train", "infer"],
                    help="train = full step; infer = forward-only (serving)")
@@ -96,7 +99,8 @@ def main():
         model = convert_to_bf16(model)
     model = model.to(device).train()
 
-    replica = FlatReplica(model)
+    replica = FlatReplica(model, lazy=bool(args.lazy_grads)
+                          and use_cuda)
     bn_tick = manage_bn_counters(model)
     optimizer = FusedAdam(model.parameters(), lr=8e-5)
     loss_fn = Config.get_loss(args.model).to(device)

@@ -60,10 +60,12 @@ class FlatReplica:
         optimizer.step()
     """
 
-    def __init__(self, model: nn.Module, process_group=None):
+    def __init__(self, model: nn.Module, process_group=None,
+                 lazy: bool = False):
         self.model = model
         self.group = process_group
         self.world_size = pdist.get_world_size()
+        self.lazy = lazy
         self.params: List[torch.nn.Parameter] = [
             p for p in model.parameters() if p.requires_grad]
 
@@ -77,21 +79,46 @@ class FlatReplica:
         for p in self.params:
             buckets.setdefault(p.dtype, []).append(p)
         self.buffers = {}
-        self.views = []
+        self._bucket_params = buckets
+        self._views = {}
         for dtype, ps in buckets.items():
             total = sum(p.numel() for p in ps)
             buf = torch.zeros(total, dtype=dtype, device=ps[0].device)
             off = 0
+            views = []
             for p in ps:
-                p.grad = buf[off:off + p.numel()].view_as(p)
+                v = buf[off:off + p.numel()].view_as(p)
+                if not lazy:
+                    p.grad = v
+                views.append(v)
                 off += p.numel()
             self.buffers[dtype] = buf
+            self._views[dtype] = views
 
     def zero_grad(self):
+        if self.lazy:
+            # backward STEALS each gradient (no per-param accumulate-add
+            # kernel); pack() copies them into the flat buffers afterwards
+            for p in self.params:
+                p.grad = None
+            return
         for buf in self.buffers.values():
             buf.zero_()
 
+    def pack(self):
+        """Lazy mode: one _foreach_copy_ of all stolen grads into the flat
+        buffers, then re-alias p.grad to the flat views so the fused
+        optimizer's packed pointers stay stable across steps."""
+        if not self.lazy:
+            return
+        for dtype, ps in self._bucket_params.items():
+            views = self._views[dtype]
+            torch._foreach_copy_(views, [p.grad for p in ps])
+            for p, v in zip(ps, views):
+                p.grad = v
+
     def allreduce(self):
+        self.pack()
         if not pdist.is_dist() or self.world_size == 1:
             return
         for buf in self.buffers.values():

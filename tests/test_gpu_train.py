@@ -155,3 +155,57 @@ def test_postprocess_gpu_matches_cpu(dev):
     a = _detect_event(out.to(dev), 0.8, 2).cpu()
     b = _detect_event(out, 0.8, 2)
     assert torch.equal(a, b)
+
+
+def test_flat_replica_lazy_grads_match(dev):
+    """Lazy FlatReplica (steal + one _foreach_copy_ pack) must produce the
+    same flat-buffer gradients as accumulate-mode, eager AND under
+    hipGraph capture/replay."""
+    import copy
+    from seist_amd.parallel.ddp import FlatReplica
+    from seist_amd.models import create_model
+    from seist_amd.engine.precision import convert_to_bf16
+
+    torch.manual_seed(0)
+    m1 = convert_to_bf16(create_model("seist_s_dpk", in_samples=2048)).to(dev)
+    m2 = copy.deepcopy(m1)
+    m1.train()
+    m2.train()
+    r1 = FlatReplica(m1)
+    r2 = FlatReplica(m2, lazy=True)
+    x = torch.randn(4, 3, 2048, device=dev, dtype=torch.bfloat16)
+
+    def run(rep, model):
+        rep.zero_grad()
+        y = model(x)
+        y.float().pow(2).mean().backward()
+        rep.allreduce()
+
+    run(r1, m1)
+    run(r2, m2)
+    for dt in r1.buffers:
+        d = (r1.buffers[dt].float() - r2.buffers[dt].float()).abs().max()
+        assert d.item() < 1e-5, f"eager lazy mismatch {dt}: {d.item()}"
+
+    # graph capture of the lazy step: replay must rewrite the same grads
+    ref = {dt: b.clone() for dt, b in r2.buffers.items()}
+    s = torch.cuda.Stream()
+    with torch.cuda.stream(s):
+        run(r2, m2)
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        run(r2, m2)
+    for dt in r2.buffers:
+        r2.buffers[dt].zero_()
+    g.replay()
+    torch.cuda.synchronize()
+    for dt in r2.buffers:
+        d = (r2.buffers[dt].float() - ref[dt].float()).abs().max()
+        assert d.item() < 1e-5, f"replay mismatch {dt}: {d.item()}"
+    # second replay must be stable too
+    g.replay()
+    torch.cuda.synchronize()
+    for dt in r2.buffers:
+        d = (r2.buffers[dt].float() - ref[dt].float()).abs().max()
+        assert d.item() < 1e-5, f"replay2 mismatch {dt}: {d.item()}"
