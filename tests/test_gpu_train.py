@@ -184,8 +184,13 @@ def test_flat_replica_lazy_grads_match(dev):
     run(r1, m1)
     run(r2, m2)
     for dt in r1.buffers:
-        d = (r1.buffers[dt].float() - r2.buffers[dt].float()).abs().max()
-        assert d.item() < 1e-5, f"eager lazy mismatch {dt}: {d.item()}"
+        # accumulate-mode and steal-mode round bf16 adds in a different
+        # order; compare relative to the gradient scale
+        scale = r1.buffers[dt].float().abs().max().item() or 1.0
+        d = (r1.buffers[dt].float()
+             - r2.buffers[dt].float()).abs().max().item() / scale
+        tol = 2e-2 if dt == torch.bfloat16 else 1e-5
+        assert d < tol, f"eager lazy mismatch {dt}: {d}"
 
     # graph capture of the lazy step: replay must rewrite the same grads
     ref = {dt: b.clone() for dt, b in r2.buffers.items()}
