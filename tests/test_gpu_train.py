@@ -167,7 +167,12 @@ def test_flat_replica_lazy_grads_match(dev):
     from seist_amd.engine.precision import convert_to_bf16
 
     torch.manual_seed(0)
-    m1 = convert_to_bf16(create_model("seist_s_dpk", in_samples=2048)).to(dev)
+    # zero drop rates: the fused attention draws dropout masks from a
+    # device-resident seed that bumps every forward, so stochastic runs
+    # are not comparable
+    kw = dict(in_samples=2048, path_drop_rate=0.0, attn_drop_rate=0.0,
+              key_drop_rate=0.0, mlp_drop_rate=0.0, other_drop_rate=0.0)
+    m1 = convert_to_bf16(create_model("seist_s_dpk", **kw)).to(dev)
     m2 = copy.deepcopy(m1)
     m1.train()
     m2.train()
