@@ -357,7 +357,9 @@ def train_worker(args, device) -> str:
         else:
             if getattr(args, "sync_bn", True):
                 enable_native_syncbn(model)
-            replica = FlatReplica(model)
+            # lazy: backward steals grads (no per-param accumulate adds);
+            # allreduce() packs them with one _foreach_copy_
+            replica = FlatReplica(model, lazy=torch.cuda.is_available())
 
     ckpt_path = None
     num_saved = 0
