@@ -100,3 +100,27 @@ def test_engine_entry_points():
     from seist_amd.parallel.ddp import FlatReplica, wrap_distributed  # noqa
     from seist_amd.utils.visualization import (  # noqa
         vis_phase_picking, vis_waves_preds_targets)
+
+
+def test_native_op_surface_complete():
+    """K1-K20 op-layer surface: every op family the models
+    dispatch through must exist on seist_amd.ops."""
+    from seist_amd import ops
+    for name in ("pointwise_conv", "pointwise_conv_cat", "conv1d",
+                 "conv_transpose1d", "bn_act", "bn_act_pw", "act_pw",
+                 "avgmax_pool1d", "max_pool1d", "global_avg_pool1d",
+                 "interp_linear", "upsample2x", "pooled_attention",
+                 "droppath_add", "layer_norm", "additive_attention_weights",
+                 "lstm", "gelu", "auto_pad", "FusedAdam"):
+        assert hasattr(ops, name), name
+
+
+def test_distributed_surface_complete():
+    """SURVEY §2.5 C1-C9 obligations."""
+    from seist_amd.parallel import dist as pdist
+    from seist_amd.parallel.ddp import (FlatReplica, enable_native_syncbn,
+                                        wrap_distributed)
+    for fn in ("init_distributed_mode", "reduce_tensor",
+               "gather_tensors_to_list", "broadcast_object", "barrier"):
+        assert hasattr(pdist, fn), fn
+    assert FlatReplica and enable_native_syncbn and wrap_distributed
