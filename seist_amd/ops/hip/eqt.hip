@@ -356,7 +356,14 @@ __global__ void lstm_fwd_kernel(const float* __restrict__ pre,  // (N,L,4H)
   const int gid = (threadIdx.x % ((long)gwaves * sa::kWave));
   const int n = blockIdx.x * spb + slot;
 
-  for (int t = threadIdx.x; t < G4 * H; t += blockDim.x) w[t] = whh[t];
+  // W staged TRANSPOSED [hh][g]: the per-hh inner loop then reads
+  // consecutive g across threads — conflict-free broadcast instead of a
+  // 32-way bank conflict on the row-major stride-H access
+  for (int t = threadIdx.x; t < G4 * H; t += blockDim.x) {
+    const int g_ = t / H;
+    const int hh_ = t - g_ * H;
+    w[hh_ * G4 + g_] = whh[t];
+  }
   float* h = state + (long)slot * (2 * H + G4);
   float* c = h + H;
   float* gates = c + H;
@@ -369,8 +376,7 @@ __global__ void lstm_fwd_kernel(const float* __restrict__ pre,  // (N,L,4H)
     const int t = dir ? (L - 1 - step) : step;
     if (active) {
       float acc = pre[((long)n * L + t) * G4 + g];
-      const float* wr = w + (long)g * H;
-      for (int hh = 0; hh < H; ++hh) acc += wr[hh] * h[hh];
+      for (int hh = 0; hh < H; ++hh) acc += w[hh * G4 + g] * h[hh];
       const int kind = g / H;
       gates[g] = (kind == 2) ? tanhf(acc)
                              : 1.0f / (1.0f + expf(-acc));
@@ -574,6 +580,8 @@ __global__ void lstm_bwd_kernel(const float* __restrict__ dy,  // (N,L,D*H)
     }
     __syncthreads();
     if (active && g < H) {
+      // w[gg*H + g]: per gg, threads h read consecutive addresses —
+      // already conflict-free in the row-major image
       float acc = 0.0f;
       for (int gg = 0; gg < G4; ++gg) acc += w[(long)gg * H + g] * dg[gg];
       dh[g] = acc;
