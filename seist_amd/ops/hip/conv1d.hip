@@ -24,6 +24,8 @@
 
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor sum_mid(const at::Tensor& in);
+at::Tensor sum_mid_to(const at::Tensor& in, at::ScalarType out_dtype);
+at::Tensor channel_sum_to(const at::Tensor& in, at::ScalarType out_dtype);
 bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
                long padl, long dilation, bool is_dx,
@@ -597,14 +599,14 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
       auto out_k = slab.select(0, k);
       at::bmm_out(out_k, dyv, xv.transpose(1, 2));
     }
-    auto dw = sum_mid(slab.view({(long)K, (long)N, (long)Co * Cig}))
+    auto dw = sum_mid_to(slab.view({(long)K, (long)N, (long)Co * Cig}),
+                         w.scalar_type())
                   .view({(long)K, (long)Co, (long)Cig})
                   .permute({1, 2, 0})
-                  .contiguous()
-                  .to(w.scalar_type());
+                  .contiguous();
     at::Tensor db;
     if (has_bias) {
-      db = channel_sum(dy).to(w.scalar_type());
+      db = channel_sum_to(dy, w.scalar_type());
     }
     return {dw, db};
   }
@@ -612,7 +614,7 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
   // grouped/depthwise stride-1 convs: matrix-core weight gradient
   if (auto dwm = dw_mfma_try(dy, x, stride, padl, groups, dilation, K)) {
     at::Tensor db;
-    if (has_bias) db = channel_sum(dy).to(w.scalar_type());
+    if (has_bias) db = channel_sum_to(dy, w.scalar_type());
     return {dwm->to(w.scalar_type()), db};
   }
 
@@ -713,6 +715,6 @@ std::vector<at::Tensor> conv_transpose1d_bwd(const at::Tensor& dy,
                           /*has_bias=*/false);
   auto dw = dws[0];
   at::Tensor db;
-  if (has_bias) db = channel_sum(dy).to(w.scalar_type());
+  if (has_bias) db = channel_sum_to(dy, w.scalar_type());
   return {dx, dw, db};
 }

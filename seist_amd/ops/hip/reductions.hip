@@ -38,8 +38,9 @@ __global__ void sum_batch_kernel(const scalar_t* __restrict__ in,
 // conv), so column-only parallelism leaves the chip idle — the 4 z-slices
 // quadruple the block count and the per-thread serial depth drops to
 // nsplit/4.
+template <typename out_t>
 __global__ void sum_batch_final_kernel(const float* __restrict__ part,
-                                       float* __restrict__ out,
+                                       out_t* __restrict__ out,
                                        int nsplit, long M) {
   __shared__ float red[8][32];
   const int jt = threadIdx.x & 31;
@@ -55,7 +56,7 @@ __global__ void sum_batch_final_kernel(const float* __restrict__ part,
     float t = 0.0f;
 #pragma unroll
     for (int z = 0; z < 8; ++z) t += red[z][jt];
-    out[j] = t;
+    out[j] = (out_t)t;
   }
 }
 
@@ -80,8 +81,9 @@ __global__ void channel_sum_kernel(const scalar_t* __restrict__ in,
 }
 
 // one wave per channel (a thread-per-channel serial loop is latency-bound)
+template <typename out_t>
 __global__ void part_sum_kernel(const float* __restrict__ part,
-                                float* __restrict__ out, int C, int nsplit) {
+                                out_t* __restrict__ out, int C, int nsplit) {
   const int lane = threadIdx.x & (sa::kWave - 1);
   const int c = blockIdx.x * (blockDim.x / sa::kWave)
                 + threadIdx.x / sa::kWave;
@@ -91,12 +93,12 @@ __global__ void part_sum_kernel(const float* __restrict__ part,
     s += part[(long)c * nsplit + j];
   }
   s = sa::warp_reduce_sum(s);
-  if (lane == 0) out[c] = s;
+  if (lane == 0) out[c] = (out_t)s;
 }
 
-template <typename scalar_t>
+template <typename scalar_t, typename out_t>
 __global__ void sum_mid_kernel(const scalar_t* __restrict__ in,
-                               float* __restrict__ out,
+                               out_t* __restrict__ out,
                                long A, long B, long M) {
   const long i = (long)blockIdx.x * kBlock + threadIdx.x;  // a*M + j
   if (i >= A * M) return;
@@ -110,38 +112,49 @@ __global__ void sum_mid_kernel(const scalar_t* __restrict__ in,
          + (float)p[(b + 2) * M] + (float)p[(b + 3) * M];
   }
   for (; b < B; ++b) s += (float)p[b * M];
-  out[i] = s;
+  out[i] = (out_t)s;
 }
 
 }  // namespace
 
 // (A, B, M) -> (A, M) fp32, middle axis reduced — used to collapse the
 // batch axis of all K tap-GEMMs of a conv weight gradient in one launch.
-at::Tensor sum_mid(const at::Tensor& in) {
+at::Tensor sum_mid_to(const at::Tensor& in, at::ScalarType out_dtype) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() >= 3);
   const long A = in.size(0);
   const long B = in.size(1);
   const long M = in.numel() / (A * B);
-  auto out = at::empty({A, M}, in.options().dtype(at::kFloat));
+  auto out = at::empty({A, M}, in.options().dtype(out_dtype));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, in.scalar_type(),
       "sum_mid", [&] {
-        hipLaunchKernelGGL((sum_mid_kernel<scalar_t>),
-                           dim3(sa::ceil_div(A * M, kBlock)), dim3(kBlock),
-                           0, stream.stream(), in.data_ptr<scalar_t>(),
-                           out.data_ptr<float>(), A, B, M);
+        using in_t = scalar_t;
+        AT_DISPATCH_FLOATING_TYPES_AND2(
+            at::ScalarType::BFloat16, at::ScalarType::Half, out_dtype,
+            "sum_mid_out", [&] {
+              using out_t = scalar_t;
+              hipLaunchKernelGGL((sum_mid_kernel<in_t, out_t>),
+                                 dim3(sa::ceil_div(A * M, kBlock)),
+                                 dim3(kBlock), 0, stream.stream(),
+                                 in.data_ptr<in_t>(),
+                                 out.data_ptr<out_t>(), A, B, M);
+            });
       });
   return out;
 }
 
-at::Tensor sum_batch(const at::Tensor& in) {
+at::Tensor sum_mid(const at::Tensor& in) {
+  return sum_mid_to(in, at::kFloat);
+}
+
+at::Tensor sum_batch_to(const at::Tensor& in, at::ScalarType out_dtype) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() >= 2);
   const long B = in.size(0);
   const long M = in.numel() / B;
   auto out_sizes = in.sizes().vec();
   out_sizes.erase(out_sizes.begin());
-  auto out = at::empty(out_sizes, in.options().dtype(at::kFloat));
+  auto out = at::empty(out_sizes, in.options().dtype(out_dtype));
   auto stream = at::hip::getCurrentHIPStream();
   // target ~512 blocks total, nsplit bounded so the final pass stays tiny
   const long col_blocks = std::max<long>(M / kBlock, 1);
@@ -157,14 +170,22 @@ at::Tensor sum_batch(const at::Tensor& in) {
                            stream.stream(), in.data_ptr<scalar_t>(),
                            part.data_ptr<float>(), B, M);
       });
-  hipLaunchKernelGGL(sum_batch_final_kernel,
-                     dim3(sa::ceil_div(M, 32)), dim3(kBlock), 0,
-                     stream.stream(), part.data_ptr<float>(),
-                     out.data_ptr<float>(), nsplit, M);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, out_dtype,
+      "sum_batch_final", [&] {
+        hipLaunchKernelGGL((sum_batch_final_kernel<scalar_t>),
+                           dim3(sa::ceil_div(M, 32)), dim3(kBlock), 0,
+                           stream.stream(), part.data_ptr<float>(),
+                           out.data_ptr<scalar_t>(), nsplit, M);
+      });
   return out;
 }
 
-at::Tensor channel_sum(const at::Tensor& in) {
+at::Tensor sum_batch(const at::Tensor& in) {
+  return sum_batch_to(in, at::kFloat);
+}
+
+at::Tensor channel_sum_to(const at::Tensor& in, at::ScalarType out_dtype) {
   TORCH_CHECK(in.is_cuda() && in.is_contiguous() && in.dim() == 3);
   const long N = in.size(0);
   const int C = in.size(1);
@@ -172,7 +193,7 @@ at::Tensor channel_sum(const at::Tensor& in) {
   const int nsplit = std::max(1, std::min<int>(
       (int)N, 2048 / std::max(C, 1)));
   auto part = at::empty({C, nsplit}, in.options().dtype(at::kFloat));
-  auto out = at::empty({C}, in.options().dtype(at::kFloat));
+  auto out = at::empty({C}, in.options().dtype(out_dtype));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, in.scalar_type(),
@@ -182,8 +203,17 @@ at::Tensor channel_sum(const at::Tensor& in) {
                            in.data_ptr<scalar_t>(), part.data_ptr<float>(),
                            C, N, L);
       });
-  hipLaunchKernelGGL(part_sum_kernel, dim3(sa::ceil_div(C, 4)), dim3(256),
-                     0, stream.stream(), part.data_ptr<float>(),
-                     out.data_ptr<float>(), C, nsplit);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, out_dtype,
+      "part_sum", [&] {
+        hipLaunchKernelGGL((part_sum_kernel<scalar_t>),
+                           dim3(sa::ceil_div(C, 4)), dim3(256), 0,
+                           stream.stream(), part.data_ptr<float>(),
+                           out.data_ptr<scalar_t>(), C, nsplit);
+      });
   return out;
+}
+
+at::Tensor channel_sum(const at::Tensor& in) {
+  return channel_sum_to(in, at::kFloat);
 }
