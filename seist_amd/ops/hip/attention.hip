@@ -438,6 +438,232 @@ unsigned long long* attn_seed_state(const at::Tensor& like) {
 
 }  // namespace
 
+
+// ---------------------------------------------------------------------------
+// MFMA training forward: the per-thread VALU loops of
+// pooled_attn_tfwd_kernel measured ~25x off roofline; here QK^T and PV run
+// on v_mfma_f32_16x16x32_bf16 wave tiles. One block = 4 waves, each owning
+// a 16-query tile of one (n, h); K/V staged once per block. Saves the SAME
+// (m, denom) stats and bit-packed dropout mask as the VALU kernel, so the
+// flash-style backward kernels are unchanged.
+// Envelope: bf16, E in {8,16,32}, Lk % 32 == 0, Lk <= 256, Lq % 16 == 0.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+typedef __bf16 sa_bf16_a;
+typedef sa_bf16_a bf16x8a __attribute__((ext_vector_type(8)));
+typedef float f32x4a __attribute__((ext_vector_type(4)));
+
+constexpr int kAT = 16;        // queries per wave tile
+constexpr int kAW = 4;         // waves per block
+constexpr int kMaxNF = 16;     // Lk/16 <= 16
+
+// NF = Lk/16 as a template constant: the accumulator array is indexed in
+// fully-unrolled loops and stays in registers (a runtime bound would
+// spill it to scratch)
+template <int NF, bool DROP>
+__global__ __launch_bounds__(256)
+void pooled_attn_tfwd_mfma_kernel(const sa_bf16_a* __restrict__ q,
+                                  const sa_bf16_a* __restrict__ k,
+                                  const sa_bf16_a* __restrict__ v,
+                                  sa_bf16_a* __restrict__ out,
+                                  float* __restrict__ stats,
+                                  unsigned* __restrict__ mask,
+                                  const unsigned long long* seed_ptr,
+                                  long Lq, int E,
+                                  float scale, float p, float inv_keep) {
+  constexpr int Lk = NF * 16;
+  constexpr int W = (NF + 1) / 2;
+  extern __shared__ sa_bf16_a smem[];
+  constexpr int Lkp = Lk + 8;
+  sa_bf16_a* k_s = smem;                       // [32][Lkp]
+  sa_bf16_a* v_s = k_s + 32 * Lkp;             // [32][Lkp]
+  sa_bf16_a* q_s = v_s + 32 * Lkp;             // [kAW][16][32]
+  sa_bf16_a* p_s = q_s + kAW * 16 * 32;        // [kAW][16][Lkp]
+
+  const long nh = blockIdx.y;
+  const long lq0 = (long)blockIdx.x * (kAW * kAT);
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int col = lane & 15;
+  const int grp = lane >> 4;           // 0..3
+  const int kbase = grp * 8;
+
+  // ---- stage K/V as [e][key] (zero-padded rows e >= E) ----
+  for (int idx = tid; idx < 32 * (Lkp / 8); idx += 256) {
+    const int e = idx / (Lkp / 8);
+    const int c8 = (idx - e * (Lkp / 8)) * 8;
+    bf16x8a kv = {}, vv = {};
+    if (e < E && c8 + 8 <= Lk) {
+      kv = *(const bf16x8a*)(k + (nh * E + e) * Lk + c8);
+      vv = *(const bf16x8a*)(v + (nh * E + e) * Lk + c8);
+    } else if (e < E) {
+      for (int j = 0; j < 8; ++j) {
+        if (c8 + j < Lk) {
+          kv[j] = k[(nh * E + e) * Lk + c8 + j];
+          vv[j] = v[(nh * E + e) * Lk + c8 + j];
+        }
+      }
+    }
+    *(bf16x8a*)(k_s + e * Lkp + c8) = kv;
+    *(bf16x8a*)(v_s + e * Lkp + c8) = vv;
+  }
+  // ---- stage this block's 64 q rows transposed to [lq][e]
+  // (e-major loop: consecutive threads read consecutive lq — coalesced;
+  // the transposed LDS writes are cheap) ----
+  for (int idx = tid; idx < 32 * 64; idx += 256) {
+    const int e = idx >> 6;
+    const int r = idx & 63;
+    const long lq = lq0 + r;
+    q_s[r * 32 + e] = (e < E && lq < Lq) ? q[(nh * E + e) * Lq + lq]
+                                         : (sa_bf16_a)0.0f;
+  }
+  __syncthreads();
+
+  // ---- scores: one MFMA per 16-key fragment ----
+  const bf16x8a a_q =
+      *(const bf16x8a*)(q_s + (wid * 16 + col) * 32 + kbase);
+  f32x4a acc[NF];
+#pragma unroll
+  for (int f = 0; f < NF; ++f) {
+    bf16x8a b;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      b[j] = k_s[(kbase + j) * Lkp + f * 16 + col];
+    }
+    acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+        a_q, b, (f32x4a){0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+  }
+
+  // ---- softmax rows (row = grp*4 + r local query), dropout, P to LDS ----
+  const unsigned long long seed = DROP ? *seed_ptr : 0ull;
+#pragma unroll 4
+  for (int r = 0; r < 4; ++r) {
+    const int row = grp * 4 + r;
+    const long lq = lq0 + wid * kAT + row;
+    float m = -INFINITY;
+#pragma unroll
+    for (int f = 0; f < NF; ++f) m = fmaxf(m, acc[f][r] * scale);
+#pragma unroll
+    for (int b = 1; b < 16; b <<= 1) {
+      m = fmaxf(m, __shfl_xor(m, b, 64));
+    }
+    float den = 0.0f;
+#pragma unroll
+    for (int f = 0; f < NF; ++f) {
+      const float u = __expf(acc[f][r] * scale - m);
+      acc[f][r] = u;    // reuse accumulator storage for u
+      den += u;
+    }
+#pragma unroll
+    for (int b = 1; b < 16; b <<= 1) {
+      den += __shfl_xor(den, b, 64);
+    }
+    const float invd = 1.0f / den;
+    unsigned keepbits = 0xffffffffu;   // bit f = keep(this lane's key in f)
+    if (DROP) {
+      keepbits = 0u;
+      const unsigned long long ridx0 = (nh * Lq + lq) * (unsigned long long)Lk;
+#pragma unroll
+      for (int f = 0; f < NF; ++f) {
+        const bool keep = rng_uniform(seed, ridx0 + f * 16 + col) >= p;
+        if (keep) keepbits |= 1u << f;
+      }
+    }
+#pragma unroll
+    for (int f = 0; f < NF; ++f) {
+      float pv = acc[f][r] * invd;
+      if (DROP) pv = (keepbits >> f) & 1u ? pv * inv_keep : 0.0f;
+      p_s[(wid * 16 + row) * Lkp + f * 16 + col] = (sa_bf16_a)pv;
+    }
+    if (col == 0 && lq < Lq) {
+      stats[(nh * Lq + lq) * 2 + 0] = m;
+      stats[(nh * Lq + lq) * 2 + 1] = den;
+    }
+    if (DROP) {
+      // assemble 32-bit mask words from the 16-lane groups via ballots
+#pragma unroll
+      for (int w = 0; w < W; ++w) {
+        const unsigned long long b0 =
+            __ballot((keepbits >> (2 * w)) & 1u);
+        const unsigned long long b1 = (2 * w + 1 < NF)
+            ? __ballot((keepbits >> (2 * w + 1)) & 1u) : 0ull;
+        if (col == 0 && lq < Lq) {
+          const unsigned lo = (unsigned)((b0 >> (16 * grp)) & 0xffffull);
+          const unsigned hi = (unsigned)((b1 >> (16 * grp)) & 0xffffull);
+          mask[(nh * Lq + lq) * W + w] = lo | (hi << 16);
+        }
+      }
+    }
+  }
+  __syncthreads();
+
+  // ---- PV: out tile (16 lq x E) ----
+  const int NE = (E + 15) / 16;
+  f32x4a acc_o[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  for (int kk = 0; kk < Lk; kk += 32) {
+    const bf16x8a a_p =
+        *(const bf16x8a*)(p_s + (wid * 16 + col) * Lkp + kk + kbase);
+    for (int ne = 0; ne < NE; ++ne) {
+      bf16x8a b =
+          *(const bf16x8a*)(v_s + (ne * 16 + col) * Lkp + kk + kbase);
+      acc_o[ne] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_p, b, acc_o[ne], 0, 0, 0);
+    }
+  }
+  for (int ne = 0; ne < NE; ++ne) {
+    const int e = ne * 16 + col;
+    if (e >= E) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long lq = lq0 + wid * kAT + grp * 4 + r;
+      if (lq < Lq) {
+        out[(nh * E + e) * Lq + lq] = (sa_bf16_a)acc_o[ne][r];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// MFMA-path host dispatch; returns false outside the envelope
+bool pooled_attn_tfwd_mfma(const at::Tensor& q, const at::Tensor& k,
+                           const at::Tensor& v, at::Tensor& out,
+                           at::Tensor& stats, at::Tensor& mask, bool drop,
+                           unsigned long long* seed, long NH, int E,
+                           long Lq, int Lk, float scale, float p,
+                           float inv_keep) {
+  if (q.scalar_type() != at::kBFloat16) return false;
+  if (Lk != 64 && Lk != 128 && Lk != 256) return false;
+  if (Lq % 16 != 0) return false;
+  const int Lkp = Lk + 8;
+  const size_t lds = sizeof(sa_bf16_a)
+      * (2 * 32 * Lkp + kAW * 16 * 32 + kAW * 16 * Lkp);
+  dim3 grid(sa::ceil_div(Lq, (long)(kAW * kAT)), NH);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto launch = [&](auto nf_, auto d_) {
+    hipLaunchKernelGGL(
+        (pooled_attn_tfwd_mfma_kernel<decltype(nf_)::value,
+                                      decltype(d_)::value>),
+        grid, dim3(256), lds, stream.stream(),
+        (const sa_bf16_a*)q.data_ptr(), (const sa_bf16_a*)k.data_ptr(),
+        (const sa_bf16_a*)v.data_ptr(), (sa_bf16_a*)out.data_ptr(),
+        stats.data_ptr<float>(),
+        drop ? (unsigned*)mask.data_ptr<int>() : nullptr, seed, Lq, E,
+        scale, p, inv_keep);
+  };
+  auto launch_nf = [&](auto nf_) {
+    if (drop) launch(nf_, std::true_type{});
+    else launch(nf_, std::false_type{});
+  };
+  if (Lk == 64) launch_nf(std::integral_constant<int, 4>{});
+  else if (Lk == 128) launch_nf(std::integral_constant<int, 8>{});
+  else launch_nf(std::integral_constant<int, 16>{});
+  return true;
+}
+
 std::vector<at::Tensor> pooled_attn_train_fwd(const at::Tensor& q,
                                               const at::Tensor& k,
                                               const at::Tensor& v,
@@ -465,6 +691,14 @@ std::vector<at::Tensor> pooled_attn_train_fwd(const at::Tensor& q,
   dim3 grid(sa::ceil_div(Lq, kBlock), N * H);
   auto stream = at::hip::getCurrentHIPStream();
   unsigned long long* seed = attn_seed_state(q);
+  if (pooled_attn_tfwd_mfma(q, k, v, out, stats, mask, drop, seed, N * H,
+                            E, Lq, Lk, scale, (float)p, inv_keep)) {
+    if (drop) {
+      hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0,
+                         stream.stream(), seed);
+    }
+    return {out, stats, mask};
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, q.scalar_type(),
       "pooled_attn_tfwd", [&] {
@@ -494,6 +728,293 @@ std::vector<at::Tensor> pooled_attn_train_fwd(const at::Tensor& q,
   return {out, stats, mask};
 }
 
+
+// ---------------------------------------------------------------------------
+// Single-pass MFMA backward: one block per (n, h) sweeps Lq in 64-query
+// chunks, recomputing P from the saved (m, denom) stats (no reductions),
+// emitting dQ per chunk and accumulating the dK/dV wave tiles across the
+// sweep — replaces the per-query and per-key VALU kernels in one launch.
+// Envelope: bf16, E in {8,16,32}, Lk in {64,128}, Lq % 16 == 0.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+// delta[nh, lq] = sum_e dout * out — lane-per-lq, coalesced per e
+__global__ void attn_delta_kernel(const sa_bf16_a* __restrict__ out,
+                                  const sa_bf16_a* __restrict__ dout,
+                                  float* __restrict__ delta,
+                                  long NHLq, long Lq, int E) {
+  const long i = (long)blockIdx.x * 256 + threadIdx.x;
+  if (i >= NHLq) return;
+  const long nh = i / Lq;
+  const long lq = i - nh * Lq;
+  float s = 0.0f;
+  for (int e = 0; e < E; ++e) {
+    const long o = (nh * E + e) * Lq + lq;
+    s += (float)dout[o] * (float)out[o];
+  }
+  delta[i] = s;
+}
+
+template <int NF, bool DROP>
+__global__ __launch_bounds__(256)
+void pooled_attn_bwd_mfma_kernel(const sa_bf16_a* __restrict__ q,
+                                 const sa_bf16_a* __restrict__ k,
+                                 const sa_bf16_a* __restrict__ v,
+                                 const sa_bf16_a* __restrict__ out,
+                                 const sa_bf16_a* __restrict__ dout,
+                                 const float* __restrict__ delta,
+                                 const float* __restrict__ stats,
+                                 const unsigned* __restrict__ mask,
+                                 sa_bf16_a* __restrict__ dq,
+                                 sa_bf16_a* __restrict__ dk,
+                                 sa_bf16_a* __restrict__ dv,
+                                 long Lq, int E, float scale,
+                                 float inv_keep) {
+  constexpr int Lk = NF * 16;
+  constexpr int W = (NF + 1) / 2;
+  constexpr int Lkp = Lk + 8;
+  extern __shared__ sa_bf16_a smem[];
+  sa_bf16_a* k_s = smem;                       // [32][Lkp]
+  sa_bf16_a* v_s = k_s + 32 * Lkp;             // [32][Lkp]
+  sa_bf16_a* q_s = v_s + 32 * Lkp;             // [64][32]  (chunk, [lq][e])
+  sa_bf16_a* dO_s = q_s + 64 * 32;             // [64][32]
+  sa_bf16_a* p_s = dO_s + 64 * 32;             // [64][Lkp]
+  sa_bf16_a* ds_s = p_s + 64 * Lkp;            // [64][Lkp]
+  float* md_s = (float*)(ds_s + 64 * Lkp);     // [64][2] m, invden
+  float* delta_s = md_s + 64 * 2;              // [64]
+  unsigned* mk_s = (unsigned*)(delta_s + 64);  // [64][W]
+
+  const long nh = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int col = lane & 15;
+  const int grp = lane >> 4;
+  const int kbase = grp * 8;
+  const int NE = (E + 15) / 16;
+
+  // ---- stage K/V [e][key], zero-padded ----
+  for (int idx = tid; idx < 32 * (Lkp / 8); idx += 256) {
+    const int e = idx / (Lkp / 8);
+    const int c8 = (idx - e * (Lkp / 8)) * 8;
+    bf16x8a kv = {}, vv = {};
+    if (e < E && c8 + 8 <= Lk) {
+      kv = *(const bf16x8a*)(k + (nh * E + e) * Lk + c8);
+      vv = *(const bf16x8a*)(v + (nh * E + e) * Lk + c8);
+    } else if (e < E) {
+      for (int j = 0; j < 8; ++j) {
+        if (c8 + j < Lk) {
+          kv[j] = k[(nh * E + e) * Lk + c8 + j];
+          vv[j] = v[(nh * E + e) * Lk + c8 + j];
+        }
+      }
+    }
+    *(bf16x8a*)(k_s + e * Lkp + c8) = kv;
+    *(bf16x8a*)(v_s + e * Lkp + c8) = vv;
+  }
+
+  // dK/dV accumulators: tiles t = wid + 4*i over (f = t % NF, ne = t / NF)
+  f32x4a acc_dv[4] = {{0,0,0,0},{0,0,0,0},{0,0,0,0},{0,0,0,0}};
+  f32x4a acc_dk[4] = {{0,0,0,0},{0,0,0,0},{0,0,0,0},{0,0,0,0}};
+  const int NT = NF * NE;
+
+  for (long lq0 = 0; lq0 < Lq; lq0 += 64) {
+    __syncthreads();
+    // ---- stage chunk: q/dOut transposed [lq][e] with e-major loops
+    // (coalesced lq reads); delta comes precomputed ----
+    for (int idx = tid; idx < 32 * 64; idx += 256) {
+      const int e = idx >> 6;
+      const int r = idx & 63;
+      const long lq = lq0 + r;
+      sa_bf16_a qv = (sa_bf16_a)0.0f, dov = (sa_bf16_a)0.0f;
+      if (e < E && lq < Lq) {
+        qv = q[(nh * E + e) * Lq + lq];
+        dov = dout[(nh * E + e) * Lq + lq];
+      }
+      q_s[r * 32 + e] = qv;
+      dO_s[r * 32 + e] = dov;
+    }
+    if (tid < 64) {
+      const long lq = lq0 + tid;
+      float dlt = 0.0f, m = 0.0f, invd = 1.0f;
+      if (lq < Lq) {
+        dlt = delta[nh * Lq + lq];
+        m = stats[(nh * Lq + lq) * 2 + 0];
+        invd = 1.0f / stats[(nh * Lq + lq) * 2 + 1];
+      }
+      delta_s[tid] = dlt;
+      md_s[tid * 2 + 0] = m;
+      md_s[tid * 2 + 1] = invd;
+    }
+    if (DROP) {
+      for (int idx = tid; idx < 64 * W; idx += 256) {
+        const long lq = lq0 + idx / W;
+        mk_s[idx] = (lq < Lq) ? mask[(nh * Lq + lq) * W + (idx % W)] : 0u;
+      }
+    }
+    __syncthreads();
+
+    // ---- scores + P~ and dP + dS for this wave's 16 lq rows ----
+    const bf16x8a a_q =
+        *(const bf16x8a*)(q_s + (wid * 16 + col) * 32 + kbase);
+    const bf16x8a a_do =
+        *(const bf16x8a*)(dO_s + (wid * 16 + col) * 32 + kbase);
+    f32x4a acc_s[NF], acc_dp[NF];
+#pragma unroll
+    for (int f = 0; f < NF; ++f) {
+      bf16x8a bk, bv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        bk[j] = k_s[(kbase + j) * Lkp + f * 16 + col];
+        bv[j] = v_s[(kbase + j) * Lkp + f * 16 + col];
+      }
+      acc_s[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_q, bk, (f32x4a){0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+      acc_dp[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+          a_do, bv, (f32x4a){0.f, 0.f, 0.f, 0.f}, 0, 0, 0);
+    }
+#pragma unroll 4
+    for (int r = 0; r < 4; ++r) {
+      const int row = grp * 4 + r;
+      const float m = md_s[(wid * 16 + row) * 2 + 0];
+      const float invd = md_s[(wid * 16 + row) * 2 + 1];
+      const float dlt = delta_s[wid * 16 + row];
+#pragma unroll
+      for (int f = 0; f < NF; ++f) {
+        const float P = __expf(acc_s[f][r] * scale - m) * invd;
+        float dP = acc_dp[f][r];
+        float Pm = P;
+        if (DROP) {
+          const unsigned wbits = mk_s[(wid * 16 + row) * W + (f >> 1)];
+          const bool keep = (wbits >> (((f & 1) << 4) + col)) & 1u;
+          dP = keep ? dP * inv_keep : 0.0f;
+          Pm = keep ? P * inv_keep : 0.0f;
+        }
+        const float dS = P * (dP - dlt);
+        p_s[(wid * 16 + row) * Lkp + f * 16 + col] = (sa_bf16_a)Pm;
+        ds_s[(wid * 16 + row) * Lkp + f * 16 + col] = (sa_bf16_a)dS;
+      }
+    }
+    __syncthreads();
+
+    // ---- dQ chunk: dq[lq, e] = scale * sum_j dS[lq,j] k[e,j] ----
+    for (int ne = 0; ne < 2; ++ne) {
+      if (ne >= NE) break;
+      f32x4a acc_q = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < Lk; kk += 32) {
+        const bf16x8a a_ds =
+            *(const bf16x8a*)(ds_s + (wid * 16 + col) * Lkp + kk + kbase);
+        bf16x8a b =
+            *(const bf16x8a*)(k_s + (ne * 16 + col) * Lkp + kk + kbase);
+        acc_q = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b, acc_q,
+                                                        0, 0, 0);
+      }
+      const int e = ne * 16 + col;
+      if (e < E) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long lq = lq0 + wid * 16 + grp * 4 + r;
+          if (lq < Lq) {
+            dq[(nh * E + e) * Lq + lq] = (sa_bf16_a)(acc_q[r] * scale);
+          }
+        }
+      }
+    }
+
+    // ---- accumulate dV/dK tiles: A = P~^T / dS^T, B = dO / q ----
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int t = wid + 4 * i;
+      if (t >= NT) break;
+      const int f = t % NF;
+      const int ne = t / NF;
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {       // lq 32-chunks of the 64
+        bf16x8a a_p, a_ds, b_do, b_q;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int lql = it * 32 + kbase + j;
+          a_p[j] = p_s[lql * Lkp + f * 16 + col];
+          a_ds[j] = ds_s[lql * Lkp + f * 16 + col];
+          b_do[j] = dO_s[lql * 32 + ne * 16 + col];
+          b_q[j] = q_s[lql * 32 + ne * 16 + col];
+        }
+        acc_dv[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_p, b_do, acc_dv[i], 0, 0, 0);
+        acc_dk[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_ds, b_q, acc_dk[i], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: dV/dK tiles (D col = e, row = key) ----
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int t = wid + 4 * i;
+    if (t >= NT) break;
+    const int f = t % NF;
+    const int ne = t / NF;
+    const int e = ne * 16 + col;
+    if (e >= E) continue;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = f * 16 + grp * 4 + r;
+      dv[(nh * E + e) * Lk + key] = (sa_bf16_a)acc_dv[i][r];
+      dk[(nh * E + e) * Lk + key] = (sa_bf16_a)(acc_dk[i][r] * scale);
+    }
+  }
+}
+
+}  // namespace
+
+// single-launch MFMA backward; returns false outside the envelope
+bool pooled_attn_bwd_mfma(const at::Tensor& q, const at::Tensor& k,
+                          const at::Tensor& v, const at::Tensor& out,
+                          const at::Tensor& dout, const at::Tensor& stats,
+                          const at::Tensor& mask, bool drop,
+                          at::Tensor& dq, at::Tensor& dk, at::Tensor& dv,
+                          long NH, int E, long Lq, int Lk, float scale,
+                          float inv_keep) {
+  if (q.scalar_type() != at::kBFloat16) return false;
+  if (Lk != 64 && Lk != 128) return false;
+  if (Lq % 16 != 0) return false;
+  const int Lkp = Lk + 8;
+  const int W = (Lk / 16 + 1) / 2;
+  const size_t lds = sizeof(sa_bf16_a)
+          * (2 * 32 * Lkp + 2 * 64 * 32 + 2 * 64 * Lkp)
+      + sizeof(float) * (64 * 2 + 64) + sizeof(unsigned) * 64 * W;
+  dim3 grid(NH);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto delta = at::empty({NH, Lq}, q.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(attn_delta_kernel,
+                     dim3(sa::ceil_div(NH * Lq, (long)256)), dim3(256), 0,
+                     stream.stream(), (const sa_bf16_a*)out.data_ptr(),
+                     (const sa_bf16_a*)dout.data_ptr(),
+                     delta.data_ptr<float>(), NH * Lq, Lq, E);
+  auto launch = [&](auto nf_, auto d_) {
+    hipLaunchKernelGGL(
+        (pooled_attn_bwd_mfma_kernel<decltype(nf_)::value,
+                                     decltype(d_)::value>),
+        grid, dim3(256), lds, stream.stream(),
+        (const sa_bf16_a*)q.data_ptr(), (const sa_bf16_a*)k.data_ptr(),
+        (const sa_bf16_a*)v.data_ptr(), (const sa_bf16_a*)out.data_ptr(),
+        (const sa_bf16_a*)dout.data_ptr(), delta.data_ptr<float>(),
+        stats.data_ptr<float>(),
+        drop ? (const unsigned*)mask.data_ptr<int>() : nullptr,
+        (sa_bf16_a*)dq.data_ptr(), (sa_bf16_a*)dk.data_ptr(),
+        (sa_bf16_a*)dv.data_ptr(), Lq, E, scale, inv_keep);
+  };
+  auto launch_nf = [&](auto nf_) {
+    if (drop) launch(nf_, std::true_type{});
+    else launch(nf_, std::false_type{});
+  };
+  if (Lk == 64) launch_nf(std::integral_constant<int, 4>{});
+  else launch_nf(std::integral_constant<int, 8>{});
+  return true;
+}
+
 std::vector<at::Tensor> pooled_attn_bwd(const at::Tensor& q,
                                         const at::Tensor& k,
                                         const at::Tensor& v,
@@ -515,6 +1036,10 @@ std::vector<at::Tensor> pooled_attn_bwd(const at::Tensor& q,
   const float inv_keep = drop ? (float)(1.0 / (1.0 - p)) : 1.0f;
   const size_t lds = sizeof(float) * 2 * E * Lk;
   auto stream = at::hip::getCurrentHIPStream();
+  if (pooled_attn_bwd_mfma(q, k, v, out, dout, stats, mask, drop, dq, dk,
+                           dv, N * H, E, Lq, Lk, scale, inv_keep)) {
+    return {dq, dk, dv};
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::ScalarType::BFloat16, at::ScalarType::Half, q.scalar_type(),
       "pooled_attn_bwd", [&] {
