@@ -804,7 +804,8 @@ class _BNActPw(torch.autograd.Function):
          shift) = ctx.saved_tensors
         dy = dy.contiguous()
         dz = ext().pw_conv_dx(dy, weight)
-        dw = ext().pw_dw_pre(dy, x, scale, shift, ctx.act).to(weight.dtype)
+        dw = ext().pw_dw_pre(dy, x, scale, shift, ctx.act,
+                             weight.dtype)
         db = ext().channel_sum(dy).to(weight.dtype) if ctx.has_bias else None
         if ctx.act_only:
             # identity-BN: dx = act_grad(x) * dz, no dgamma/dbeta passes

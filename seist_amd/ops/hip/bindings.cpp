@@ -70,7 +70,8 @@ at::Tensor bn_bwd_dx_eval(const at::Tensor& dy, const at::Tensor& x,
                           long act);
 at::Tensor pw_dw_pre(const at::Tensor& dy, const at::Tensor& x,
                      const c10::optional<at::Tensor>& scale,
-                     const c10::optional<at::Tensor>& shift, long act);
+                     const c10::optional<at::Tensor>& shift, long act,
+                     c10::optional<at::ScalarType> out_dtype);
 std::vector<at::Tensor> bn_act_fwd_from_sums(
     const at::Tensor& x, const at::Tensor& sums, double count,
     const at::Tensor& gamma, const at::Tensor& beta,

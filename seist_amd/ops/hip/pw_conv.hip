@@ -24,6 +24,10 @@ at::Tensor bn_sums_only(const at::Tensor& x);
 bool pw_mfma_gemm(const at::Tensor& x, const at::Tensor& w,
                   const c10::optional<at::Tensor>& bias, at::Tensor& y,
                   bool trans, at::Tensor* stats_out = nullptr);
+at::Tensor pw_dw_pre(const at::Tensor& dy, const at::Tensor& x,
+                     const c10::optional<at::Tensor>& scale,
+                     const c10::optional<at::Tensor>& shift, long act,
+                     c10::optional<at::ScalarType> out_dtype);
 
 namespace {
 
@@ -179,13 +183,18 @@ std::vector<at::Tensor> pw_conv_bwd(const at::Tensor& dy, const at::Tensor& x,
         });
   }
 
-  // dw[o][i] = sum_n dy_n (Co x L) @ x_n^T (L x Ci) — a plain batched GEMM:
-  // run it on the MFMA matrix cores via rocBLAS (guide rule: hand-write the
-  // fused hot ops, use the BLAS library for plain GEMMs), reduce the batch
-  // axis in fp32. The bespoke pw_dw_kernel above measured 79% of the whole
-  // training step (rocprofv3, profiles/); this path is >10x faster.
-  auto dw = sum_batch_to(at::bmm(dy, x.transpose(1, 2)),
-                         w.scalar_type());
+  // dw[o][i] = sum_n dy_n (Co x L) @ x_n^T (L x Ci). Square-ish shapes
+  // go to the in-tree split-K MFMA kernel (measured 20-25 us vs
+  // hipblaslt+sum_batch's 28 us at 32..96 channels); rectangular MLP
+  // shapes stay on the library GEMM, which wins there.
+  at::Tensor dw;
+  const int Cmax = std::max(Co, Ci), Cmin = std::min(Co, Ci);
+  if (dy.scalar_type() == at::kBFloat16 && Cmax <= 128 && Cmin >= 32) {
+    dw = pw_dw_pre(dy, x, c10::nullopt, c10::nullopt, 0,
+                   w.scalar_type());
+  } else {
+    dw = sum_batch_to(at::bmm(dy, x.transpose(1, 2)), w.scalar_type());
+  }
   at::Tensor db;
   if (has_bias) {
     db = channel_sum_to(dy, w.scalar_type());

@@ -960,11 +960,13 @@ void pw_dw_kernel(const sa_bf16* __restrict__ dy,
 }  // namespace
 
 at::Tensor sum_batch(const at::Tensor& in);
+at::Tensor sum_batch_to(const at::Tensor& in, at::ScalarType out_dtype);
 
 // dw = sum_n dy_n @ act(x_n * scale + shift)^T  (fp32 out)
 at::Tensor pw_dw_pre(const at::Tensor& dy, const at::Tensor& x,
                      const c10::optional<at::Tensor>& scale,
-                     const c10::optional<at::Tensor>& shift, long act) {
+                     const c10::optional<at::Tensor>& shift, long act,
+                     c10::optional<at::ScalarType> out_dtype) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
   const int N = dy.size(0), Co = dy.size(1);
   const long L = dy.size(2);
@@ -987,5 +989,5 @@ at::Tensor pw_dw_pre(const at::Tensor& dy, const at::Tensor& x,
                        (const sa_bf16*)x.data_ptr(), nullptr, nullptr, 0,
                        slab.data_ptr<float>(), N, Co, Ci, L);
   }
-  return sum_batch(slab);
+  return sum_batch_to(slab, out_dtype.value_or(at::kFloat));
 }
