@@ -584,8 +584,11 @@ def test_pointwise_conv_cat_fused(widths):
 
     assert torch.allclose(y1.float(), y2.float(), atol=1e-2)
     for a, bb in zip(g1, g2):
-        assert torch.allclose(a.float(), bb.float(), atol=5e-2), \
-            (a - bb).abs().max().item()
+        # dw paths differ (split-K MFMA vs bmm+sum): compare at bf16
+        # rounding scale relative to the gradient magnitude
+        scale = bb.float().abs().max().item() or 1.0
+        d = (a.float() - bb.float()).abs().max().item() / scale
+        assert d < 1e-2, d
 
 
 def test_bn_act_pw_fused_matches_composite():
