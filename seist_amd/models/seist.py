@@ -265,8 +265,9 @@ class MultiScaleMixedConv(nn.Module):
         for proj, norm, conv in zip(self.projs, self.norms, self.convs):
             xi = run_conv_bn(proj, norm, x)
             outs.append(xi + conv(xi))
-        x = torch.cat(outs, dim=1)
-        return _norm(self.out_norm, x)
+        if isinstance(self.out_norm, nn.BatchNorm1d):
+            return ops.bn_act_cat(outs, self.out_norm)
+        return _norm(self.out_norm, torch.cat(outs, dim=1))
 
 
 class AttentionBlock(nn.Module):
@@ -384,8 +385,12 @@ class MultiPathTransformerLayer(nn.Module):
                                   self.gconv_droppath.drop_prob,
                                   self.training)
             outs.append(x2)
-        x = torch.cat(outs, dim=1)
-        x = _norm(self.norm2, x)
+        if len(outs) > 1:
+            x = ops.bn_act_cat(outs, self.norm2) \
+                if isinstance(self.norm2, nn.BatchNorm1d) \
+                else _norm(self.norm2, torch.cat(outs, dim=1))
+        else:
+            x = _norm(self.norm2, outs[0])
         return ops.droppath_add(x, self.mlp(x),
                                 self.mlp_droppath.drop_prob, self.training)
 

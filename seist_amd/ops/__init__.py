@@ -58,6 +58,7 @@ from .functional import (  # noqa: E402,F401
     avgmax_pool1d,
     droppath_add,
     bn_act,
+    bn_act_cat,
     bn_act_pw,
     act_pw,
     conv1d,

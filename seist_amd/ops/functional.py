@@ -888,3 +888,53 @@ def act_pw(x, act, weight, bias, module=None):
     ones, zeros = cache
     return _BNActPw.apply(x, ones, zeros, zeros, ones,
                           weight.contiguous(), bias, act_id, False, True)
+
+
+# ---------------------------------------------------------------------------
+# BN(+act) over a virtual channel-concat (MSMC/MPT concat->norm sites)
+# ---------------------------------------------------------------------------
+
+
+class _BNActCat(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gamma, beta, running_mean, running_var, training,
+                momentum, eps, act, *xs):
+        y, mean, invstd = ext().bn_act_cat_fwd(
+            list(xs), gamma, beta, running_mean, running_var, training,
+            momentum, eps, act)
+        ctx.save_for_backward(gamma, beta, mean, invstd, *xs)
+        ctx.training = training
+        ctx.act = act
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        gamma, beta, mean, invstd, *xs = ctx.saved_tensors
+        outs = ext().bn_act_cat_bwd(dy.contiguous(), list(xs), gamma, beta,
+                                    mean, invstd, ctx.training, ctx.act)
+        *dxs, dgamma, dbeta = outs
+        return (dgamma, dbeta, None, None, None, None, None, None) \
+            + tuple(dxs)
+
+
+def bn_act_cat(xs, bn, act: str = "none"):
+    """BatchNorm(+act) over the channel-concat of ``xs`` with the concat
+    virtual: the kernels route channel reads by range, and backward writes
+    per-input contiguous gradients (no cat, no narrow+copy)."""
+    act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
+    if (use_native(xs[0]) and 2 <= len(xs) <= 3
+            and not getattr(bn, "_sync_bn", False)):
+        y = _BNActCat.apply(bn.weight, bn.bias, bn.running_mean,
+                            bn.running_var, bn.training, bn.momentum,
+                            bn.eps, act_id, *[x.contiguous() for x in xs])
+        if bn.training and bn.track_running_stats \
+                and not getattr(bn, "_managed_nbt", False):
+            bn.num_batches_tracked += 1
+        return y
+    y = bn_act(torch.cat(xs, dim=1), bn.weight, bn.bias, bn.running_mean,
+               bn.running_var, bn.training, bn.momentum, bn.eps, act=act,
+               sync=getattr(bn, "_sync_bn", False))
+    if bn.training and bn.track_running_stats \
+            and not getattr(bn, "_managed_nbt", False):
+        bn.num_batches_tracked += 1
+    return y

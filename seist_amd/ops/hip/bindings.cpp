@@ -54,6 +54,15 @@ std::vector<at::Tensor> bn_act_fwd_with_part(
     const c10::optional<at::Tensor>& running_var, double momentum, double eps,
     long act);
 at::Tensor bn_part_to_sums(const at::Tensor& part);
+std::vector<at::Tensor> bn_act_cat_fwd(
+    std::vector<at::Tensor> xs, const at::Tensor& gamma,
+    const at::Tensor& beta, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, bool training,
+    double momentum, double eps, long act);
+std::vector<at::Tensor> bn_act_cat_bwd(
+    const at::Tensor& dy, std::vector<at::Tensor> xs,
+    const at::Tensor& gamma, const at::Tensor& beta, const at::Tensor& mean,
+    const at::Tensor& invstd, bool training, long act);
 std::vector<at::Tensor> bn_finalize_only(
     const at::Tensor& sums, double count,
     const c10::optional<at::Tensor>& running_mean,
@@ -172,6 +181,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "conv1d forward + BN stats partials (fusion step 1)");
   m.def("bn_act_fwd_with_part", &bn_act_fwd_with_part,
         "BN+act forward from producer-collected partials");
+  m.def("bn_act_cat_fwd", &bn_act_cat_fwd,
+        "BN+act over a virtual channel-concat (fwd)");
+  m.def("bn_act_cat_bwd", &bn_act_cat_bwd,
+        "BN+act concat backward (per-input contiguous dx)");
   m.def("bn_part_to_sums", &bn_part_to_sums,
         "(C,nsplit,2) partial slab -> (C,2) sums");
   m.def("bn_finalize_only", &bn_finalize_only,

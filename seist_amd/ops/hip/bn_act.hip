@@ -17,10 +17,36 @@ namespace {
 
 constexpr int kBlock = 256;
 
-template <typename scalar_t>
+// Channel-routed row lookup for the virtual channel-concat forms: the BN
+// runs over cat(x0, x1[, x2]) without the cat existing (MSMC/MPT sites,
+// reference models/seist.py:308-318, 486-504). cb1 == C degenerates to
+// the single-tensor form.
+template <typename T>
+__device__ __forceinline__ const T* bn_row(const T* x0, const T* x1,
+                                           const T* x2, int cb1, int cb2,
+                                           int C, long n, int c, long L) {
+  if (c < cb1) return x0 + (n * cb1 + c) * L;
+  if (c < cb2) return x1 + (n * (long)(cb2 - cb1) + (c - cb1)) * L;
+  return x2 + (n * (long)(C - cb2) + (c - cb2)) * L;
+}
+
+template <typename T>
+__device__ __forceinline__ T* bn_row_mut(T* x0, T* x1, T* x2, int cb1,
+                                         int cb2, int C, long n, int c,
+                                         long L) {
+  if (c < cb1) return x0 + (n * cb1 + c) * L;
+  if (c < cb2) return x1 + (n * (long)(cb2 - cb1) + (c - cb1)) * L;
+  return x2 + (n * (long)(C - cb2) + (c - cb2)) * L;
+}
+
+
+template <typename scalar_t, bool MULTI = false>
 __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
                                float* __restrict__ part,  // (C, nsplit, 2)
-                               int C, long NL, long L) {
+                               int C, long NL, long L,
+                               const scalar_t* x1 = nullptr,
+                               const scalar_t* x2 = nullptr,
+                               int cb1 = 0, int cb2 = 0) {
   __shared__ float red[kBlock / sa::kWave];
   const int c = blockIdx.x;
   const int split = blockIdx.y;
@@ -32,7 +58,8 @@ __global__ void bn_sums_kernel(const scalar_t* __restrict__ x,
 
   float s = 0.0f, s2 = 0.0f;
   for (long n = n0; n < n1; ++n) {
-    const scalar_t* xr = x + (n * C + c) * L;
+    const scalar_t* xr = MULTI ? bn_row(x, x1, x2, cb1, cb2, C, n, c, L)
+                               : x + (n * C + c) * L;
     for (long l = threadIdx.x; l < L; l += kBlock) {
       const float v = (float)xr[l];
       s += v;
@@ -90,27 +117,37 @@ __global__ void bn_finalize_kernel(const float* __restrict__ part,
 // dispatch-rate bound, not bandwidth bound
 constexpr int kEwTile = 4;
 
-template <typename scalar_t>
+template <typename scalar_t, bool MULTI = false>
 __global__ void bn_apply_kernel(const scalar_t* __restrict__ x,
                                 scalar_t* __restrict__ y,
                                 const float* __restrict__ mean,
                                 const float* __restrict__ invstd,
                                 const float* __restrict__ gamma,
                                 const float* __restrict__ beta,
-                                int C, long L, long total, int act) {
+                                int C, long L, long total, int act,
+                                const scalar_t* x1 = nullptr,
+                                const scalar_t* x2 = nullptr,
+                                int cb1 = 0, int cb2 = 0) {
 #pragma unroll
   for (int t = 0; t < kEwTile; ++t) {
     const long i =
         ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
     if (i >= total) return;
-    const int c = (int)((i / L) % C);
-    const float xh = ((float)x[i] - mean[c]) * invstd[c];
+    const long row = i / L;
+    const int c = (int)(row % C);
+    float xv;
+    if (MULTI) {
+      xv = (float)bn_row(x, x1, x2, cb1, cb2, C, row / C, c, L)[i - row * L];
+    } else {
+      xv = (float)x[i];
+    }
+    const float xh = (xv - mean[c]) * invstd[c];
     const float pre = xh * gamma[c] + beta[c];
     y[i] = (scalar_t)sa::act_fwd(pre, act);
   }
 }
 
-template <typename scalar_t>
+template <typename scalar_t, bool MULTI = false>
 __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
                                    const scalar_t* __restrict__ x,
                                    const float* __restrict__ mean,
@@ -118,7 +155,10 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
                                    float* __restrict__ part,  // (C,nsplit,2)
-                                   int C, long NL, long L, int act) {
+                                   int C, long NL, long L, int act,
+                                   const scalar_t* x1 = nullptr,
+                                   const scalar_t* x2 = nullptr,
+                                   int cb1 = 0, int cb2 = 0) {
   __shared__ float red[kBlock / sa::kWave];
   const int c = blockIdx.x;
   const int split = blockIdx.y;
@@ -131,7 +171,8 @@ __global__ void bn_bwd_sums_kernel(const scalar_t* __restrict__ dy,
   const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
   float s1 = 0.0f, s2 = 0.0f;
   for (long n = n0; n < n1; ++n) {
-    const scalar_t* xr = x + (n * C + c) * L;
+    const scalar_t* xr = MULTI ? bn_row(x, x1, x2, cb1, cb2, C, n, c, L)
+                               : x + (n * C + c) * L;
     const scalar_t* dyr = dy + (n * C + c) * L;
     for (long l = threadIdx.x; l < L; l += kBlock) {
       const float xh = ((float)xr[l] - m) * is;
@@ -171,7 +212,7 @@ __global__ void bn_part_reduce_kernel(const float* __restrict__ part,
   out[c * 2 + 1] = s2;
 }
 
-template <typename scalar_t, bool TRAINING>
+template <typename scalar_t, bool TRAINING, bool MULTI = false>
 __global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
                                  const scalar_t* __restrict__ x,
                                  scalar_t* __restrict__ dx,
@@ -180,15 +221,25 @@ __global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
                                  const float* __restrict__ gamma,
                                  const float* __restrict__ beta,
                                  const float* __restrict__ sums,  // dbeta,dgamma
-                                 int C, long L, long total, long NL, int act) {
+                                 int C, long L, long total, long NL, int act,
+                                 const scalar_t* x1 = nullptr,
+                                 const scalar_t* x2 = nullptr,
+                                 scalar_t* dx1 = nullptr,
+                                 scalar_t* dx2 = nullptr,
+                                 int cb1 = 0, int cb2 = 0) {
 #pragma unroll
   for (int t = 0; t < kEwTile; ++t) {
     const long i =
         ((long)blockIdx.x * kEwTile + t) * kBlock + threadIdx.x;
     if (i >= total) return;
-    const int c = (int)((i / L) % C);
+    const long row = i / L;
+    const int c = (int)(row % C);
+    const long l = i - row * L;
     const float m = mean[c], is = invstd[c], g = gamma[c], b = beta[c];
-    const float xh = ((float)x[i] - m) * is;
+    const float xv = MULTI
+        ? (float)bn_row(x, x1, x2, cb1, cb2, C, row / C, c, L)[l]
+        : (float)x[i];
+    const float xh = (xv - m) * is;
     float d = (float)dy[i];
     if (act != sa::ACT_NONE) d *= sa::act_grad(xh * g + b, act);
     float v;
@@ -199,7 +250,11 @@ __global__ void bn_bwd_dx_kernel(const scalar_t* __restrict__ dy,
     } else {
       v = d * g * is;
     }
-    dx[i] = (scalar_t)v;
+    if (MULTI) {
+      bn_row_mut(dx, dx1, dx2, cb1, cb2, C, row / C, c, L)[l] = (scalar_t)v;
+    } else {
+      dx[i] = (scalar_t)v;
+    }
   }
 }
 
@@ -575,4 +630,154 @@ std::vector<at::Tensor> bn_act_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dbeta = split[0].to(gamma.scalar_type());
   auto dgamma = split[1].to(gamma.scalar_type());
   return {dx, dgamma, dbeta};
+}
+
+// ---------------------------------------------------------------------------
+// Virtual channel-concat forms: BN(+act) over cat(xs) without the cat
+// (MSMC/MPT concat->norm sites). Backward writes per-input contiguous
+// gradients, so the cat's backward narrow+copy pass disappears too.
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> bn_act_cat_fwd(
+    std::vector<at::Tensor> xs, const at::Tensor& gamma,
+    const at::Tensor& beta, const c10::optional<at::Tensor>& running_mean,
+    const c10::optional<at::Tensor>& running_var, bool training,
+    double momentum, double eps, long act) {
+  TORCH_CHECK(xs.size() >= 2 && xs.size() <= 3);
+  const int N = xs[0].size(0);
+  const long L = xs[0].size(2);
+  int C = 0;
+  for (auto& x : xs) {
+    TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.size(0) == N
+                && x.size(2) == L);
+    C += x.size(1);
+  }
+  const int cb1 = xs[0].size(1);
+  const int cb2 = cb1 + xs[1].size(1);
+  const long NL = (long)N * L;
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = xs[0].options().dtype(at::kFloat);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+
+  at::Tensor mean, invstd;
+  const bool has_running = running_mean.has_value() && running_mean->defined();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, xs[0].scalar_type(),
+      "bn_cat", [&] {
+        const scalar_t* x0 = xs[0].data_ptr<scalar_t>();
+        const scalar_t* x1 = xs[1].data_ptr<scalar_t>();
+        const scalar_t* x2 = xs.size() > 2 ? xs[2].data_ptr<scalar_t>() : x1;
+        if (training) {
+          const int nsplit = pick_nsplit(N, C);
+          auto part = at::empty({C, nsplit, 2}, opts);
+          mean = at::empty({C}, opts);
+          invstd = at::empty({C}, opts);
+          hipLaunchKernelGGL((bn_sums_kernel<scalar_t, true>),
+                             dim3(C, nsplit), dim3(kBlock), 0,
+                             stream.stream(), x0, part.data_ptr<float>(),
+                             C, NL, L, x1, x2, cb1,
+                             xs.size() > 2 ? cb2 : C);
+          hipLaunchKernelGGL(bn_finalize_kernel, dim3(sa::ceil_div(C, 4)),
+                             dim3(256), 0, stream.stream(),
+                             part.data_ptr<float>(), nsplit, 1,
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             has_running ? running_mean->data_ptr<float>()
+                                         : nullptr,
+                             has_running ? running_var->data_ptr<float>()
+                                         : nullptr,
+                             C, NL, (float)momentum, (float)eps);
+        } else {
+          TORCH_CHECK(has_running, "eval mode requires running stats");
+          mean = running_mean->to(at::kFloat).contiguous();
+          invstd = at::rsqrt(running_var->to(at::kFloat) + eps).contiguous();
+        }
+        auto y = at::empty({N, C, L}, xs[0].options());
+        hipLaunchKernelGGL((bn_apply_kernel<scalar_t, true>),
+                           dim3(sa::ceil_div(total, (long)kBlock * kEwTile)),
+                           dim3(kBlock), 0, stream.stream(), x0,
+                           y.data_ptr<scalar_t>(), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), g32.data_ptr<float>(),
+                           b32.data_ptr<float>(), C, L, total, (int)act,
+                           x1, x2, cb1, xs.size() > 2 ? cb2 : C);
+        mean = mean;  // keep in scope
+        invstd = invstd;
+        xs[0] = y;  // reuse slot 0 to smuggle y out of the dispatch lambda
+      });
+  return {xs[0], mean, invstd};
+}
+
+std::vector<at::Tensor> bn_act_cat_bwd(
+    const at::Tensor& dy, std::vector<at::Tensor> xs,
+    const at::Tensor& gamma, const at::Tensor& beta, const at::Tensor& mean,
+    const at::Tensor& invstd, bool training, long act) {
+  const int N = xs[0].size(0);
+  const long L = xs[0].size(2);
+  int C = 0;
+  for (auto& x : xs) C += x.size(1);
+  const int cb1 = xs[0].size(1);
+  const int cb2 = cb1 + xs[1].size(1);
+  const long NL = (long)N * L;
+  const long total = (long)N * C * L;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto opts = xs[0].options().dtype(at::kFloat);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+
+  const int nsplit = pick_nsplit(N, C);
+  auto part = at::empty({C, nsplit, 2}, opts);
+  auto sums = at::empty({C, 2}, opts);
+  std::vector<at::Tensor> dxs;
+  for (auto& x : xs) dxs.push_back(at::empty_like(x));
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, xs[0].scalar_type(),
+      "bn_cat_bwd", [&] {
+        const scalar_t* x0 = xs[0].data_ptr<scalar_t>();
+        const scalar_t* x1 = xs[1].data_ptr<scalar_t>();
+        const scalar_t* x2 = xs.size() > 2 ? xs[2].data_ptr<scalar_t>() : x1;
+        scalar_t* d0 = dxs[0].data_ptr<scalar_t>();
+        scalar_t* d1 = dxs[1].data_ptr<scalar_t>();
+        scalar_t* d2 = xs.size() > 2 ? dxs[2].data_ptr<scalar_t>() : d1;
+        const int b2 = xs.size() > 2 ? cb2 : C;
+        hipLaunchKernelGGL((bn_bwd_sums_kernel<scalar_t, true>),
+                           dim3(C, nsplit), dim3(kBlock), 0, stream.stream(),
+                           dy.data_ptr<scalar_t>(), x0,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           g32.data_ptr<float>(), b32.data_ptr<float>(),
+                           part.data_ptr<float>(), C, NL, L, (int)act,
+                           x1, x2, cb1, b2);
+        hipLaunchKernelGGL(bn_part_reduce_kernel, dim3(sa::ceil_div(C, 4)),
+                           dim3(256), 0, stream.stream(),
+                           part.data_ptr<float>(), sums.data_ptr<float>(),
+                           C, nsplit, 1);
+        if (training) {
+          hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, true, true>),
+                             dim3(sa::ceil_div(total,
+                                               (long)kBlock * kEwTile)),
+                             dim3(kBlock), 0, stream.stream(),
+                             dy.data_ptr<scalar_t>(), x0, d0,
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             g32.data_ptr<float>(), b32.data_ptr<float>(),
+                             sums.data_ptr<float>(), C, L, total, NL,
+                             (int)act, x1, x2, d1, d2, cb1, b2);
+        } else {
+          hipLaunchKernelGGL((bn_bwd_dx_kernel<scalar_t, false, true>),
+                             dim3(sa::ceil_div(total,
+                                               (long)kBlock * kEwTile)),
+                             dim3(kBlock), 0, stream.stream(),
+                             dy.data_ptr<scalar_t>(), x0, d0,
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             g32.data_ptr<float>(), b32.data_ptr<float>(),
+                             sums.data_ptr<float>(), C, L, total, NL,
+                             (int)act, x1, x2, d1, d2, cb1, b2);
+        }
+      });
+  auto split = sums.unbind(1);
+  auto dbeta = split[0].to(gamma.scalar_type());
+  auto dgamma = split[1].to(gamma.scalar_type());
+  std::vector<at::Tensor> outs = dxs;
+  outs.push_back(dgamma);
+  outs.push_back(dbeta);
+  return outs;
 }

@@ -649,3 +649,34 @@ def test_act_pw_fused_matches_composite():
     for a, bb in zip(g1, g2):
         scale = bb.float().abs().max().item() or 1.0
         assert (a.float() - bb.float()).abs().max().item() / scale < 5e-2
+
+
+@pytest.mark.parametrize("widths", [(24, 24), (16, 24, 8)])
+def test_bn_act_cat_matches_composite(widths):
+    """Virtual channel-concat BN == bn_act over torch.cat (fwd + grads +
+    running stats)."""
+    import torch.nn as nn
+    torch.manual_seed(9)
+    N, L = 5, 512
+    C = sum(widths)
+    xs = [torch.randn(N, c, L, device="cuda:0", dtype=torch.bfloat16,
+                      requires_grad=True) for c in widths]
+    bn = nn.BatchNorm1d(C).to("cuda:0").train()
+    y1 = ops.bn_act_cat(xs, bn, act="gelu")
+    rm1, rv1 = bn.running_mean.clone(), bn.running_var.clone()
+    dy = torch.randn_like(y1)
+    g1 = torch.autograd.grad(y1, xs + [bn.weight, bn.bias], dy)
+
+    bn2 = nn.BatchNorm1d(C).to("cuda:0").train()
+    xs2 = [x.detach().clone().requires_grad_(True) for x in xs]
+    y2 = ops.bn_act(torch.cat(xs2, dim=1), bn2.weight, bn2.bias,
+                    bn2.running_mean, bn2.running_var, True, bn2.momentum,
+                    bn2.eps, act="gelu")
+    g2 = torch.autograd.grad(y2, xs2 + [bn2.weight, bn2.bias], dy)
+
+    assert torch.allclose(y1.float(), y2.float(), atol=1e-2)
+    assert torch.allclose(rm1, bn2.running_mean, atol=1e-4)
+    assert torch.allclose(rv1, bn2.running_var, atol=1e-4)
+    for a, bb in zip(g1, g2):
+        scale = bb.float().abs().max().item() or 1.0
+        assert (a.float() - bb.float()).abs().max().item() / scale < 1e-2
