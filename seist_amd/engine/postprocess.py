@@ -204,7 +204,7 @@ def _detect_event(outputs: torch.Tensor, prob_threshold: float,
     # size the run table by the batch's actual maximum run count (det
     # traces have a handful of runs; the worst-case L/2 table made the
     # sort dominate the step) — one small D2H sync, off the graphed path
-    M = int(rid_s[:, -1].max().item()) + 1 if N > 0 else 1
+    M = int(S.sum(dim=1).max().item()) + 1 if N > 0 else 1
     M = max(M, 2)
     cols = torch.arange(L, device=dev).expand(N, L)
     start_tab = torch.full((N, M), -1, dtype=torch.long, device=dev)
