@@ -108,12 +108,25 @@ class FlatReplica:
     def pack(self):
         """Lazy mode: one _foreach_copy_ of all stolen grads into the flat
         buffers, then re-alias p.grad to the flat views so the fused
-        optimizer's packed pointers stay stable across steps."""
+        optimizer's packed pointers stay stable across steps. Params that
+        received no gradient this step zero their slice (stale values
+        from the previous step must not survive the all-reduce)."""
         if not self.lazy:
             return
         for dtype, ps in self._bucket_params.items():
             views = self._views[dtype]
-            torch._foreach_copy_(views, [p.grad for p in ps])
+            dst, src, missing = [], [], []
+            for p, v in zip(ps, views):
+                if p.grad is None:
+                    missing.append(v)
+                elif p.grad.data_ptr() != v.data_ptr():
+                    dst.append(v)
+                    src.append(p.grad)
+                # else: already packed (double pack without a new backward)
+            if dst:
+                torch._foreach_copy_(dst, src)
+            if missing:
+                torch._foreach_zero_(missing)
             for p, v in zip(ps, views):
                 p.grad = v
 
