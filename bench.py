@@ -131,6 +131,12 @@ def main():
                 out = out.float()
             loss = loss_fn(out, t)
             loss.backward()
+            # pack INSIDE the captured region: the replayed backward
+            # writes the graph-pool grad tensors, and only a captured
+            # _foreach_copy_ moves them into the flat buffers every
+            # replay (an eager pack would see p.grad already re-aliased
+            # and skip — silently desynchronizing ranks)
+            replica.pack()
             return loss
 
         def exchange():

@@ -219,3 +219,47 @@ def test_flat_replica_lazy_grads_match(dev):
     for dt in r2.buffers:
         d = (r2.buffers[dt].float() - ref[dt].float()).abs().max()
         assert d.item() < 1e-5, f"replay2 mismatch {dt}: {d.item()}"
+
+
+def test_flat_replica_lazy_split_graph(dev):
+    """The bench's distributed pattern: compute (zero+backward+pack) is
+    captured, all-reduce stays eager. The captured pack must refresh the
+    flat buffers on EVERY replay."""
+    import copy
+    from seist_amd.parallel.ddp import FlatReplica
+    from seist_amd.models import create_model
+    from seist_amd.engine.precision import convert_to_bf16
+
+    torch.manual_seed(1)
+    kw = dict(in_samples=2048, path_drop_rate=0.0, attn_drop_rate=0.0,
+              key_drop_rate=0.0, mlp_drop_rate=0.0, other_drop_rate=0.0)
+    m = convert_to_bf16(create_model("seist_s_dpk", **kw)).to(dev).train()
+    rep = FlatReplica(m, lazy=True)
+    x = torch.randn(4, 3, 2048, device=dev, dtype=torch.bfloat16)
+
+    def compute():
+        rep.zero_grad()
+        y = m(x)
+        y.float().pow(2).mean().backward()
+        rep.pack()
+
+    for _ in range(2):
+        compute()
+    torch.cuda.synchronize()
+    ref = {dt: b.clone() for dt, b in rep.buffers.items()}
+
+    s = torch.cuda.Stream()
+    with torch.cuda.stream(s):
+        compute()
+    torch.cuda.current_stream().wait_stream(s)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        compute()
+    for _ in range(3):
+        for b in rep.buffers.values():
+            b.zero_()
+        g.replay()
+        torch.cuda.synchronize()
+        for dt in rep.buffers:
+            d = (rep.buffers[dt].float() - ref[dt].float()).abs().max()
+            assert d.item() < 1e-6, f"split-graph pack stale {dt}: {d.item()}"
