@@ -140,7 +140,7 @@ class MLP(nn.Module):
         self.lin1 = nn.Conv1d(ffwd, out_dim, kernel_size=1, bias=bias)
         self.dropout = nn.Dropout(mlp_drop_rate)
 
-    def forward(self, x):
+    def forward(self, x, apply_dropout: bool = True):
         x = ops.pointwise_conv(x, self.lin0.weight, self.lin0.bias)
         if _is_gelu(self.act):
             # GELU fused into lin1's LDS staging (never hits HBM)
@@ -149,7 +149,7 @@ class MLP(nn.Module):
         else:
             x = self.act(x)
             x = ops.pointwise_conv(x, self.lin1.weight, self.lin1.bias)
-        return self.dropout(x)
+        return self.dropout(x) if apply_dropout else x
 
 
 class DSConvNormAct(nn.Module):
@@ -227,9 +227,9 @@ class GroupConvBlock(nn.Module):
             y = self.mlp.act(y)
             y = ops.pointwise_conv(y, self.mlp.lin1.weight,
                                    self.mlp.lin1.bias)
-        y = self.mlp.dropout(y)
-        return ops.droppath_add(x, y, self.droppath1.drop_prob,
-                                self.training)
+        # mlp dropout fused into the residual pass
+        return ops.droppath_dropout_add(x, y, self.droppath1.drop_prob,
+                                        self.mlp.dropout.p, self.training)
 
 
 class MultiScaleMixedConv(nn.Module):
@@ -297,7 +297,7 @@ class AttentionBlock(nn.Module):
         self.out_proj = nn.Conv1d(io_dim, io_dim, kernel_size=1, bias=qkv_bias)
         self.proj_dropout = nn.Dropout(proj_drop_rate)
 
-    def forward(self, x):
+    def forward(self, x, apply_proj_dropout: bool = True):
         N, C, L = x.size()
         H = self.num_heads
         q = ops.pointwise_conv(x, self.q_proj.weight, self.q_proj.bias)
@@ -318,7 +318,7 @@ class AttentionBlock(nn.Module):
                                    training=self.training)
         out = out.reshape(N, C, L)
         out = ops.pointwise_conv(out, self.out_proj.weight, self.out_proj.bias)
-        return self.proj_dropout(out)
+        return self.proj_dropout(out) if apply_proj_dropout else out
 
 
 class MultiPathTransformerLayer(nn.Module):
@@ -375,9 +375,10 @@ class MultiPathTransformerLayer(nn.Module):
         outs = []
         if self.has_attn:
             x1 = run_conv_bn(self.attn_proj, self.norm0, x)
-            x1 = ops.droppath_add(x1, self.attention(x1),
-                                  self.attn_droppath.drop_prob,
-                                  self.training)
+            x1 = ops.droppath_dropout_add(
+                x1, self.attention(x1, apply_proj_dropout=False),
+                self.attn_droppath.drop_prob,
+                self.attention.proj_dropout.p, self.training)
             outs.append(x1)
         if self.has_conv:
             x2 = run_conv_bn(self.conv_proj, self.norm1, x)
@@ -391,8 +392,9 @@ class MultiPathTransformerLayer(nn.Module):
                 else _norm(self.norm2, torch.cat(outs, dim=1))
         else:
             x = _norm(self.norm2, outs[0])
-        return ops.droppath_add(x, self.mlp(x),
-                                self.mlp_droppath.drop_prob, self.training)
+        return ops.droppath_dropout_add(
+            x, self.mlp(x, apply_dropout=False),
+            self.mlp_droppath.drop_prob, self.mlp.dropout.p, self.training)
 
 
 class HeadDetectionPicking(nn.Module):

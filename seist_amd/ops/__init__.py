@@ -57,6 +57,7 @@ from .functional import (  # noqa: E402,F401
     pointwise_conv_stats,
     avgmax_pool1d,
     droppath_add,
+    droppath_dropout_add,
     bn_act,
     bn_act_cat,
     bn_act_pw,

@@ -938,3 +938,45 @@ def bn_act_cat(xs, bn, act: str = "none"):
             and not getattr(bn, "_managed_nbt", False):
         bn.num_batches_tracked += 1
     return y
+
+
+# ---------------------------------------------------------------------------
+# fused residual + DropPath + elementwise Dropout (K14)
+# ---------------------------------------------------------------------------
+
+_DPD_BASE = [0]
+
+
+class _DropPathDropoutAdd(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, y, path_p, drop_p, base):
+        z, slot = ext().droppath_dropout_add(x, y, path_p, drop_p, base)
+        ctx.save_for_backward(slot)
+        ctx.conf = (path_p, drop_p, base)
+        return z
+
+    @staticmethod
+    def backward(ctx, dz):
+        (slot,) = ctx.saved_tensors
+        path_p, drop_p, base = ctx.conf
+        dz = dz.contiguous()
+        dy = ext().droppath_dropout_scale(dz, slot, path_p, drop_p, base)
+        return dz, dy, None, None, None
+
+
+def droppath_dropout_add(x, y, path_p: float, drop_p: float,
+                         training: bool):
+    """z = x + DropPath(path_p)(Dropout(drop_p)(y)) in ONE pass: both
+    masks are regenerated from a seed the kernel snapshots per call, so
+    no mask tensors exist and hipGraph replays draw fresh masks."""
+    if not training or (path_p == 0.0 and drop_p == 0.0):
+        return droppath_add(x, y, 0.0, False)
+    if not use_native(x):
+        y = F.dropout(y, p=drop_p, training=True) if drop_p > 0.0 else y
+        return droppath_add(x, y, path_p, True)
+    # distinct RNG stream per call site; the captured value is frozen per
+    # graph node while the device seed varies per replay
+    _DPD_BASE[0] = (_DPD_BASE[0] + 1) & 0x3FFFFFFF
+    base = _DPD_BASE[0] * (1 << 28)
+    return _DropPathDropoutAdd.apply(x.contiguous(), y.contiguous(),
+                                     float(path_p), float(drop_p), base)

@@ -436,6 +436,13 @@ unsigned long long* attn_seed_state(const at::Tensor& like) {
   return (unsigned long long*)st[dev].data_ptr<long>();
 }
 
+// public bump for other dropout-consuming kernels (rowscale.hip)
+void bump_attn_seed(const at::Tensor& ref) {
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0, stream.stream(),
+                     attn_seed_state(ref));
+}
+
 }  // namespace
 
 

@@ -149,6 +149,13 @@ at::Tensor adam_pack(std::vector<at::Tensor> params,
                      std::vector<at::Tensor> masters, bool has_master);
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor channel_sum(const at::Tensor& in);
+std::vector<at::Tensor> droppath_dropout_add(const at::Tensor& x,
+                                             const at::Tensor& y,
+                                             double path_p, double drop_p,
+                                             long base);
+at::Tensor droppath_dropout_scale(const at::Tensor& dz,
+                                  const at::Tensor& slot, double path_p,
+                                  double drop_p, long base);
 at::Tensor row_scale_add(const at::Tensor& x, const at::Tensor& y,
                          const c10::optional<at::Tensor>& mask, double scale);
 at::Tensor row_scale(const at::Tensor& y,
@@ -229,6 +236,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sum_batch", &sum_batch, "batch-axis sum to fp32");
   m.def("channel_sum", &channel_sum, "per-channel sum to fp32");
   m.def("row_scale_add", &row_scale_add, "z = x + mask[n]*scale*y");
+  m.def("droppath_dropout_add", &droppath_dropout_add,
+        "fused residual + DropPath + Dropout (replay-safe seed slot)");
+  m.def("droppath_dropout_scale", &droppath_dropout_scale,
+        "backward mask-scale for droppath_dropout_add");
   m.def("row_scale", &row_scale, "z = mask[n]*scale*y");
   m.def("adam_step_packed", &adam_step_packed, "fused adam step");
 }

@@ -680,3 +680,33 @@ def test_bn_act_cat_matches_composite(widths):
     for a, bb in zip(g1, g2):
         scale = bb.float().abs().max().item() or 1.0
         assert (a.float() - bb.float()).abs().max().item() / scale < 1e-2
+
+
+def test_droppath_dropout_add_statistics_and_backward():
+    """Fused residual+DropPath+Dropout: mask statistics within binomial
+    bounds; backward mask EXACTLY matches forward (replay-safe slot)."""
+    torch.manual_seed(11)
+    N, C, L = 64, 8, 1024
+    x = torch.zeros(N, C, L, device="cuda:0", dtype=torch.bfloat16)
+    y = torch.ones(N, C, L, device="cuda:0", dtype=torch.bfloat16,
+                   requires_grad=True)
+    pp, dp = 0.25, 0.2
+    z = ops.droppath_dropout_add(x, y, pp, dp, training=True)
+    zf = z.float()
+    # per-row: either all zero (path dropped) or mean ~= 1 (dropout scaled)
+    row = zf.reshape(N, -1)
+    alive = row.abs().sum(1) > 0
+    frac_alive = alive.float().mean().item()
+    assert abs(frac_alive - (1 - pp)) < 0.2, frac_alive
+    live_mean = row[alive].mean().item() * (1 - pp)  # undo path 1/keep
+    assert abs(live_mean - 1.0) < 0.05, live_mean
+    # backward must regenerate the same mask: d(z)/d(y) == z (y == 1, x == 0)
+    g = torch.autograd.grad(z, y, torch.ones_like(z))[0]
+    assert torch.equal(g.float(), zf)
+
+
+def test_droppath_dropout_add_eval_passthrough():
+    x = torch.randn(4, 8, 64, device="cuda:0", dtype=torch.bfloat16)
+    y = torch.randn_like(x)
+    z = ops.droppath_dropout_add(x, y, 0.3, 0.2, training=False)
+    assert torch.allclose(z.float(), (x + y).float(), atol=1e-2)
