@@ -424,6 +424,8 @@ void pooled_attn_bwd_kv_kernel(const scalar_t* __restrict__ q,
   }
 }
 
+}  // namespace
+
 unsigned long long* attn_seed_state(const at::Tensor& like) {
   static std::array<at::Tensor, 16> st;
   const int dev = like.get_device();
@@ -442,6 +444,8 @@ void bump_attn_seed(const at::Tensor& ref) {
   hipLaunchKernelGGL(bump_seed_kernel, dim3(1), dim3(1), 0, stream.stream(),
                      attn_seed_state(ref));
 }
+
+namespace {
 
 }  // namespace
 
