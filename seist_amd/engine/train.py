@@ -360,6 +360,11 @@ def train_worker(args, device) -> str:
             # lazy: backward steals grads (no per-param accumulate adds);
             # allreduce() packs them with one _foreach_copy_
             replica = FlatReplica(model, lazy=torch.cuda.is_available())
+    elif torch.cuda.is_available():
+        # single GPU: the lazy replica gives the fused optimizer stable
+        # flat-view grad pointers (plain set_to_none steals fresh tensors
+        # every step, forcing a repack of the Adam chunk metadata per step)
+        replica = FlatReplica(model, lazy=True)
 
     ckpt_path = None
     num_saved = 0
