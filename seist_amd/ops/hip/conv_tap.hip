@@ -236,6 +236,11 @@ bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
   const int Cg_out = Cout / G;
   const int Cg_in = Cin / G;
   if (K < 1 || K > 24) return false;
+  static const bool no_tap = getenv("SEIST_AMD_NO_TAP") != nullptr;
+  if (no_tap && G == 1) return false;  // A/B: dense goes to conv_mfma/VALU
+  // measured: at reduce-width < 8 channels the im2col-in-LDS kernel wins
+  // (e.g. Ci=3,K=11,L=8192: 121 vs 211 us) — let it take those
+  if (G == 1 && Cin < 8) return false;
   if (G > 1) {
     // grouped tiles need group-aligned 16-channel windows
     if (Cg_out != Cg_in || Cin != Cout) return false;
