@@ -187,3 +187,39 @@ def test_upsample2x_matches_interpolate():
     x.grad = None
     y_ref.backward(g)
     assert torch.allclose(gx, x.grad)
+
+
+def test_droppath_dropout_add_cpu_fallback():
+    """CPU path: dropout + droppath composite semantics; eval identity."""
+    import torch
+    from seist_amd import ops
+    torch.manual_seed(0)
+    x = torch.zeros(32, 4, 256)
+    y = torch.ones(32, 4, 256)
+    z = ops.droppath_dropout_add(x, y, 0.0, 0.0, training=True)
+    assert torch.allclose(z, x + y)
+    z = ops.droppath_dropout_add(x, y, 0.5, 0.5, training=False)
+    assert torch.allclose(z, x + y)
+    z = ops.droppath_dropout_add(x, y, 0.25, 0.2, training=True)
+    row = z.reshape(32, -1)
+    alive = row.abs().sum(1) > 0
+    assert 0.4 < alive.float().mean().item() < 1.0
+    live_mean = row[alive].mean().item() * 0.75
+    assert abs(live_mean - 1.0) < 0.15
+
+
+def test_bn_act_cat_cpu_fallback():
+    import torch
+    import torch.nn as nn
+    from seist_amd import ops
+    torch.manual_seed(1)
+    xs = [torch.randn(4, 6, 64, requires_grad=True),
+          torch.randn(4, 10, 64, requires_grad=True)]
+    bn = nn.BatchNorm1d(16).train()
+    y = ops.bn_act_cat(xs, bn, act="relu")
+    y.sum().backward()
+    bn2 = nn.BatchNorm1d(16).train()
+    xs2 = [x.detach().clone().requires_grad_(True) for x in xs]
+    ref = torch.relu(bn2(torch.cat(xs2, dim=1)))
+    assert torch.allclose(y, ref, atol=1e-5)
+    assert all(x.grad is not None for x in xs)
