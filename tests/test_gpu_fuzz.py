@@ -129,3 +129,80 @@ def test_bn_pool_interp_fuzz(dev, trial):
     y_ref.backward(g)
     _cmp(xg.grad, xc.grad, tol[0] * 4, tol[1] * 2,
          f"interp bwd {out_len} " + msg)
+
+
+@pytest.mark.parametrize("seed", range(6))
+def test_fuzz_new_fused_ops(seed):
+    """Randomized shapes through the round-2 fused ops: cat-BN, cat-pw,
+    dropout-residual, pools."""
+    import torch.nn as nn
+    from seist_amd import ops
+
+    rng = torch.Generator().manual_seed(1000 + seed)
+
+    def ri(lo, hi):
+        return int(torch.randint(lo, hi + 1, (1,), generator=rng))
+
+    dev = "cuda:0"
+    N = ri(2, 6)
+    L = ri(33, 700)
+    dtype = torch.bfloat16 if seed % 2 == 0 else torch.float32
+
+    # --- cat-BN ---
+    npieces = ri(2, 3)
+    widths = [ri(4, 40) for _ in range(npieces)]
+    C = sum(widths)
+    xs = [torch.randn(N, c, L, device=dev, dtype=dtype, requires_grad=True)
+          for c in widths]
+    bn = nn.BatchNorm1d(C).to(dev).train()
+    act = ["none", "gelu", "relu"][seed % 3]
+    y = ops.bn_act_cat(xs, bn, act=act)
+    bn2 = nn.BatchNorm1d(C).to(dev).train()
+    xs2 = [x.detach().clone().requires_grad_(True) for x in xs]
+    ref = ops.bn_act(torch.cat(xs2, 1), bn2.weight, bn2.bias,
+                     bn2.running_mean, bn2.running_var, True, bn2.momentum,
+                     bn2.eps, act=act)
+    assert torch.allclose(y.float(), ref.float(), atol=5e-2), \
+        (y - ref).abs().max().item()
+    dy = torch.randn_like(y)
+    g1 = torch.autograd.grad(y, xs + [bn.weight, bn.bias], dy)
+    g2 = torch.autograd.grad(ref, xs2 + [bn2.weight, bn2.bias], dy)
+    for a, b in zip(g1, g2):
+        s = b.float().abs().max().item() or 1.0
+        assert (a.float() - b.float()).abs().max().item() / s < 5e-2
+
+    # --- cat-pw (bf16 only path; fallback otherwise — both must agree) ---
+    Co = ri(8, 64)
+    w = (torch.randn(Co, C, device=dev, dtype=dtype, generator=None) * 0.1
+         ).requires_grad_(True)
+    xs3 = [x.detach().clone().requires_grad_(True) for x in xs]
+    z1 = ops.pointwise_conv_cat(xs3, w, None)
+    xs4 = [x.detach().clone().requires_grad_(True) for x in xs]
+    w2 = w.detach().clone().requires_grad_(True)
+    z2 = ops.pointwise_conv(torch.cat(xs4, 1), w2, None)
+    s = z2.float().abs().max().item() or 1.0
+    assert (z1.float() - z2.float()).abs().max().item() / s < 2e-2
+    dz = torch.randn_like(z1)
+    h1 = torch.autograd.grad(z1, xs3 + [w], dz)
+    h2 = torch.autograd.grad(z2, xs4 + [w2], dz)
+    for a, b in zip(h1, h2):
+        s = b.float().abs().max().item() or 1.0
+        assert (a.float() - b.float()).abs().max().item() / s < 2e-2
+
+    # --- dropout residual: backward mask equals forward ---
+    x0 = torch.zeros(N, widths[0], L, device=dev, dtype=dtype)
+    y0 = torch.ones(N, widths[0], L, device=dev, dtype=dtype,
+                    requires_grad=True)
+    pp = [0.0, 0.3][seed % 2]
+    dp = [0.2, 0.0][seed % 2]
+    z = ops.droppath_dropout_add(x0, y0, pp, dp, training=True)
+    g = torch.autograd.grad(z, y0, torch.ones_like(z))[0]
+    assert torch.equal(g.float(), z.float())
+
+    # --- pools at odd lengths ---
+    k = ri(2, 5)
+    xp = torch.randn(N, widths[0], L, device=dev, dtype=torch.float32,
+                     requires_grad=True)
+    yp = ops.max_pool1d(xp, k, ceil_mode=True)
+    ref = torch.nn.functional.max_pool1d(xp.detach(), k, ceil_mode=True)
+    assert torch.allclose(yp, ref, atol=1e-6)
