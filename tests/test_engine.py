@@ -150,3 +150,20 @@ def test_demo_predict_end_to_end(tmp_path):
         capture_output=True, text=True, timeout=600, cwd=root)
     assert res.returncode == 0, res.stderr[-1500:]
     assert glob.glob(str(tmp_path / "*demo_prediction*.png"))
+
+
+def test_demo_predict_with_real_pretrained(tmp_path):
+    """demo_predict.py against one of the reference's shipped pretrained
+    checkpoints — the deployment recipe on real weights."""
+    import subprocess
+    import sys
+    ckpt = "/root/reference/pretrained/seist_m_dpk_diting.pth"
+    if not os.path.exists(ckpt):
+        pytest.skip("reference pretrained dir absent")
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    res = subprocess.run(
+        [sys.executable, "demo_predict.py", "--checkpoint", ckpt,
+         "--model-name", "seist_m_dpk", "--save-dir", str(tmp_path)],
+        capture_output=True, text=True, timeout=600, cwd=root)
+    assert res.returncode == 0, res.stderr[-1500:]
+    assert glob.glob(str(tmp_path / "*demo_prediction*.png"))
