@@ -5,9 +5,13 @@ FocalLoss, BinaryFocalLoss, MSELoss, CombinationLoss, MousaviLoss, plus the
 ``HuberLoss`` re-export). All operate on probabilities (post-sigmoid/softmax
 model outputs) like the reference; epsilon = 1e-6.
 
-These run as eager PyTorch on purpose: at batch 500 the loss is <1% of the
-step (rocprof, profiles/step_profile_r01.md), so the fusion budget is spent
-on the conv/BN/attention chains instead.
+BCE and CE (the flagship/phasenet losses) dispatch to the fused K15 kernel
+on GPU for fp32 scalar-weight inputs — ONE pass forward (grid-stride sum),
+one elementwise pass backward (`ops/hip/loss.hip`) — instead of the ~10
+eager elementwise+reduce kernels. The eager composites below remain the
+semantics ground truth and run on CPU / per-channel-weight / non-fp32
+inputs. The remaining losses stay eager on purpose: they are <1% of any
+step they appear in (rocprof, profiles/step_profile_r01.md).
 """
 
 from typing import Tuple
@@ -15,6 +19,8 @@ from typing import Tuple
 import torch
 import torch.nn as nn
 from torch.nn import HuberLoss  # noqa: F401  (re-export, parity with reference)
+
+from ..ops.functional import LOSS_BCE, LOSS_CE, fused_prob_loss
 
 _EPS = 1e-6
 
@@ -36,6 +42,9 @@ class CELoss(nn.Module):
         _register_weight(self, weight)
 
     def forward(self, preds, targets):
+        fused = fused_prob_loss(preds, targets, self.weight, LOSS_CE)
+        if fused is not None:
+            return fused
         loss = -targets * torch.log(preds + _EPS) * self.weight
         return loss.sum(1).mean()
 
@@ -48,6 +57,9 @@ class BCELoss(nn.Module):
         _register_weight(self, weight)
 
     def forward(self, preds, targets):
+        fused = fused_prob_loss(preds, targets, self.weight, LOSS_BCE)
+        if fused is not None:
+            return fused
         loss = -(
             targets * torch.log(preds + _EPS)
             + (1.0 - targets) * torch.log(1.0 - preds + _EPS)

@@ -72,6 +72,9 @@ from .functional import (  # noqa: E402,F401
     pointwise_conv_cat,
     pooled_attention,
     upsample2x,
+    fused_prob_loss,
+    LOSS_BCE,
+    LOSS_CE,
 )
 from .eqt import (  # noqa: E402,F401
     additive_attention_weights,

@@ -980,3 +980,61 @@ def droppath_dropout_add(x, y, path_p: float, drop_p: float,
     base = _DPD_BASE[0] * (1 << 28)
     return _DropPathDropoutAdd.apply(x.contiguous(), y.contiguous(),
                                      float(path_p), float(drop_p), base)
+
+
+# ---------------------------------------------------------------------------
+# fused probability-input losses (K15): BCE / CE forward + backward
+# ---------------------------------------------------------------------------
+
+LOSS_BCE = 0
+LOSS_CE = 1
+
+
+class _FusedProbLoss(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, p, t, w, kind, inv_div):
+        ctx.kind = kind
+        ctx.inv_div = inv_div
+        ctx.save_for_backward(p, t, w)
+        return ext().loss_sum_fwd(p, t, w, kind, inv_div)
+
+    @staticmethod
+    def backward(ctx, gout):
+        p, t, w = ctx.saved_tensors
+        dp = ext().loss_sum_bwd(p, t, w, gout.contiguous(), ctx.kind,
+                                ctx.inv_div)
+        return dp, None, None, None, None
+
+
+def fused_prob_loss(preds: torch.Tensor, targets: torch.Tensor,
+                    weight: torch.Tensor, kind: int):
+    """K15: single-pass BCE/CE over probability inputs (reference
+    models/loss.py:8-61 semantics, eps=1e-6, scalar weight). Returns the
+    0-dim fp32 loss, or ``None`` when the fused path does not apply (CPU,
+    non-fp32, per-channel weight, weight still on host) — callers fall
+    back to the eager composite, which stays the numerics ground truth.
+
+    * BCE: mean over ALL elements -> inv_div = 1/numel
+    * CE : ``loss.sum(1).mean()`` -> inv_div = size(1)/numel
+    ``weight`` and grad_output are read via device pointers so the loss is
+    hipGraph-capturable inside the benchmark's captured compute graph.
+    """
+    if not (preds.is_cuda
+            and preds.dtype == torch.float32
+            and targets.dtype == torch.float32
+            and targets.is_cuda
+            and preds.shape == targets.shape
+            and weight.dim() == 0
+            and weight.is_cuda
+            and not targets.requires_grad):
+        return None
+    if kind == LOSS_CE and preds.dim() < 2:
+        return None
+    if not use_native(preds):  # ALLOW_FALLBACK debug escape hatch
+        return None
+    p = preds.contiguous()
+    t = targets.contiguous()
+    div = p.numel() if kind == LOSS_BCE else p.numel() // p.size(1)
+    if div == 0:
+        return None
+    return _FusedProbLoss.apply(p, t, weight, kind, 1.0 / div)

@@ -104,6 +104,11 @@ at::Tensor max_pool1d_bwd(const at::Tensor& dy, const at::Tensor& argmax,
                           long k, long in_len);
 at::Tensor gap_fwd(const at::Tensor& x);
 at::Tensor gap_bwd(const at::Tensor& dy, long in_len);
+at::Tensor loss_sum_fwd(const at::Tensor& p, const at::Tensor& t,
+                        const at::Tensor& w, long kind, double inv_div);
+at::Tensor loss_sum_bwd(const at::Tensor& p, const at::Tensor& t,
+                        const at::Tensor& w, const at::Tensor& gout,
+                        long kind, double inv_div);
 at::Tensor avgmax_pool_bwd(const at::Tensor& dy, const at::Tensor& argmax,
                            long k, long in_len);
 at::Tensor interp_linear_fwd(const at::Tensor& x, long out_len);
@@ -214,6 +219,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("max_pool1d_bwd", &max_pool1d_bwd, "max pool backward");
   m.def("gap_fwd", &gap_fwd, "global average pool forward");
   m.def("gap_bwd", &gap_bwd, "global average pool backward");
+  m.def("loss_sum_fwd", &loss_sum_fwd, "fused BCE/CE loss forward");
+  m.def("loss_sum_bwd", &loss_sum_bwd, "fused BCE/CE loss backward");
   m.def("avgmax_pool_bwd", &avgmax_pool_bwd, "fused avg+max pool backward");
   m.def("interp_linear_fwd", &interp_linear_fwd, "linear interp forward");
   m.def("interp_linear_bwd", &interp_linear_bwd, "linear interp backward");

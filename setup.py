@@ -45,6 +45,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "rowscale.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
         os.path.join(HIP_DIR, "eqt.hip"),
+        os.path.join(HIP_DIR, "loss.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

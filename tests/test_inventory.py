@@ -111,7 +111,8 @@ def test_native_op_surface_complete():
                  "avgmax_pool1d", "max_pool1d", "global_avg_pool1d",
                  "interp_linear", "upsample2x", "pooled_attention",
                  "droppath_add", "layer_norm", "additive_attention_weights",
-                 "lstm", "gelu", "auto_pad", "FusedAdam"):
+                 "lstm", "gelu", "auto_pad", "FusedAdam",
+                 "fused_prob_loss"):
         assert hasattr(ops, name), name
 
 
