@@ -40,6 +40,10 @@ bool conv_tap_mfma(const at::Tensor& x, const at::Tensor& w,
                    const c10::optional<at::Tensor>& bias, at::Tensor& y,
                    long padl, long dilation, long groups, bool is_dx,
                    at::Tensor* stats_out = nullptr);
+bool conv_smallc_mfma(const at::Tensor& x, const at::Tensor& w,
+                      const c10::optional<at::Tensor>& bias, at::Tensor& y,
+                      long padl, long dilation, long groups, bool is_dx,
+                      at::Tensor* stats_out = nullptr);
 bool conv_tap_s_mfma(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias, at::Tensor& y,
                      long stride, long padl, long dilation, long groups,
@@ -400,6 +404,11 @@ at::Tensor conv1d_fwd(const at::Tensor& x, const at::Tensor& w,
   // tap-gather kernel stages each channel row once; the im2col kernel
   // remains as the fp32 fallback
   if (stride == 1
+      && conv_smallc_mfma(x, w, bias, y, padl, dilation, groups,
+                          /*is_dx=*/false)) {
+    return y;
+  }
+  if (stride == 1
       && conv_tap_mfma(x, w, bias, y, padl, dilation, groups,
                        /*is_dx=*/false)) {
     return y;
@@ -477,6 +486,11 @@ std::vector<at::Tensor> conv1d_fwd_stats(
   auto y = at::empty({N, Co, Lo}, x.options());
   at::Tensor part;
   if (stride == 1
+      && conv_smallc_mfma(x, w, bias, y, padl, dilation, groups,
+                          /*is_dx=*/false, &part)) {
+    return {y, part};
+  }
+  if (stride == 1
       && conv_tap_mfma(x, w, bias, y, padl, dilation, groups,
                        /*is_dx=*/false, &part)) {
     return {y, part};
@@ -508,6 +522,11 @@ void conv1d_dx_into(const at::Tensor& dy, const at::Tensor& w,
   const int Cog = Co / groups;
   auto stream = at::hip::getCurrentHIPStream();
 
+  if (stride == 1
+      && conv_smallc_mfma(dy, w, c10::nullopt, dx, padl, dilation, groups,
+                          /*is_dx=*/true)) {
+    return;
+  }
   if (stride == 1
       && conv_tap_mfma(dy, w, c10::nullopt, dx, padl, dilation, groups,
                        /*is_dx=*/true)) {

@@ -760,3 +760,30 @@ def test_fused_prob_loss_weight_and_fallback(dev):
     # vector weight: fused path must decline (broadcast semantics differ)
     w = torch.tensor([1.0, 2.0, 0.5], device=dev)
     assert ops.fused_prob_loss(p, t, w, ops.LOSS_BCE) is None
+
+
+@pytest.mark.parametrize("geo", [
+    (8, 3, 7, 512), (8, 16, 7, 512), (1, 8, 11, 512), (16, 8, 9, 512),
+    (2, 27, 3, 512), (8, 32, 5, 512), (16, 32, 7, 512), (8, 8, 2, 512),
+    (3, 5, 16, 512), (15, 31, 13, 510),
+])
+def test_conv_smallc_fwd_dx(dev, geo):
+    """Small-C dense conv kernel (conv_smallc.hip) vs eager fp32, incl. the
+    odd-Lout fallback case (last geo routes back to tap/im2col)."""
+    co, ci, k, L = geo
+    torch.manual_seed(co * 100 + ci)
+    x = torch.randn(9, ci, L, device=dev, dtype=torch.bfloat16)
+    w = (torch.randn(co, ci, k, device=dev, dtype=torch.bfloat16)
+         * (ci * k) ** -0.5)
+    b = torch.randn(co, device=dev, dtype=torch.bfloat16)
+    pl, pr = ops.auto_pad_lr(L, k, 1)
+    xr = x.clone().requires_grad_(True)
+    y = ops.conv1d(xr, w, b, stride=1, padding=(pl, pr))
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.float().requires_grad_(True)
+    yr = F.conv1d(F.pad(xf, (pl, pr)), w.float(), b.float())
+    yr.backward(dy.float())
+    _cmp(y, yr, atol=0.02, rtol=0.02, msg=f"fwd {geo}")
+    _cmp(xr.grad, xf.grad, atol=0.02, rtol=0.02, msg=f"dx {geo}")
