@@ -9,6 +9,9 @@ im2col/tap result computed in the same process via F.conv1d eager fp32.
 
 import argparse
 import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 import torch.nn.functional as F
