@@ -25,6 +25,7 @@
 at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor sum_mid(const at::Tensor& in);
 at::Tensor sum_mid_to(const at::Tensor& in, at::ScalarType out_dtype);
+at::Tensor sum_batch_to(const at::Tensor& in, at::ScalarType out_dtype);
 at::Tensor channel_sum_to(const at::Tensor& in, at::ScalarType out_dtype);
 bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
@@ -627,6 +628,31 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
     if (has_bias) {
       db = channel_sum_to(dy, w.scalar_type());
     }
+    return {dw, db};
+  }
+
+  // strided dense convs (phasenet encoder / transposed-conv grad): the
+  // direct accumulation kernel below is ~100x off roofline here (measured
+  // 1.25 ms/call at Co=8, profiles/step_profile_r02.md). One padded copy
+  // of x makes every tap an arithmetic-sequence view, so a zero-copy
+  // (N, Cig, K, Lo) as_strided with strides (Ci*Lp, Lp, dil, s)
+  // materialized once (= im2col) turns dw into ONE batched GEMM over the
+  // matrix cores plus one batch-axis sum.
+  if (groups == 1 && stride > 1
+      && (long)N * Cig * K * Lo <= (long)256 * 1024 * 1024) {
+    auto xp = (padl > 0 || padr > 0)
+        ? at::constant_pad_nd(x, {padl, padr})
+        : x;
+    const long Lp = xp.size(2);
+    auto xs = at::as_strided(xp, {(long)N, (long)Cig, (long)K, Lo},
+                             {(long)Ci * Lp, Lp, dilation, stride});
+    auto xc = xs.reshape({(long)N, (long)Cig * K, Lo});  // one im2col copy
+    auto slab = at::bmm(dy, xc.transpose(1, 2));         // (N, Co, Cig*K)
+    auto dw = sum_batch_to(slab.view({(long)N, (long)Co * Cig * K}),
+                           w.scalar_type())
+                  .view({(long)Co, (long)Cig, (long)K});
+    at::Tensor db;
+    if (has_bias) db = channel_sum_to(dy, w.scalar_type());
     return {dw, db};
   }
 
