@@ -26,6 +26,9 @@ at::Tensor sum_batch(const at::Tensor& in);
 at::Tensor sum_mid(const at::Tensor& in);
 at::Tensor sum_mid_to(const at::Tensor& in, at::ScalarType out_dtype);
 at::Tensor sum_batch_to(const at::Tensor& in, at::ScalarType out_dtype);
+at::Tensor conv_dw_smallc(const at::Tensor& dy, const at::Tensor& x,
+                          long K, long padl, long stride, long dilation,
+                          long groups, at::ScalarType out_dtype);
 at::Tensor channel_sum_to(const at::Tensor& in, at::ScalarType out_dtype);
 bool conv_mfma(const at::Tensor& x, const at::Tensor& w,
                const c10::optional<at::Tensor>& bias, at::Tensor& y,
@@ -596,6 +599,17 @@ std::vector<at::Tensor> conv1d_dw_db(const at::Tensor& dy,
   // so one batched GEMM covers every group.
   // grouped convs keep the direct kernel: at Cog=Cig=8 the per-tap
   // batched GEMM is launch/overhead-bound and measured slower.
+  // small-channel dense convs: one MFMA kernel + one batch sum replaces
+  // the K-tap bmm slab + sum_mid chain (measured ~4-9 ms/step of GEMM +
+  // reduce on eqt/ditingmotion; profiles/step_profile_r02.md)
+  if (auto dws = conv_dw_smallc(dy, x, K, padl, stride, dilation, groups,
+                                w.scalar_type());
+      dws.defined()) {
+    at::Tensor db;
+    if (has_bias) db = channel_sum_to(dy, w.scalar_type());
+    return {dws, db};
+  }
+
   if (groups == 1 && stride == 1) {
     // one bmm per tap into a (K, N, Co, Ci) slab, one middle-axis sum over
     // N for all taps, one strided scatter into the (Co, Ci, K) layout

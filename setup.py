@@ -43,6 +43,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "dw_mfma.hip"),
         os.path.join(HIP_DIR, "conv_tap.hip"),
         os.path.join(HIP_DIR, "conv_smallc.hip"),
+        os.path.join(HIP_DIR, "conv_dw_smallc.hip"),
         os.path.join(HIP_DIR, "rowscale.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
         os.path.join(HIP_DIR, "eqt.hip"),

@@ -830,3 +830,24 @@ def test_conv_transpose_dw_bmm_route(dev):
     yr = F.conv_transpose1d(x.float(), wf, None, stride=4)
     yr.backward(dy.float())
     _cmp(w.grad, wf.grad, atol=0.05, rtol=0.02, msg="transpose dw")
+
+
+@pytest.mark.parametrize("geo", [
+    (8, 16, 11, 8192), (1, 8, 11, 4096), (16, 16, 9, 4096),
+    (2, 43, 3, 512), (8, 40, 3, 2048), (64, 16, 3, 128),
+])
+def test_conv_dw_smallc_route(dev, geo):
+    """Small-C dense conv weight grad (conv_dw_smallc.hip) vs eager fp32."""
+    co, ci, k, L = geo
+    torch.manual_seed(ci + k)
+    x = torch.randn(21, ci, L, device=dev, dtype=torch.bfloat16)
+    w = (torch.randn(co, ci, k, device=dev, dtype=torch.bfloat16)
+         * (ci * k) ** -0.5).requires_grad_(True)
+    pl, pr = ops.auto_pad_lr(L, k, 1)
+    y = ops.conv1d(x, w, None, stride=1, padding=(pl, pr))
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    wf = w.detach().float().requires_grad_(True)
+    yr = F.conv1d(F.pad(x.float(), (pl, pr)), wf)
+    yr.backward(dy.float())
+    _cmp(w.grad, wf.grad, atol=0.05, rtol=0.02, msg=f"dw smallc {geo}")
