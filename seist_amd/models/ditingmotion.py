@@ -84,7 +84,8 @@ class SideLayer(nn.Module):
         N, C, L = x.size()
         if C * L != self.linear_in_dim:
             # official model expects (2, 128); interpolate to fit other shapes
-            x = F.interpolate(x, self.linear_in_dim // self.conv_out_channels)
+            x = ops.nearest_resize(x,
+                                   self.linear_in_dim // self.conv_out_channels)
         x1 = self.flatten(x)
         x2 = self.relu(self.lin0(x1))
         x3 = self.sigmoid(self.lin1(x2))

@@ -73,6 +73,7 @@ from .functional import (  # noqa: E402,F401
     pooled_attention,
     upsample2x,
     fused_prob_loss,
+    nearest_resize,
     LOSS_BCE,
     LOSS_CE,
 )

@@ -1038,3 +1038,20 @@ def fused_prob_loss(preds: torch.Tensor, targets: torch.Tensor,
     if div == 0:
         return None
     return _FusedProbLoss.apply(p, t, weight, kind, 1.0 / div)
+
+
+def nearest_resize(x: torch.Tensor, out_len: int) -> torch.Tensor:
+    """``F.interpolate(x, out_len)`` (nearest, the reference's default) with
+    fast paths: an integer-ratio downsample IS a strided slice
+    (src = floor(i * Lin/Lout) = i * step), which replaces the
+    launch-bound at::native nearest kernels (measured 6 calls x ~130 us
+    per ditingmotion step at (500, 2, 1024)->16); 2x upsample routes to
+    the native kernel."""
+    L = x.size(-1)
+    if out_len == L:
+        return x
+    if out_len < L and L % out_len == 0:
+        return x[..., :: L // out_len]
+    if out_len == 2 * L:
+        return upsample2x(x)
+    return F.interpolate(x, out_len)

@@ -223,3 +223,22 @@ def test_bn_act_cat_cpu_fallback():
     ref = torch.relu(bn2(torch.cat(xs2, dim=1)))
     assert torch.allclose(y, ref, atol=1e-5)
     assert all(x.grad is not None for x in xs)
+
+
+def test_nearest_resize_matches_interpolate():
+    import torch.nn.functional as F
+    from seist_amd import ops
+    torch.manual_seed(0)
+    for L, out in [(1024, 16), (512, 8), (256, 8), (100, 7), (64, 128),
+                   (128, 128)]:
+        x = torch.randn(3, 2, L, requires_grad=True)
+        y = ops.nearest_resize(x, out)
+        ref = F.interpolate(x, out) if out != L else x
+        assert torch.equal(y.reshape(-1), ref.reshape(-1)), (L, out)
+        if out != L:
+            g = torch.randn_like(ref)
+            (dx,) = torch.autograd.grad(y.sum() * 0 + (y * g).sum(), [x])
+            xr = x.detach().clone().requires_grad_(True)
+            (dxr,) = torch.autograd.grad(
+                (F.interpolate(xr, out) * g).sum(), [xr])
+            assert torch.allclose(dx, dxr), (L, out)
