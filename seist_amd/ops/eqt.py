@@ -206,8 +206,11 @@ def lstm(x: torch.Tensor, module: torch.nn.LSTM) -> torch.Tensor:
     # 2.3 ms/step on eqt for tiny K=16 / tall-skinny dW shapes). The
     # recurrence itself stays fp32 — only pre-activation output rounding
     # changes, matching the bf16 conv stack that produced x.
-    bf16_pre = (x.dtype == torch.bfloat16
-                and os.environ.get("SEIST_AMD_LSTM_PRE_FP32") != "1")
+    if (os.environ.get("SEIST_AMD_LSTM_PRE_FP32") == "1"
+            and x.dtype != torch.float32):
+        x = x.contiguous().float()  # A/B escape: old all-fp32 projection
+        y = y.float()
+    bf16_pre = x.dtype == torch.bfloat16
     training = torch.is_grad_enabled() and (
         x.requires_grad
         or any(p.requires_grad for p in module.parameters()))
