@@ -9,8 +9,6 @@ used by the numerics tests as ground truth.
 
 from typing import Optional, Tuple
 
-import os
-
 import torch
 import torch.nn.functional as F
 
@@ -198,19 +196,15 @@ def lstm(x: torch.Tensor, module: torch.nn.LSTM) -> torch.Tensor:
     if not use_native(x):
         out, _ = module(x.float() if x.dtype != torch.float32 else x)
         return out
+    # pre-projections stay fp32 even for bf16 activations: a bf16 GEMM
+    # pair here measured SLOWER same-box (eqt 20.49 vs 19.79 ms/step —
+    # hipblaslt's tall-skinny fp32 dW tunings beat the bf16 ones at
+    # K=16 shapes), so the upcast is the measured-faster path.
     x = x.contiguous()
+    if x.dtype != torch.float32:
+        x = x.float()
     N, L, _ = x.shape
     y = x.new_empty(N, L, dirs * H, dtype=torch.float32)
-    # bf16 activations keep the pre-projection GEMM in bf16 (fp32
-    # accumulate inside hipblaslt; the fp32 GEMM pair here measured
-    # 2.3 ms/step on eqt for tiny K=16 / tall-skinny dW shapes). The
-    # recurrence itself stays fp32 — only pre-activation output rounding
-    # changes, matching the bf16 conv stack that produced x.
-    if (os.environ.get("SEIST_AMD_LSTM_PRE_FP32") == "1"
-            and x.dtype != torch.float32):
-        x = x.contiguous().float()  # A/B escape: old all-fp32 projection
-        y = y.float()
-    bf16_pre = x.dtype == torch.bfloat16
     training = torch.is_grad_enabled() and (
         x.requires_grad
         or any(p.requires_grad for p in module.parameters()))
@@ -218,10 +212,7 @@ def lstm(x: torch.Tensor, module: torch.nn.LSTM) -> torch.Tensor:
         sfx = "_reverse" if dir_ == 1 else ""
         w_ih = getattr(module, f"weight_ih_l0{sfx}")
         w_hh = getattr(module, f"weight_hh_l0{sfx}")
-        if bf16_pre:
-            pre = x.matmul(w_ih.to(torch.bfloat16).t()).float()
-        else:
-            pre = x.matmul(w_ih.t())
+        pre = x.matmul(w_ih.t())
         if module.bias:
             b_ih = getattr(module, f"bias_ih_l0{sfx}")
             b_hh = getattr(module, f"bias_hh_l0{sfx}")
