@@ -246,7 +246,9 @@ bool conv_smallc_mfma(const at::Tensor& x, const at::Tensor& w,
   const int Cm = y.size(1);
   const long Lout = y.size(2);
   const int K = w.size(2);
-  if (K < 1 || K > 16) return false;
+  // K=1 is a pointwise GEMM: the KP=8 tap padding would waste 7/8 of the
+  // reduce — leave it to the tap/im2col paths
+  if (K < 2 || K > 16) return false;
   if (CiR > 32) return false;
   if (Lout & 1) return false;  // packed dword stores need even rows
   static const bool off = getenv("SEIST_AMD_NO_SMALLC") != nullptr;
