@@ -85,9 +85,8 @@ class BiLSTMBlock(nn.Module):
         # LSTM weights stay fp32; cast bf16 encoder activations up.
         # On GPU the recurrence runs as the persistent K11 kernel (input
         # projections = one GEMM), not MIOpen's per-step launches.
-        # keep bf16 into the pre-projection GEMM (ops.lstm runs the
-        # recurrence in fp32 either way; the fp32 GEMM pair here was
-        # 2.3 ms/step of tiny-K hipblaslt time)
+        # ops.lstm owns the fp32 upcast (bf16 pre-projection GEMMs
+        # measured slower same-box — see the note in ops/eqt.py)
         x = ops.lstm(x.permute(0, 2, 1).contiguous(), self.bilstm)
         x = self.dropout(x).permute(0, 2, 1)
         x = run_conv(self.conv, x)

@@ -1026,6 +1026,7 @@ def fused_prob_loss(preds: torch.Tensor, targets: torch.Tensor,
             and preds.shape == targets.shape
             and weight.dim() == 0
             and weight.is_cuda
+            and not weight.requires_grad
             and not targets.requires_grad):
         return None
     if kind == LOSS_CE and preds.dim() < 2:
