@@ -199,12 +199,42 @@ def _w_syncbn_model(rank):
         (y - y_ref[lo:hi]).abs().max().item()
 
 
+def _w_flat_replica_mixed(rank):
+    """eqtransformer's shape of replica: bf16 conv params + fp32 LSTM/LN
+    params -> TWO flat buckets, one all-reduce each (lazy mode, the
+    bench default). Grads of both dtypes must land averaged."""
+    from seist_amd.parallel.ddp import FlatReplica
+    torch.manual_seed(200 + rank)
+    net = torch.nn.Sequential(torch.nn.Linear(4, 8), torch.nn.Linear(8, 2))
+    net[0].to(torch.bfloat16)  # mixed precision module tree
+    rep = FlatReplica(net, lazy=True)
+    assert set(rep.buffers) == {torch.bfloat16, torch.float32}
+    torch.manual_seed(rank)
+    x = torch.randn(16, 4, dtype=torch.bfloat16)
+    rep.zero_grad()
+    loss = net[1](net[0](x).float()).sum()
+    loss.backward()
+    local = [p.grad.clone() for p in net.parameters()]
+    rep.allreduce()
+    for p, lg in zip(net.parameters(), local):
+        lst = [torch.zeros_like(lg) for _ in range(WORLD)]
+        dist.all_gather(lst, lg)
+        manual = ((lst[0].float() + lst[1].float()) / 2).to(lg.dtype)
+        tol = 2e-2 if lg.dtype == torch.bfloat16 else 1e-6
+        assert torch.allclose(p.grad.float(), manual.float(), atol=tol), \
+            p.grad.dtype
+
+
 def test_collectives():
     _spawn(_w_collectives, 29511)
 
 
 def test_flat_replica():
     _spawn(_w_flat_replica, 29512)
+
+
+def test_flat_replica_mixed_dtype():
+    _spawn(_w_flat_replica_mixed, 29517)
 
 
 def test_metrics_sync():
