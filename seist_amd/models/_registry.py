@@ -57,11 +57,40 @@ def save_checkpoint(
     use_ddp: bool = False,
     use_compile: bool = False,
 ) -> None:
+    """Checkpoints are always written with fp32 weights (the
+    reference .pth format): under the bf16 training policy the fused
+    optimizer's fp32 MASTER copies are saved in place of the bf16
+    parameters (full-precision trajectory, not a bf16 round-trip), and
+    any remaining non-fp32 floating tensors are upcast."""
     os.makedirs(os.path.dirname(save_path) or ".", exist_ok=True)
+    mdl = _unwrap(model)
+    sd = mdl.state_dict()
+    masters = {}
+    if optimizer is not None:
+        for prm, st in getattr(optimizer, "state", {}).items():
+            m = st.get("master") if isinstance(st, dict) else None
+            if isinstance(m, torch.Tensor):
+                masters[prm] = m
+    needs_fix = masters or any(
+        torch.is_tensor(v) and v.is_floating_point()
+        and v.dtype != torch.float32 for v in sd.values())
+    if needs_fix:
+        pmap = dict(mdl.named_parameters())
+        out = {}
+        for k, v in sd.items():
+            prm = pmap.get(k)
+            if prm is not None and prm in masters:
+                out[k] = masters[prm].detach().clone()
+            elif (torch.is_tensor(v) and v.is_floating_point()
+                  and v.dtype != torch.float32):
+                out[k] = v.float()
+            else:
+                out[k] = v
+        sd = out
     ckpt = {
         "epoch": epoch,
         "optimizer_dict": optimizer.state_dict() if optimizer is not None else None,
-        "model_dict": _unwrap(model).state_dict(),
+        "model_dict": sd,
         "loss": loss,
         "use_compile": use_compile,
         "use_ddp": use_ddp,
