@@ -150,7 +150,13 @@ def _pick_phase(outputs: torch.Tensor, prob_threshold: float,
 
     k = min(topk, L)
     heights = torch.where(cand, x, torch.full_like(x, float("-inf")))
-    vals, idx = heights.topk(k, dim=1)           # height-ordered, desc
+    # reference candidate order (postprocess.py:96 `np.argsort(h)[::-1]`):
+    # height descending with EXACT ties broken toward the larger sample
+    # index — stable-sort the flipped trace and map indices back (ties do
+    # occur in practice on saturated sigmoid outputs)
+    svals, sidx = heights.flip(1).sort(dim=1, descending=True, stable=True)
+    vals = svals[:, :k]
+    idx = (L - 1) - sidx[:, :k]
     valid = vals > float("-inf")
 
     if min_peak_dist > 1:

@@ -89,3 +89,63 @@ def test_pick_phase_matches_reference_loops():
     a = _pick_phase(out, 0.6, 25, 3, -7)
     b = ref._pick_phase(out, 0.6, 25, 3, -7)
     assert torch.equal(a, b)
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_pick_phase_matches_reference_sweep():
+    """Batched picker vs reference loop across shapes, thresholds and
+    adversarial inputs (boundary peaks, plateaus, saturated, ties)."""
+    ref = load_ref_module("training/postprocess.py", "ref_post_sweep")
+    rng = np.random.default_rng(7)
+    cases = [torch.tensor(rng.random((n, L)), dtype=torch.float32)
+             for n, L in ((4, 128), (8, 300), (5, 512))]
+    adv = torch.zeros(6, 64)
+    adv[0, 0] = 1.0
+    adv[0, 63] = 0.9           # boundary peaks
+    adv[1, 10:20] = 0.8        # plateau of equal values
+    adv[2, :] = 0.99           # saturated trace
+    adv[3, 5] = adv[3, 25] = adv[3, 45] = 0.7  # exact-tie peaks
+    adv[5, 32] = 0.4999        # just below threshold
+    cases.append(adv)
+    for out in cases:
+        for th, mpd, topk in ((0.5, 10, 3), (0.3, 2, 2), (0.7, 50, 1)):
+            a = _pick_phase(out, th, mpd, topk, -7)
+            b = ref._pick_phase(out, th, mpd, topk, -7)
+            assert torch.equal(a, b), (tuple(out.shape), th, mpd, topk)
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_detect_event_matches_reference_loops():
+    """Batched run-table event detector vs the reference per-sample loop
+    (the reference's obspy trigger_onset is injected with our
+    obspy-equivalent implementation, so only the batching differs)."""
+    ref = load_ref_module("training/postprocess.py", "ref_post_det")
+    ref.trigger_onset = trigger_onset
+    rng = np.random.default_rng(11)
+    walks = np.clip(np.cumsum(rng.normal(0, 0.15, (8, 400)), axis=1)
+                    * 0.2 + 0.5, 0.0, 1.0).astype(np.float32)
+    cases = [torch.tensor(walks)]
+    adv = torch.zeros(5, 64)
+    adv[0, 0:64] = 0.9          # run covering the whole trace
+    adv[1, 0:5] = 0.9           # run starting at 0
+    adv[2, 60:64] = 0.9         # run ending at the last sample
+    adv[3, 10:20] = 0.9
+    adv[3, 30:40] = 0.8         # two equal-length runs (stable tie)
+    cases.append(adv)
+    for out in cases:
+        for th, topk in ((0.5, 2), (0.3, 1), (0.8, 4)):
+            a = _detect_event(out, th, topk)
+            b = ref._detect_event(out, th, topk)
+            assert torch.equal(a, b), (tuple(out.shape), th, topk)
+
+
+def test_pick_phase_mpd1_topk_no_crash():
+    """Divergence-by-bugfix: the reference applies topk only inside its
+    `if mpd > 1` block (reference training/postprocess.py:95-99) and then
+    crashes broadcasting >topk peaks into the topk-wide row
+    (postprocess.py:184). Our batched picker applies topk for any mpd."""
+    rng = np.random.default_rng(5)
+    out = torch.tensor(rng.random((4, 256)), dtype=torch.float32)
+    ph = _pick_phase(out, 0.3, 1, 2, -7)
+    assert ph.shape == (4, 2)
+    assert (ph != -7).any()
