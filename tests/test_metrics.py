@@ -119,3 +119,33 @@ def test_det_matches_reference():
     for n in names:
         assert ours.get_metric(n) == pytest.approx(theirs.get_metric(n),
                                                    abs=1e-6)
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+@pytest.mark.parametrize("seed", [1, 2, 3, 4])
+def test_phase_metrics_reference_sweep(seed):
+    """Batched greedy phase ordering (_order_phases, K argmin rounds) vs
+    the reference loop across random draws plus tie/padding adversaries:
+    all-padded rows, duplicate predicted times, equidistant pairs."""
+    refm = load_ref_utils().metrics
+    names = ["precision", "recall", "f1", "mean", "rmse", "mae", "mape"]
+    torch.manual_seed(seed)
+    K = 3
+    targets = torch.randint(-5, 8192, (12, K))
+    preds = targets + torch.randint(-12, 12, (12, K))
+    preds[0] = -10000000                  # all padding
+    preds[1, 1] = preds[1, 0]             # duplicate predicted time
+    targets[2, 1] = targets[2, 0]         # duplicate target time
+    if K >= 2:
+        # two preds exactly equidistant from one target (argmin tie)
+        targets[3, 0] = 1000
+        preds[3, 0] = 996
+        preds[3, 1] = 1004
+    ours = _ours("ppk", names)
+    ours.compute(targets.clone(), preds.clone())
+    theirs = refm.Metrics(task="ppk", metric_names=names, sampling_rate=50,
+                          time_threshold=0.1, num_samples=8192, device=DEV)
+    theirs.compute(targets.clone(), preds.clone())
+    for n in names:
+        assert ours.get_metric(n) == pytest.approx(
+            theirs.get_metric(n), rel=1e-5, abs=1e-6), (seed, n)
