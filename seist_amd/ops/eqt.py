@@ -63,6 +63,8 @@ class _LayerNorm(torch.autograd.Function):
 def layer_norm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
                eps: float = 1e-5) -> torch.Tensor:
     """LayerNorm over the last dimension (EQT transformer stage, K8)."""
+    if x.device.type == "cpu" and torch.jit.is_tracing():
+        return F.layer_norm(x, x.shape[-1:], gamma, beta, eps)
     return _LayerNorm.apply(x.contiguous(), gamma, beta, eps)
 
 

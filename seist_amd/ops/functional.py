@@ -25,6 +25,15 @@ _ACT_NONE = 0
 _ACT_GELU = 1
 _ACT_RELU = 2
 
+def _trace_eager(x: torch.Tensor) -> bool:
+    """True while torch.jit is tracing CPU tensors: run the plain eager
+    composite instead of the autograd.Function wrapper (the tracer cannot
+    trace custom Functions whose inputs are nn.Parameters). Keeps the
+    models jit-traceable like the reference (its explicit-padding design
+    exists for torch.jit/onnx export, reference models/seist.py:24)."""
+    return x.device.type == "cpu" and torch.jit.is_tracing()
+
+
 # ---------------------------------------------------------------------------
 # padding semantics (reference models/seist.py:12-48 `_auto_pad_1d`)
 # ---------------------------------------------------------------------------
@@ -101,6 +110,8 @@ def pointwise_conv(x: torch.Tensor, weight: torch.Tensor,
         weight = weight.squeeze(-1)
     if x.dtype != weight.dtype:
         x = x.to(weight.dtype)
+    if _trace_eager(x):
+        return F.conv1d(x, weight.unsqueeze(-1), bias)
     return _PointwiseConv.apply(x.contiguous(), weight.contiguous(), bias)
 
 
@@ -227,6 +238,10 @@ def conv1d(x: torch.Tensor, weight: torch.Tensor,
     if (weight.size(-1) == 1 and stride == 1 and padl == 0 and padr == 0
             and groups == 1):
         return pointwise_conv(x, weight, bias)
+    if _trace_eager(x):
+        xp = F.pad(x, (padl, padr)) if (padl or padr) else x
+        return F.conv1d(xp, weight, bias, stride=stride, groups=groups,
+                        dilation=dilation)
     return _Conv1d.apply(x.contiguous(), weight.contiguous(), bias, stride,
                          padl, padr, groups, dilation)
 
@@ -424,6 +439,14 @@ def bn_act(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
     are reduced across ranks (one (C,2)-float collective before finalize,
     one more in backward) — K7's SyncBN obligation.
     """
+    if _trace_eager(x):
+        y = F.batch_norm(x.float(), running_mean, running_var, gamma.float(),
+                         beta.float(), training, momentum, eps)
+        if act == "gelu":
+            y = F.gelu(y)
+        elif act == "relu":
+            y = F.relu(y)
+        return y.to(x.dtype)
     act_id = {"none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}[act]
     return _BNAct.apply(x.contiguous(), gamma, beta, running_mean, running_var,
                         training, momentum, eps, act_id, sync, process_group,
@@ -475,6 +498,9 @@ def avgmax_pool1d(x: torch.Tensor, k: int) -> torch.Tensor:
         # reference LocalAwareAggregationBlock builds no pools for k == 1
         # (models/seist.py:79-84) — identity, not avg+max.
         return x
+    if _trace_eager(x):
+        return (F.avg_pool1d(x, k, ceil_mode=True)
+                + F.max_pool1d(x, k, ceil_mode=True))
     return _AvgMaxPool.apply(x.contiguous(), k)
 
 
@@ -509,6 +535,8 @@ def max_pool1d(x: torch.Tensor, k: int,
                ceil_mode: bool = False) -> torch.Tensor:
     """MaxPool1d with stride == kernel (K12): EQT/MagNet/DiTingMotion
     encoder pools."""
+    if _trace_eager(x):
+        return F.max_pool1d(x, k, ceil_mode=ceil_mode)
     return _MaxPool1d.apply(x.contiguous(), k, ceil_mode)
 
 
@@ -531,6 +559,8 @@ class _GlobalAvgPool(torch.autograd.Function):
 
 def global_avg_pool1d(x: torch.Tensor) -> torch.Tensor:
     """AdaptiveAvgPool1d(1) (K12): classification/regression heads."""
+    if _trace_eager(x):
+        return x.mean(dim=-1, keepdim=True)
     return _GlobalAvgPool.apply(x.contiguous())
 
 
@@ -565,6 +595,9 @@ def interp_linear(x: torch.Tensor, out_len: int) -> torch.Tensor:
     """K13: 1d linear resize to arbitrary size."""
     if out_len == x.size(-1):
         return x
+    if _trace_eager(x):
+        return F.interpolate(x, size=out_len, mode="linear",
+                             align_corners=False)
     return _InterpLinear.apply(x.contiguous(), out_len)
 
 
@@ -660,6 +693,8 @@ def droppath_add(x: torch.Tensor, y: torch.Tensor, drop_prob: float,
     """Fused residual + stochastic depth: replaces ``x + DropPath(p)(y)``
     (one elementwise pass; the reference chain is bernoulli/div/mul/add)."""
     if drop_prob == 0.0 or not training:
+        if _trace_eager(x):
+            return x + y
         return _DropPathAdd.apply(x.contiguous(), y.contiguous(), None, 1.0)
     keep = 1.0 - drop_prob
     mask = torch.bernoulli(
@@ -704,6 +739,8 @@ def conv_transpose1d(x: torch.Tensor, weight: torch.Tensor,
                      bias: Optional[torch.Tensor] = None,
                      stride: int = 1) -> torch.Tensor:
     """ConvTranspose1d, weight (Ci, Co, K), no padding (PhaseNet's usage)."""
+    if _trace_eager(x):
+        return F.conv_transpose1d(x, weight, bias, stride=stride)
     return _ConvTranspose1d.apply(x.contiguous(), weight.contiguous(), bias,
                                   stride)
 
@@ -730,6 +767,8 @@ class _Upsample2x(torch.autograd.Function):
 
 def upsample2x(x: torch.Tensor) -> torch.Tensor:
     """Nearest-neighbour x2 upsample along the last dim."""
+    if _trace_eager(x):
+        return F.interpolate(x, scale_factor=2, mode="nearest")
     return _Upsample2x.apply(x.contiguous())
 
 
