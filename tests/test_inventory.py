@@ -1,6 +1,8 @@
 """Component-inventory audit: pins the SURVEY.md §2 capability surface so
 a regression that silently drops a model/dataset/loss/flag fails loudly."""
 
+import os
+
 import pytest
 
 
@@ -125,3 +127,37 @@ def test_distributed_surface_complete():
                "gather_tensors_to_list", "broadcast_object", "barrier"):
         assert hasattr(pdist, fn), fn
     assert FlatReplica and enable_native_syncbn and wrap_distributed
+
+
+def test_cli_defaults_match_reference():
+    """Every reference CLI flag exists here with the same default
+    (reference main.py:8-179), modulo two deliberate divergences:
+    the dataset path (environment) and use_torch_compile (the reference
+    defaults its CUDA compile stack on; the MI355X path uses hipGraphs +
+    native kernels, torch.compile stays opt-in)."""
+    import sys
+    import types
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from _refload import load_ref_module, reference_available
+    if not reference_available():
+        import pytest
+        pytest.skip("reference absent")
+    if "torch.utils.tensorboard" not in sys.modules:
+        tb = types.ModuleType("torch.utils.tensorboard")
+        tb.SummaryWriter = object
+        sys.modules["torch.utils.tensorboard"] = tb
+    argv = sys.argv
+    sys.argv = ["main.py"]
+    try:
+        ref = load_ref_module("main.py", "ref_main_cli")
+        theirs = vars(ref.get_args())
+    finally:
+        sys.argv = argv
+    from seist_amd.cli import get_args
+    ours = vars(get_args([]))
+    missing = set(theirs) - set(ours)
+    assert not missing, f"reference flags missing here: {sorted(missing)}"
+    allowed = {"data", "use_torch_compile"}
+    diffs = {k for k in set(theirs) & set(ours)
+             if theirs[k] != ours[k] and k not in allowed}
+    assert not diffs, f"default drift: {sorted(diffs)}"
