@@ -193,3 +193,39 @@ def test_bf16_precision_train_cpu(tmp_path):
     import numpy as np
     for f in losses:
         assert np.isfinite(np.load(f)).all()
+
+
+def test_scheduler_matches_reference_conversion():
+    """build_scheduler's fractional warmup/down conversion and CyclicLR
+    parameters reproduce the reference's LR sequence exactly
+    (reference training/train.py:328-354)."""
+    import pytest
+    from types import SimpleNamespace
+    from seist_amd.engine.train import build_scheduler
+    for (w, d, steps) in ((2.0, 3.0, 100), (0.2, 0.3, 50), (0.0, 0.0, 40),
+                          (5.0, 0.5, 64)):
+        args = SimpleNamespace(use_lr_scheduler=True, warmup_steps=w,
+                               down_steps=d, steps=steps, base_lr=8e-5,
+                               max_lr=1e-3, lr_scheduler_mode="exp_range",
+                               start_epoch=0)
+        opt = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=8e-5)
+        sch = build_scheduler(args, opt, steps_per_epoch=10)
+        ours = []
+        for _ in range(steps):
+            opt.step()
+            sch.step()
+            ours.append(sch.get_last_lr()[0])
+        rw = w if w >= 1 else (int(steps * w) if w > 0 else 1)
+        rd = d if d >= 1 else (int(steps * d) if d > 0 else steps - rw)
+        opt2 = torch.optim.SGD([torch.nn.Parameter(torch.zeros(1))], lr=8e-5)
+        ref = torch.optim.lr_scheduler.CyclicLR(
+            opt2, base_lr=8e-5, max_lr=1e-3, step_size_up=int(rw),
+            step_size_down=int(rd), mode="exp_range",
+            gamma=8e-5 ** ((steps * 2) ** -1), cycle_momentum=False,
+            last_epoch=-1)
+        theirs = []
+        for _ in range(steps):
+            opt2.step()
+            ref.step()
+            theirs.append(ref.get_last_lr()[0])
+        assert ours == pytest.approx(theirs, rel=1e-12), (w, d)
