@@ -149,3 +149,51 @@ def test_pick_phase_mpd1_topk_no_crash():
     ph = _pick_phase(out, 0.3, 1, 2, -7)
     assert ph.shape == (4, 2)
     assert (ph != -7).any()
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_result_saver_csv_matches_reference(tmp_path):
+    """ResultSaver CSV format parity: same columns, same cell encoding
+    (onehot argmax, ppk/spk padding removal, comma-joined multi-values)
+    as the reference saver (training/postprocess.py:253-338)."""
+    import pandas as pd
+    from seist_amd.engine.postprocess import ResultSaver
+
+    ref = load_ref_module("training/postprocess.py", "ref_saver")
+    meta = {"key": ["a", "b", "c"], "snr": [1.5, 2.5, 3.5]}
+    targets = {
+        "ppk": torch.tensor([[100, 200], [300, -10000000], [-1, -1]]),
+        "spk": torch.tensor([[150, -10000000], [350, 400], [500, 600]]),
+        "pmp": torch.tensor([[0.9, 0.1], [0.2, 0.8], [0.6, 0.4]]),
+    }
+    results = {
+        "ppk": torch.tensor([[110, -10000000], [290, 310], [-2, -2]]),
+        "spk": torch.tensor([[160, 170], [-10000000, -10000000],
+                             [510, 590]]),
+        "pmp": torch.tensor([[0.3, 0.7], [0.55, 0.45], [0.1, 0.9]]),
+    }
+    names = ["ppk", "spk", "pmp"]
+
+    ours = ResultSaver(item_names=list(names))
+    ours.append(dict(meta), {k: v.clone() for k, v in targets.items()},
+                {k: v.clone() for k, v in results.items()})
+    p1 = tmp_path / "ours.csv"
+    ours.save_as_csv(str(p1))
+
+    theirs = ref.ResultSaver(item_names=list(names))
+    theirs.append(dict(meta), {k: v.clone() for k, v in targets.items()},
+                  {k: v.clone() for k, v in results.items()})
+    p2 = tmp_path / "theirs.csv"
+    theirs.save_as_csv(str(p2))
+
+    a = pd.read_csv(p1)
+    b = pd.read_csv(p2)
+    assert list(a.columns) == list(b.columns)
+    for col in a.columns:
+        av = a[col].fillna("").astype(str).tolist()
+        bv = b[col].fillna("").astype(str).tolist()
+        # divergence-by-bugfix: the reference's `v[i]==""` typo
+        # (training/postprocess.py:266 — comparison, not assignment)
+        # leaves empty pick lists as the literal "[]"; ours writes ""
+        bv = ["" if x == "[]" else x for x in bv]
+        assert av == bv, col
