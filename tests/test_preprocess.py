@@ -374,3 +374,26 @@ def test_native_dataset_picklable():
     _ = ds[0]
     ds2 = pickle.loads(pickle.dumps(ds))
     _ = ds2[1]
+
+
+@pytest.mark.skipif(not reference_available(), reason="reference absent")
+def test_cal_snr_matches_reference():
+    """SNR helper vs reference utils/misc.py:228-274 across 1-3 channel
+    data, boundary picks, and zero-noise windows."""
+    from seist_amd.utils import cal_snr
+    ref = load_ref_module("utils/misc.py", "ref_misc_snr")
+    rng = np.random.default_rng(2)
+    cases = []
+    for pick in (1000, 50, 5, 1900, 1999):
+        cases.append((rng.standard_normal((3, 2000)).astype(np.float32),
+                      pick))
+    flat = np.zeros((3, 2000), dtype=np.float32)
+    flat[:, 1200:1400] = 1.0
+    cases.append((flat, 1200))
+    for data, pick in cases:
+        a = cal_snr(data.copy(), pick)
+        b = ref.cal_snr(data.copy(), pick)
+        a = np.asarray(a, dtype=np.float64)
+        b = np.asarray(b, dtype=np.float64)
+        assert a.shape == b.shape, pick
+        assert np.allclose(a, b, equal_nan=True), (pick, a, b)
