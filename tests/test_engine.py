@@ -229,3 +229,20 @@ def test_scheduler_matches_reference_conversion():
             ref.step()
             theirs.append(ref.get_last_lr()[0])
         assert ours == pytest.approx(theirs, rel=1e-12), (w, d)
+
+
+def test_standalone_test_mode(tmp_path):
+    """--mode test --checkpoint ... (the reference README's test recipe,
+    README.md:186-224): builds the test split, loads the checkpoint, and
+    writes the results CSV."""
+    args = _args(tmp_path, ["--model-name", "phasenet", "--mode", "train"])
+    args.distributed = False
+    main_worker(args, torch.device("cpu"))
+    ckpts = sorted(glob.glob(str(tmp_path / "*" / "checkpoints" / "*.pth")))
+    assert ckpts
+    args2 = _args(tmp_path, ["--model-name", "phasenet", "--mode", "test"])
+    args2.checkpoint = ckpts[-1]
+    args2.distributed = False
+    main_worker(args2, torch.device("cpu"))
+    csvs = glob.glob(str(tmp_path / "*" / "test_results_*.csv"))
+    assert len(csvs) >= 1
