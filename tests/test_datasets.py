@@ -78,3 +78,64 @@ def test_sos_reader_feeds_preprocessor(sos_root):
     assert t.shape[-1] == 1024  # soft labels rasterized to the window
     import json
     assert json.loads(meta_json)["fname"] == "ev0.npz"
+
+
+def test_diting_light_reader_with_fake_h5(tmp_path, monkeypatch):
+    """DiTing reader logic without real archives: a fake h5py exercises
+    the key zero-padding quirk ("123.45" -> "000123.4500", reference
+    datasets/diting.py:137-138), channel transpose, polarity/clarity
+    encoding, baz wrap and ms->ml magnitude conversion."""
+    import seist_amd.data.diting as dt
+
+    cols = {
+        "part": [0, 0], "key": ["123.45", "7.1"], "ev_id": [1, 2],
+        "evmag": ["3.0", "2.0"], "mag_type": ["ms", "ml"],
+        "p_pick": [4000, 5000], "p_clarity": ["I", "E"],
+        "p_motion": ["U", "D"], "s_pick": [6000, 7000],
+        "net": ["AA", "BB"], "sta_id": [1, 2], "dis": [10.0, 20.0],
+        "st_mag": ["3.0", "2.0"], "baz": ["370.0", "45.0"],
+        "P_residual": ["0", "0"], "S_residual": ["0", "0"],
+    }
+    for c in ("Z_P", "Z_S", "N_P", "N_S", "E_P", "E_S"):
+        cols[f"{c}_amplitude_snr"] = [5.0, 6.0]
+        cols[f"{c}_power_snr"] = [5.0, 6.0]
+    pd.DataFrame(cols).to_csv(tmp_path / "DiTing330km_light.csv")
+
+    requested = []
+
+    class FakeFile:
+        def __init__(self, path, mode):
+            assert path.endswith("DiTing330km_part_0.hdf5")
+
+        def __enter__(self):
+            return self
+
+        def __exit__(self, *a):
+            return False
+
+        def get(self, key):
+            requested.append(key)
+            rng = np.random.default_rng(0)
+            return rng.standard_normal((8192, 3)).astype(np.float64)
+
+    import types
+    fake = types.SimpleNamespace(File=FakeFile)
+    monkeypatch.setattr(dt, "_h5py", lambda: fake)
+
+    ds = build_dataset("diting_light", seed=0, mode="train",
+                       data_dir=str(tmp_path), shuffle=False,
+                       data_split=False)
+    assert len(ds) == 2
+    ev, meta = ds[0]
+    assert requested[-1] == "earthquake/000123.4500"   # padding quirk
+    assert ev["data"].shape == (3, 8192)               # transposed
+    assert ev["ppks"] == [4000] and ev["spks"] == [6000]
+    assert ev["pmp"] == [0] and ev["clr"] == [0]       # U->0, I->0
+    assert ev["baz"] == [10.0]                         # 370 % 360
+    # ms->ml magnitude conversion: (3.0 + 1.08) / 1.13
+    assert ev["emg"][0] == pytest.approx((3.0 + 1.08) / 1.13, rel=1e-6)
+
+    ev2, _ = ds[1]
+    assert requested[-1] == "earthquake/000007.1000"
+    assert ev2["pmp"] == [1] and ev2["clr"] == [1]     # D->1, E->1
+    assert ev2["emg"][0] == pytest.approx(2.0)         # ml unchanged
