@@ -139,3 +139,62 @@ def test_diting_light_reader_with_fake_h5(tmp_path, monkeypatch):
     assert requested[-1] == "earthquake/000007.1000"
     assert ev2["pmp"] == [1] and ev2["clr"] == [1]     # D->1, E->1
     assert ev2["emg"][0] == pytest.approx(2.0)         # ml unchanged
+
+
+def test_pnw_reader_with_fake_h5(tmp_path, monkeypatch):
+    """PNW reader logic without the archive: seisbench trace_name parsing
+    ("bucket$n,:c,:l" -> data/<bucket>[n]), NaN zeroing, 4-class polarity
+    map, and the '|'-joined snr string parse (reference
+    datasets/pnw.py:102-138)."""
+    import seist_amd.data.pnw as pw
+
+    df = pd.DataFrame({
+        "trace_name": ["bucket3$7,:3,:6000", "bucket1$0,:3,:6000"],
+        "trace_P_polarity": ["positive", "undecidable"],
+        "preferred_source_magnitude_type": ["ml", "ml"],
+        "preferred_source_magnitude": [9.5, 2.0],   # clipped to 8
+        "trace_snr_db": ["10.0|nan|12.5", "3.0|4.0|5.0"],
+        "trace_P_arrival_sample": [1000, 2000],
+        "trace_S_arrival_sample": [1500, 2500],
+    })
+    df.to_csv(tmp_path / "comcat_metadata.csv", index=False)
+
+    class FakeDset:
+        def __init__(self, bucket):
+            self.bucket = bucket
+
+        def __getitem__(self, n):
+            rng = np.random.default_rng(n)
+            arr = rng.standard_normal((3, 6000)).astype(np.float64)
+            arr[0, 0] = np.nan          # reader must nan_to_num
+            return arr
+
+    class FakeFile:
+        def __init__(self, path, mode):
+            assert path.endswith("comcat_waveforms.hdf5")
+
+        def __enter__(self):
+            return self
+
+        def __exit__(self, *a):
+            return False
+
+        def get(self, key):
+            assert key.startswith("data/bucket")
+            return FakeDset(key)
+
+    import types
+    monkeypatch.setattr(pw, "_h5py",
+                        lambda: types.SimpleNamespace(File=FakeFile))
+
+    ds = build_dataset("pnw", seed=0, mode="train", data_dir=str(tmp_path),
+                       shuffle=False, data_split=False)
+    assert len(ds) == 2
+    ev, meta = ds[0]
+    assert ev["data"].shape == (3, 6000)
+    assert np.isfinite(ev["data"]).all()         # NaN zeroed
+    assert ev["pmp"] == [0]                      # positive -> 0
+    assert float(ev["emg"][0]) == 8.0            # clipped
+    assert ev["snr"].tolist() == [10.0, 0.0, 12.5]
+    ev2, _ = ds[1]
+    assert ev2["pmp"] == [2]                     # undecidable -> 2
