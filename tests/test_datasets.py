@@ -198,3 +198,50 @@ def test_pnw_reader_with_fake_h5(tmp_path, monkeypatch):
     assert ev["snr"].tolist() == [10.0, 0.0, 12.5]
     ev2, _ = ds[1]
     assert ev2["pmp"] == [2]                     # undecidable -> 2
+
+
+def test_diting_multipart_concat_and_split(tmp_path, monkeypatch):
+    """Full DiTing variant: multi-part CSV concat, deterministic
+    seed-shuffle and train/val/test row split."""
+    import seist_amd.data.diting as dt
+
+    def part_df(part, n):
+        cols = {
+            "part": [part] * n, "key": [f"{100 + 50 * part + i}.0" for i in range(n)],
+            "ev_id": list(range(n)), "evmag": ["1.0"] * n,
+            "mag_type": ["ml"] * n, "p_pick": [4000] * n,
+            "p_clarity": ["I"] * n, "p_motion": ["U"] * n,
+            "s_pick": [6000] * n, "net": ["AA"] * n,
+            "sta_id": list(range(n)), "dis": [1.0] * n,
+            "st_mag": ["1.0"] * n, "baz": ["0.0"] * n,
+            "P_residual": ["0"] * n, "S_residual": ["0"] * n,
+        }
+        for c in ("Z_P", "Z_S", "N_P", "N_S", "E_P", "E_S"):
+            cols[f"{c}_amplitude_snr"] = [5.0] * n
+            cols[f"{c}_power_snr"] = [5.0] * n
+        return pd.DataFrame(cols)
+
+    part_df(0, 6).to_csv(tmp_path / "DiTing330km_part_0.csv")
+    part_df(1, 4).to_csv(tmp_path / "DiTing330km_part_1.csv")
+    monkeypatch.setattr(dt.DiTing, "_part_range", (0, 2))
+
+    tr = build_dataset("diting", seed=3, mode="train",
+                       data_dir=str(tmp_path), shuffle=True, data_split=True,
+                       train_size=0.8, val_size=0.1)
+    va = build_dataset("diting", seed=3, mode="val", data_dir=str(tmp_path),
+                       shuffle=True, data_split=True, train_size=0.8,
+                       val_size=0.1)
+    te = build_dataset("diting", seed=3, mode="test", data_dir=str(tmp_path),
+                       shuffle=True, data_split=True, train_size=0.8,
+                       val_size=0.1)
+    assert len(tr) + len(va) + len(te) == 10
+    assert len(tr) == 8 and len(va) == 1 and len(te) == 1
+    # deterministic: same seed -> identical split membership
+    tr2 = build_dataset("diting", seed=3, mode="train",
+                        data_dir=str(tmp_path), shuffle=True,
+                        data_split=True, train_size=0.8, val_size=0.1)
+    keys = lambda ds: [ds._meta_data.iloc[i]["key"] for i in range(len(ds))]
+    assert keys(tr) == keys(tr2)
+    # disjointness across modes
+    all_keys = keys(tr) + keys(va) + keys(te)
+    assert len(set(all_keys)) == 10
