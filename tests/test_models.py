@@ -206,3 +206,21 @@ def test_seist_l_five_heads_train_step(name):
     assert torch.isfinite(loss).item()
     grads = [p.grad for p in m.parameters() if p.requires_grad]
     assert all(g is not None and torch.isfinite(g).all() for g in grads)
+
+
+def test_torch_compile_smoke():
+    """--use-torch-compile parity capability: dynamo traces through the
+    custom autograd Functions (fwd + bwd). Small model/window to bound
+    the CPU inductor compile time."""
+    import warnings
+    import torch
+    from seist_amd.models import create_model
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        m = create_model("phasenet", in_channels=3, in_samples=1024)
+        mc = torch.compile(m)
+        x = torch.randn(2, 3, 1024)
+        y = mc(x)
+        assert y.shape == (2, 3, 1024)
+        y.float().pow(2).mean().backward()
+        assert all(p.grad is not None for p in m.parameters())
