@@ -210,15 +210,16 @@ def test_seist_l_five_heads_train_step(name):
 
 def test_torch_compile_smoke():
     """--use-torch-compile parity capability: dynamo traces through the
-    custom autograd Functions (fwd + bwd). Small model/window to bound
-    the CPU inductor compile time."""
+    custom autograd Functions (fwd + bwd). backend="eager" exercises the
+    dynamo capture (the part our op layer could break) without paying
+    ~2 min of CPU inductor codegen, which is torch's own machinery."""
     import warnings
     import torch
     from seist_amd.models import create_model
     with warnings.catch_warnings():
         warnings.simplefilter("ignore")
         m = create_model("phasenet", in_channels=3, in_samples=1024)
-        mc = torch.compile(m)
+        mc = torch.compile(m, backend="eager")
         x = torch.randn(2, 3, 1024)
         y = mc(x)
         assert y.shape == (2, 3, 1024)
